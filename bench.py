@@ -1,0 +1,185 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: the J1644-4559 single-pulse pipeline on MI355X.
+
+One step = one full baseband block (default 2^30 samples, 2-bit, the
+J1644-4559 observation config of BASELINE.json) through the native engine:
+
+  H2D → unpack → R2C FFT (2^30) → RFI s1 + manual zap + coherent dedispersion
+  (fused) → waterfall batched C2C (2048 ch) → spectral-kurtosis RFI →
+  time-series + boxcar single-pulse detection → detection-counter readback,
+
+with double-buffered slots (2 HIP streams/GPU).  Multi-GPU: one rank per GPU
+(torch.distributed over RCCL/xGMI), each rank processing its own baseband
+stream (weak scaling — the reference shards streams/polarizations/beams the
+same way) and all-reducing the per-block detection counters.
+
+Usage: python bench.py --gpus N --steps K --warmup W
+(For N>1 the driver launches via torch.distributed.run; RANK/LOCAL_RANK/
+WORLD_SIZE/MASTER_* are read from the environment.)
+
+Prints ONE JSON line on rank 0 with the whole-job Msamples/s.
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import numpy as np
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=8)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--n", type=int, default=2**30,
+                    help="baseband_input_count per block (J1644: 2^30)")
+    ap.add_argument("--channels", type=int, default=2**11)
+    ap.add_argument("--bits", type=int, default=2)
+    ap.add_argument("--phase-table", action="store_true",
+                    help="cache dedispersion phase factors (fixed DM); "
+                    "default recomputes fp64 phase per block like the reference")
+    ap.add_argument("--no-rfi", action="store_true")
+    args = ap.parse_args()
+
+    import torch
+
+    world_size = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    n_gpus = world_size if world_size > 1 else args.gpus
+    distributed = world_size > 1
+
+    assert torch.cuda.is_available(), "bench.py requires a GPU"
+    torch.cuda.set_device(local_rank)
+
+    if distributed:
+        torch.distributed.init_process_group(backend="nccl")
+
+    from srtb_amd.ops import native
+    C = native()
+
+    # J1644-4559 observation config (reference userspace/srtb_config_1644-4559.cfg)
+    n = args.n
+    s = args.channels
+    freq_low = 1437.0     # 1405 + 64/2
+    bandwidth = -64.0
+    sample_rate = 128e6
+    dm = -478.80
+    rfi_threshold = 1.5
+    sk_threshold = 1.05
+    snr = 8.0
+    max_boxcar = 256
+    nc = n // 2
+    # manual zap 1418-1422 MHz → bin range (negative bandwidth)
+    lo = round((1422.0 - freq_low) / bandwidth * (nc - 1))
+    hi = round((1418.0 - freq_low) / bandwidth * (nc - 1))
+    zap_ranges = [[int(lo), int(hi)]]
+
+    eng = C.PipelineEngine(
+        n=n, nbits=args.bits, channels=s, freq_low=freq_low,
+        bandwidth=bandwidth, sample_rate=sample_rate, dm=dm,
+        rfi_threshold=rfi_threshold, sk_threshold=sk_threshold,
+        snr_threshold=snr, max_boxcar=max_boxcar, nsamps_reserved=0,
+        zap_ranges=zap_ranges, use_phase_table=args.phase_table,
+        enable_rfi_s1=not args.no_rfi, enable_sk=True, n_slots=2)
+
+    # synthetic 2-bit baseband noise, pinned, one buffer per slot
+    rng = np.random.default_rng(1234 + rank)
+    raw_bytes = eng.raw_bytes
+    pinned = []
+    for _ in range(2):
+        t = torch.from_numpy(rng.integers(0, 256, raw_bytes, dtype=np.uint8))
+        pinned.append(t.pin_memory())
+
+    counts_dev = torch.zeros(4, dtype=torch.int64, device="cuda")
+
+    def agg(res):
+        c = [res["zero_count"]] + [cnt for _, cnt in res["counts"][:3]]
+        counts_dev.copy_(torch.tensor(c, dtype=torch.int64))
+        if distributed:
+            torch.distributed.all_reduce(counts_dev)
+        return counts_dev
+
+    def run_steps(k):
+        prev = None
+        for i in range(k):
+            slot = eng.submit(pinned[i % 2])
+            if prev is not None:
+                agg(eng.wait(prev))
+            prev = slot
+        agg(eng.wait(prev))
+        eng.synchronize()
+
+    # warmup
+    run_steps(args.warmup)
+
+    if distributed:
+        torch.distributed.barrier()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    run_steps(args.steps)
+    torch.cuda.synchronize()
+    if distributed:
+        torch.distributed.barrier()
+    t1 = time.perf_counter()
+
+    elapsed = t1 - t0
+    # max over ranks
+    if distributed:
+        e = torch.tensor([elapsed], dtype=torch.float64, device="cuda")
+        torch.distributed.all_reduce(e, op=torch.distributed.ReduceOp.MAX)
+        elapsed = float(e.item())
+
+    samples_total = float(n) * args.steps * n_gpus
+    msamps = samples_total / elapsed / 1e6
+    ms_per_step = elapsed / args.steps * 1e3
+    per_gpu_sps = float(n) * args.steps / elapsed
+    real_time_ratio = per_gpu_sps / sample_rate
+
+    if rank == 0:
+        out = {
+            "metric": "baseband Msamples/sec/node + real-time ratio, "
+                      "J1644-4559 config at 1/2/4/8 MI355X",
+            "value": round(msamps, 1),
+            "unit": "Msamples/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            # BASELINE.md implied number: >= 1000 Msamples/s per stream
+            # (real-time at the 1 Gsps production config) on A40-class hardware
+            "vs_baseline": round(msamps / 1000.0, 2),
+            "dtype": "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": "J1644-4559 coherent-dedispersion single-pulse pipeline",
+                "baseband_input_count": n,
+                "baseband_input_bits": args.bits,
+                "spectrum_channel_count": s,
+                "dm": dm,
+                "freq_low_MHz": freq_low,
+                "bandwidth_MHz": bandwidth,
+                "sample_rate": sample_rate,
+                "rfi": (not args.no_rfi),
+                "phase_table": bool(args.phase_table),
+                "real_time_ratio_per_gpu": round(real_time_ratio, 1),
+                "parallelism": f"stream-sharded dp{n_gpus}",
+                "global_batch": n_gpus,
+                "seq_len": n,
+            },
+        }
+        print(json.dumps(out), flush=True)
+
+    if distributed:
+        torch.distributed.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

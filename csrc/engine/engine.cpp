@@ -1,0 +1,252 @@
+#include "engine.h"
+
+#include <cmath>
+#include <cstring>
+#include <stdexcept>
+
+namespace srtb_hip {
+
+namespace {
+constexpr double kD = 4.148808e3;  // dispersion constant (MHz^2 pc^-1 cm^3 s)
+}
+
+PipelineEngine::PipelineEngine(const EngineConfig& cfg, int n_slots)
+    : cfg_(cfg), n_slots_(n_slots) {
+  n_ = cfg.baseband_input_count;
+  nc_ = n_ / 2;
+  s_ = cfg.spectrum_channel_count;
+  if (s_ > nc_) s_ = nc_;  // reference watfft clamps (fft_pipe.hpp:300-306)
+  l_ = nc_ / s_;
+  const size_t reserved_bins = cfg.nsamps_reserved / s_;
+  ts_count_ = (l_ > reserved_bins) ? l_ - reserved_bins : l_;
+
+  const int bits = std::abs(cfg.baseband_input_bits);
+  raw_bytes_ = n_ * (size_t)bits / 8;
+
+  // boxcar ladder: 2, 4, ..., while <= max && < ts_count
+  for (size_t L = 2; L <= cfg.max_boxcar_length && L < ts_count_; L *= 2)
+    boxcar_lengths_.push_back(L);
+  n_boxcars_ = (int)boxcar_lengths_.size();
+
+  // SK corrected thresholds (reference rfi_mitigation.hpp:292-307)
+  {
+    const double M = (double)l_;
+    double hi = cfg.sk_threshold, lo = 2.0 - hi;
+    if (lo > hi) std::swap(lo, hi);
+    const double corr = (M - 1.0) / (M + 1.0);
+    sk_lo_ = (float)(lo * corr + 1.0);
+    sk_hi_ = (float)(hi * corr + 1.0);
+  }
+  // normalization coefficient (reference rfi_mitigation_pipe.hpp:60-66)
+  norm_coeff_ = (float)std::pow((double)nc_ * (double)nc_ / (double)s_, -0.5);
+
+  f_min_ = cfg.freq_low;
+  f_c_ = cfg.freq_low + cfg.bandwidth;
+  df_ = cfg.bandwidth / (double)nc_;
+
+  const int np = reduce_partials();
+  for (int i = 0; i < n_slots_; ++i) slots_.emplace_back(new Slot());
+  for (auto& sp : slots_) {
+    Slot& s = *sp;
+    check_hip(hipStreamCreateWithFlags(&s.stream, hipStreamNonBlocking),
+              "stream create");
+    check_hip(hipEventCreateWithFlags(&s.done, hipEventDisableTiming),
+              "event create");
+    check_hip(hipMalloc(&s.raw, raw_bytes_), "raw alloc");
+    check_hip(hipMalloc(&s.samples, n_ * sizeof(float)), "samples alloc");
+    check_hip(hipMalloc(&s.spec, (nc_ + 1) * sizeof(float2)), "spec alloc");
+    check_hip(hipMalloc(&s.s2s4, s_ * sizeof(float2)), "s2s4 alloc");
+    check_hip(hipMalloc(&s.flags, s_), "flags alloc");
+    check_hip(hipMalloc(&s.ts, ts_count_ * sizeof(float)), "ts alloc");
+    check_hip(hipMalloc(&s.cumsum, ts_count_ * sizeof(float)), "cumsum alloc");
+    check_hip(hipMalloc(&s.box, ts_count_ * sizeof(float)), "box alloc");
+    check_hip(hipMalloc(&s.scan_scratch, 4096 * sizeof(float)), "scan alloc");
+    check_hip(hipMalloc(&s.partials, np * sizeof(double)), "partials alloc");
+    check_hip(hipMalloc(&s.mean_power, sizeof(double)), "mean alloc");
+    check_hip(hipMalloc(&s.sums, 2 * sizeof(double)), "sums alloc");
+    const int ncnt = 1 + 1 + n_boxcars_;  // zero_count + raw + boxcars
+    check_hip(hipMalloc(&s.counters, ncnt * sizeof(unsigned)), "cnt alloc");
+    check_hip(hipMalloc(&s.thresholds, (1 + n_boxcars_) * sizeof(float)),
+              "thr alloc");
+    check_hip(hipHostMalloc(&s.h_counters, ncnt * sizeof(unsigned)),
+              "pinned cnt");
+    check_hip(hipHostMalloc(&s.h_thresholds, (1 + n_boxcars_) * sizeof(float)),
+              "pinned thr");
+    s.plans.create(n_, l_, s_, s.stream);
+  }
+
+  if (cfg.use_phase_table) {
+    check_hip(hipMalloc(&phase_table_, nc_ * sizeof(float2)), "table alloc");
+    check_hip(dedisp_phase_table(phase_table_, nc_, f_min_, f_c_, df_, cfg.dm,
+                                 slots_[0]->stream),
+              "phase table");
+    check_hip(hipStreamSynchronize(slots_[0]->stream), "table sync");
+  }
+}
+
+PipelineEngine::~PipelineEngine() {
+  for (auto& sp : slots_) {
+    if (sp->stream) hipStreamSynchronize(sp->stream);
+  }
+  for (auto& sp : slots_) {
+    Slot& s = *sp;
+    s.plans.destroy();
+    hipFree(s.raw);
+    hipFree(s.samples);
+    hipFree(s.spec);
+    hipFree(s.s2s4);
+    hipFree(s.flags);
+    hipFree(s.ts);
+    hipFree(s.cumsum);
+    hipFree(s.box);
+    hipFree(s.scan_scratch);
+    hipFree(s.partials);
+    hipFree(s.mean_power);
+    hipFree(s.sums);
+    hipFree(s.counters);
+    hipFree(s.thresholds);
+    hipHostFree(s.h_counters);
+    hipHostFree(s.h_thresholds);
+    if (s.done) hipEventDestroy(s.done);
+    if (s.stream) hipStreamDestroy(s.stream);
+  }
+  if (phase_table_) hipFree(phase_table_);
+}
+
+void PipelineEngine::enqueue_chain(Slot& s, const void* dev_raw) {
+  hipStream_t st = s.stream;
+  const uint8_t* raw = static_cast<const uint8_t*>(dev_raw);
+
+  // 1. unpack (+ window fused; default rectangle → none)
+  check_hip(unpack(raw, s.samples, n_, cfg_.baseband_input_bits, nullptr, st),
+            "unpack");
+  // 2. R2C forward (out-of-place; Nyquist bin written but ignored: the
+  //    downstream count is Nc — reference drops it, fft_pipe.hpp:77)
+  s.plans.exec_r2c(s.samples, s.spec);
+  // 3. mean |X|^2 over Nc
+  if (cfg_.enable_rfi_s1)
+    check_hip(mean_power(s.spec, nc_, s.partials, s.mean_power, st), "meanp");
+  // 4. fused RFI s1 + manual zap + dedispersion (single pass over 8·Nc bytes)
+  check_hip(rfi_dedisperse_fused(
+                s.spec, nc_, cfg_.enable_rfi_s1 ? s.mean_power : nullptr,
+                cfg_.rfi_threshold, norm_coeff_, cfg_.zap_ranges,
+                cfg_.n_zap_ranges, f_min_, f_c_, df_, cfg_.dm, phase_table_,
+                st),
+            "rfi+dedisp");
+  // 5. waterfall: batched backward C2C in place → [S][L]
+  s.plans.exec_c2c_backward(s.spec);
+  float2* wf = s.spec;
+
+  const int ncnt = 2 + n_boxcars_;
+  check_hip(hipMemsetAsync(s.counters, 0, ncnt * sizeof(unsigned), st),
+            "memset counters");
+
+  const uint8_t* ts_flags = nullptr;
+  if (cfg_.enable_sk) {
+    // 6. spectral kurtosis: row stats → flags (+zero count) → zap rows
+    check_hip(sk_row_stats(wf, s_, l_, s.s2s4, st), "sk stats");
+    check_hip(sk_flags(wf, s.s2s4, s_, l_, sk_lo_, sk_hi_, s.flags,
+                       s.counters + 0, st),
+              "sk flags");
+    check_hip(sk_zap_rows(wf, s.flags, s_, l_, st), "sk zap");
+    ts_flags = s.flags;
+  }
+  // 7. time series over non-zapped rows
+  check_hip(time_series(wf, ts_flags, s_, l_, ts_count_, s.ts, st), "ts");
+  // 8. baseline subtract
+  check_hip(sum_sumsq(s.ts, ts_count_, s.partials, s.sums, st), "ts sum");
+  check_hip(subtract_mean(s.ts, ts_count_, s.sums, st), "ts sub");
+  // 9. raw-series detection
+  check_hip(sum_sumsq(s.ts, ts_count_, s.partials, s.sums, st), "ts var");
+  check_hip(count_above(s.ts, ts_count_, s.sums + 1, cfg_.snr_threshold,
+                        s.counters + 1, s.thresholds + 0, st),
+            "count raw");
+  // 10. boxcar ladder from the inclusive scan
+  if (n_boxcars_ > 0) {
+    check_hip(inclusive_scan(s.ts, s.cumsum, ts_count_, s.scan_scratch, st),
+              "scan");
+    for (int b = 0; b < n_boxcars_; ++b) {
+      const size_t L = boxcar_lengths_[b];
+      const size_t n_out = ts_count_ - L;
+      check_hip(boxcar(s.cumsum, s.box, n_out, L, st), "boxcar");
+      check_hip(sum_sumsq(s.box, n_out, s.partials, s.sums, st), "box var");
+      check_hip(count_above(s.box, n_out, s.sums + 1, cfg_.snr_threshold,
+                            s.counters + 2 + b, s.thresholds + 1 + b, st),
+                "box count");
+    }
+  }
+  // 11. result counters → pinned host
+  check_hip(hipMemcpyAsync(s.h_counters, s.counters, ncnt * sizeof(unsigned),
+                           hipMemcpyDeviceToHost, st),
+            "res d2h");
+  check_hip(hipMemcpyAsync(s.h_thresholds, s.thresholds,
+                           (1 + n_boxcars_) * sizeof(float),
+                           hipMemcpyDeviceToHost, st),
+            "thr d2h");
+  check_hip(hipEventRecord(s.done, st), "event record");
+  s.busy = true;
+}
+
+int PipelineEngine::submit(const void* host_bytes, size_t nbytes) {
+  if (nbytes != raw_bytes_) throw std::runtime_error("submit: wrong size");
+  const int id = next_slot_;
+  next_slot_ = (next_slot_ + 1) % n_slots_;
+  Slot& s = *slots_[id];
+  if (s.busy) {
+    check_hip(hipEventSynchronize(s.done), "slot wait");
+    s.busy = false;
+  }
+  check_hip(hipMemcpyAsync(s.raw, host_bytes, nbytes, hipMemcpyHostToDevice,
+                           s.stream),
+            "raw h2d");
+  enqueue_chain(s, s.raw);
+  return id;
+}
+
+int PipelineEngine::submit_device(const void* dev_bytes, size_t nbytes) {
+  if (nbytes != raw_bytes_) throw std::runtime_error("submit: wrong size");
+  const int id = next_slot_;
+  next_slot_ = (next_slot_ + 1) % n_slots_;
+  Slot& s = *slots_[id];
+  if (s.busy) {
+    check_hip(hipEventSynchronize(s.done), "slot wait");
+    s.busy = false;
+  }
+  enqueue_chain(s, dev_bytes);
+  return id;
+}
+
+BlockResult PipelineEngine::wait(int slot) {
+  Slot& s = *slots_.at(slot);
+  check_hip(hipEventSynchronize(s.done), "wait");
+  s.busy = false;
+  BlockResult r;
+  r.zero_count = s.h_counters[0];
+  r.counts.emplace_back(1u, s.h_counters[1]);
+  for (int b = 0; b < n_boxcars_; ++b)
+    r.counts.emplace_back((unsigned)boxcar_lengths_[b], s.h_counters[2 + b]);
+  r.thresholds.assign(s.h_thresholds, s.h_thresholds + 1 + n_boxcars_);
+  return r;
+}
+
+void PipelineEngine::synchronize() {
+  for (auto& sp : slots_) {
+    check_hip(hipStreamSynchronize(sp->stream), "sync");
+    sp->busy = false;
+  }
+}
+
+float2* PipelineEngine::waterfall_ptr(int slot) { return slots_.at(slot)->spec; }
+float* PipelineEngine::time_series_ptr(int slot) { return slots_.at(slot)->ts; }
+float* PipelineEngine::cumsum_ptr(int slot) { return slots_.at(slot)->cumsum; }
+hipStream_t PipelineEngine::stream(int slot) { return slots_.at(slot)->stream; }
+
+float* PipelineEngine::compute_boxcar(int slot, size_t L) {
+  Slot& s = *slots_.at(slot);
+  const size_t n_out = ts_count_ - L;
+  check_hip(boxcar(s.cumsum, s.box, n_out, L, s.stream), "boxcar recompute");
+  check_hip(hipStreamSynchronize(s.stream), "boxcar sync");
+  return s.box;
+}
+
+}  // namespace srtb_hip

@@ -1,0 +1,143 @@
+// PipelineEngine — the MI355X-native per-GPU streaming engine.
+//
+// Runs the full per-block chain of the reference's streaming data path
+//  with zero host synchronization inside a block:
+//
+//   H2D(raw) → unpack(+window) → R2C FFT → mean-power reduce →
+//   fused {RFI-s1 zap + normalize + manual zap + coherent dedispersion} →
+//   batched backward C2C (waterfall) → SK row stats → flags → zap rows →
+//   time series → baseline subtract → threshold counts → prefix scan →
+//   boxcar ladder (2,4,…,max) with per-length thresholds →
+//   D2H(result counters) → event
+//
+// Two slots, each with its own HIP stream, device buffers and hipFFT plans:
+// while slot A computes, slot B's next block uploads (the reference instead
+// waits after every kernel — SURVEY.md §2b note).  Results (detection
+// counters) land in pinned memory; wait(slot) is the only host sync.
+
+#pragma once
+
+#include <hip/hip_runtime.h>
+
+#include <array>
+#include <cstdint>
+#include <memory>
+#include <vector>
+
+#include "../fft/fft_plans.h"
+#include "../include/srtb_kernels.h"
+
+namespace srtb_hip {
+
+struct EngineConfig {
+  size_t baseband_input_count = 1ull << 28;  // N real samples per block
+  int baseband_input_bits = 8;               // 1/2/4/8/-8/16/-16/32/-32
+  size_t spectrum_channel_count = 1ull << 15;  // S
+  double freq_low = 1000.0;    // MHz
+  double bandwidth = 500.0;    // MHz (may be negative)
+  double sample_rate = 1e9;    // samples/s
+  double dm = 0.0;
+  float rfi_threshold = 10.0f;          // mitigate_rfi_average_method_threshold
+  float sk_threshold = 1.1f;            // mitigate_rfi_spectral_kurtosis_threshold
+  float snr_threshold = 6.0f;           // signal_detect_signal_noise_threshold
+  size_t max_boxcar_length = 1024;
+  size_t nsamps_reserved = 0;           // overlap (real samples), from host
+  int n_zap_ranges = 0;
+  ZapRange zap_ranges[16] = {};
+  bool use_phase_table = false;  // cache dedispersion factors for fixed DM
+  bool enable_rfi_s1 = true;
+  bool enable_sk = true;
+};
+
+struct BlockResult {
+  unsigned zero_count = 0;  // zapped-channel count (gate detections on host)
+  // (boxcar_length, signal_count) — boxcar_length 1 is the raw series
+  std::vector<std::pair<unsigned, unsigned>> counts;
+  std::vector<float> thresholds;
+};
+
+class PipelineEngine {
+ public:
+  explicit PipelineEngine(const EngineConfig& cfg, int n_slots = 2);
+  ~PipelineEngine();
+  PipelineEngine(const PipelineEngine&) = delete;
+  PipelineEngine& operator=(const PipelineEngine&) = delete;
+
+  // Upload raw bytes (host, ideally pinned) and enqueue the whole chain on
+  // the slot's stream.  Blocks only if the slot's previous block is still in
+  // flight.  Returns the slot used.
+  int submit(const void* host_bytes, size_t nbytes);
+
+  // Enqueue the chain reading raw bytes already on the device (e.g. from a
+  // torch tensor); caller guarantees lifetime until wait().
+  int submit_device(const void* dev_bytes, size_t nbytes);
+
+  // Wait for a slot's chain and return its detection counters.
+  BlockResult wait(int slot);
+
+  // Wait for all outstanding work.
+  void synchronize();
+
+  // Geometry
+  size_t n() const { return n_; }
+  size_t nc() const { return nc_; }
+  size_t n_channels() const { return s_; }
+  size_t waterfall_len() const { return l_; }
+  size_t ts_count() const { return ts_count_; }
+  size_t raw_bytes() const { return raw_bytes_; }
+  int n_boxcars() const { return n_boxcars_; }
+  int n_slots() const { return n_slots_; }
+
+  // Device pointers of a slot (valid after wait(slot)) — for tests, dumps and
+  // the display path.
+  float2* waterfall_ptr(int slot);   // [S][L]
+  float* time_series_ptr(int slot);  // [ts_count]
+  float* cumsum_ptr(int slot);       // [ts_count]
+  // Recompute one boxcar series into the slot's boxcar buffer and return it
+  // (for dumping a detected series).
+  float* compute_boxcar(int slot, size_t boxcar_length);
+
+  hipStream_t stream(int slot);
+
+ private:
+  struct Slot;
+  void enqueue_chain(Slot& s, const void* dev_raw);
+
+  EngineConfig cfg_;
+  size_t n_, nc_, s_, l_, ts_count_, raw_bytes_;
+  int n_boxcars_ = 0;
+  std::vector<size_t> boxcar_lengths_;
+  float sk_lo_ = 0, sk_hi_ = 0;
+  float norm_coeff_ = 1.0f;
+  double f_min_ = 0, f_c_ = 0, df_ = 0;
+  int n_slots_ = 2;
+
+  float2* phase_table_ = nullptr;  // shared across slots (read-only)
+
+  struct Slot {
+    hipStream_t stream = nullptr;
+    hipEvent_t done = nullptr;
+    bool busy = false;
+    uint8_t* raw = nullptr;         // device raw bytes
+    float* samples = nullptr;       // [N] unpacked
+    float2* spec = nullptr;         // [Nc+1] spectrum / waterfall (in-place)
+    float2* s2s4 = nullptr;         // [S]
+    uint8_t* flags = nullptr;       // [S]
+    float* ts = nullptr;            // [ts_count]
+    float* cumsum = nullptr;        // [ts_count]
+    float* box = nullptr;           // [ts_count]
+    float* scan_scratch = nullptr;  // [4096]
+    double* partials = nullptr;     // reduce partials + out scalars
+    double* mean_power = nullptr;   // -> inside partials block
+    double* sums = nullptr;         // [2]
+    unsigned* counters = nullptr;   // [1 + n_boxcars + 1] zero_count + counts
+    float* thresholds = nullptr;    // [1 + n_boxcars]
+    unsigned* h_counters = nullptr; // pinned result mirror
+    float* h_thresholds = nullptr;  // pinned
+    FftPlanSet plans;
+  };
+  std::vector<std::unique_ptr<Slot>> slots_;
+  int next_slot_ = 0;
+};
+
+}  // namespace srtb_hip

@@ -1,0 +1,173 @@
+// srtb_amd — MI355X-native kernel API (CDNA4 / gfx950, wave64).
+//
+// Raw-pointer + hipStream_t interface so the same kernels serve the torch
+// extension, the native engine, and the standalone tools.  Each function
+// enqueues asynchronously on the given stream and returns the first HIP error.
+//
+// Semantics mirror the reference pipeline (citations per kernel in the .hip
+// files); the implementations are designed for gfx950: 64-wide wavefronts,
+// vectorized (float4 / uint4) global access, grid-stride loops capped per
+// CDNA guideline G11, two-pass deterministic reductions with fp64 partials.
+
+#pragma once
+
+#include <hip/hip_runtime.h>
+#include <cstddef>
+#include <cstdint>
+
+namespace srtb_hip {
+
+// ---------------- unpack (reference unpack.hpp:43-403) ----------------
+// nbits: 1, 2, 4 = unsigned MSB-first packed; 8/-8, 16/-16, 32/-32 = u/int
+// cast.  window (nullable) multiplies out[i] by window[i] (fused FFT window).
+hipError_t unpack(const uint8_t* in, float* out, size_t out_count, int nbits,
+                  const float* window, hipStream_t stream);
+
+// int8 samples interleaved per-sample: p0 s0, p1 s0, p0 s1, ...
+hipError_t unpack_interleaved_2pol(const int8_t* in, float* out0, float* out1,
+                                   size_t count_per_pol, const float* window,
+                                   hipStream_t stream);
+
+// SNAP-1 "1 1 2 2": groups of 4 int8 = [s0p0 s1p0 s0p1 s1p1]
+hipError_t unpack_naocpsr_snap1(const int8_t* in, float* out0, float* out1,
+                                size_t count_per_pol, const float* window,
+                                hipStream_t stream);
+
+// GZNU A1: 4-byte words cycling over n_streams (2 or 4) ADCs; 4-stream
+// variant is offset-binary (^0x80).  outs[s] may be null for unused streams.
+hipError_t unpack_gznupsr_a1(const uint8_t* in, float* out0, float* out1,
+                             float* out2, float* out3, int n_streams,
+                             size_t count_per_stream, const float* window,
+                             hipStream_t stream);
+
+// ---------------- reductions ----------------
+// Deterministic two-pass mean of |x|^2 over n complex bins.
+// partials: device scratch of n_partials doubles (n_partials = reduce grid),
+// out: 1 double = mean.  Use reduce_partials() to size the scratch.
+int reduce_partials();  // number of fp64 partials the reductions use
+
+hipError_t mean_power(const float2* in, size_t n, double* partials,
+                      double* out_mean, hipStream_t stream);
+
+// sum and sum-of-squares over a float array (two outputs in one pass):
+// out[0] = sum, out[1] = sum of squares
+hipError_t sum_sumsq(const float* in, size_t n, double* partials,
+                     double* out2, hipStream_t stream);
+
+// ---------------- RFI stage 1 + coherent dedispersion ----------------
+// Standalone RFI s1: zap |x|^2 > threshold*mean, else scale by norm_coeff
+// (reference rfi_mitigation_pipe.hpp:50-80).  mean is a device scalar.
+hipError_t rfi_s1(float2* spec, size_t n, const double* mean_power,
+                  float threshold, float norm_coeff, hipStream_t stream);
+
+// Manual zap of inclusive bin range (reference rfi_mitigation.hpp:97-157).
+hipError_t zap_bins(float2* spec, size_t lo, size_t hi, hipStream_t stream);
+
+// Standalone coherent dedispersion, fp64 phase (reference
+// coherent_dedispersion.hpp:133-248): x[i] *= exp(-2pi*i*frac(k)),
+// k = D*1e6*dm/f*((f-f_c)/f_c)^2, f = f_min + df*i.
+hipError_t dedisperse(float2* spec, size_t n, double f_min, double f_c,
+                      double df, double dm, hipStream_t stream);
+
+// Precompute the per-bin phase-factor table (table mode for fixed DM).
+hipError_t dedisp_phase_table(float2* table, size_t n, double f_min,
+                              double f_c, double df, double dm,
+                              hipStream_t stream);
+
+// Fused hot-path kernel: RFI-s1 zap + normalize + manual zap ranges +
+// dedispersion phase rotation in ONE pass over the spectrum (the reference
+// runs three separate kernels + waits; fusing removes two full HBM
+// round-trips of the 4 GB spectrum).  zap_ranges: up to 16 inclusive [lo,hi]
+// bin pairs, read from constant args.  factor_table: optional precomputed
+// phases (else computed on the fly in fp64).
+struct ZapRange { unsigned long long lo, hi; };
+hipError_t rfi_dedisperse_fused(float2* spec, size_t n,
+                                const double* mean_power, float threshold,
+                                float norm_coeff, const ZapRange* ranges,
+                                int n_ranges, double f_min, double f_c,
+                                double df, double dm,
+                                const float2* factor_table,
+                                hipStream_t stream);
+
+// ---------------- spectral kurtosis (stage 2) ----------------
+// Per-row <sum |x|^2, sum |x|^4> of a [rows][len] waterfall; one workgroup
+// per row (reference multi_mapreduce + rfi_mitigation.hpp:310-340).
+hipError_t sk_row_stats(const float2* wf, size_t rows, size_t len,
+                        float2* s2s4, hipStream_t stream);
+
+// Decide zap flags from SK statistic: flag[i] = SK outside [lo_, hi_]
+// (corrected thresholds precomputed on host), and count rows whose first
+// sample is zero OR flagged (the reference's zapped-channel count observes
+// column 0 after zapping).  zero_count must be zeroed beforehand.
+hipError_t sk_flags(const float2* wf, const float2* s2s4, size_t rows,
+                    size_t len, float lo_corrected, float hi_corrected,
+                    uint8_t* flags, unsigned* zero_count, hipStream_t stream);
+
+// Zero out flagged rows.
+hipError_t sk_zap_rows(float2* wf, const uint8_t* flags, size_t rows,
+                       size_t len, hipStream_t stream);
+
+// ---------------- signal detection ----------------
+// ts[j] = sum_i |wf[i][j]|^2 over non-flagged rows, j < ts_count
+// (reference signal_detect_pipe.hpp:305-316; flags may be null).
+hipError_t time_series(const float2* wf, const uint8_t* flags, size_t rows,
+                       size_t len, size_t ts_count, float* ts,
+                       hipStream_t stream);
+
+// ts[i] -= sum/n (baseline subtract; sum is a device scalar from sum_sumsq).
+hipError_t subtract_mean(float* ts, size_t n, const double* sum,
+                         hipStream_t stream);
+
+// threshold = snr * sqrt(sumsq/n) (device scalar sumsq of the zero-mean
+// series); appends count of ts[i] > threshold into out_count (atomic; zero it
+// first) and writes the threshold to out_threshold (reference
+// signal_detect.hpp:33-67).
+hipError_t count_above(const float* ts, size_t n, const double* sumsq,
+                       float snr, unsigned* out_count, float* out_threshold,
+                       hipStream_t stream);
+
+// Inclusive prefix sum (float in → float out); scratch: >= 2048 floats.
+hipError_t inclusive_scan(const float* in, float* out, size_t n,
+                          float* scratch, hipStream_t stream);
+
+// box[i] = cumsum[i+L] - cumsum[i], i < n_out (reference
+// signal_detect_pipe.hpp:387-423).
+hipError_t boxcar(const float* cumsum, float* out, size_t n_out, size_t L,
+                  hipStream_t stream);
+
+// ---------------- display / spectrum simplification ----------------
+// Area-averaged power resample of [rows][len] complex waterfall to [H][W]
+// floats; one workgroup (64 lanes) per output pixel with LDS tree reduce
+// (reference resample_spectrum_3, simplify_spectrum.hpp:423-620; wg=64 was
+// measured fastest on wave64 GCN — kept for CDNA4).
+hipError_t resample_power_2d(const float2* wf, size_t rows, size_t len,
+                             float* out, int out_h, int out_w,
+                             hipStream_t stream);
+
+// sum over img (H*W floats) into partials/out via sum_sumsq; then
+// img[i] *= 1/(2*mean) with mean = sum/n (reference simplify_spectrum.hpp:627-644).
+hipError_t normalize_by_mean(float* img, size_t n, const double* sum,
+                             hipStream_t stream);
+
+// intensity [0,1] → ARGB32 lerp color_0..color_1, else overflow color
+// (reference simplify_spectrum.hpp:700-731, config.hpp:60-68).
+hipError_t generate_pixmap(const float* intensity, uint32_t* out, size_t n,
+                           uint32_t color0, uint32_t color1,
+                           uint32_t color_overflow, hipStream_t stream);
+
+// ---------------- misc device algorithms ----------------
+// Running-mean 1-bit threshold, time-major [nsamp][nchan]
+// (reference algorithm/running_mean.hpp:31-77).
+hipError_t running_mean_init(const float* data, size_t nsamp, size_t nchan,
+                             size_t windowsize, float* ave, hipStream_t stream);
+hipError_t running_mean(const float* data, size_t nsamp, size_t nchan,
+                        uint8_t* out, size_t windowsize, float* ave,
+                        hipStream_t stream);
+
+// correlator pointwise: corr[i] = scale * f1[i]*conj(f2[i]); mag[i] = |corr[i]|
+// (reference src/correlator.cpp:116-140; mag may be null).
+hipError_t correlate_pointwise(const float2* f1, const float2* f2,
+                               float2* corr, float* mag, size_t n, float scale,
+                               hipStream_t stream);
+
+}  // namespace srtb_hip

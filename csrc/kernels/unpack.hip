@@ -1,0 +1,334 @@
+// Unpack kernels: packed baseband bytes → float32 samples, FFT window fused.
+//
+// Capability parity with reference userspace/include/srtb/unpack.hpp:43-403
+// (1/2/4-bit MSB-first unsigned fields; 8/16/32-bit signed/unsigned casts;
+// 2-pol per-sample interleave; SNAP-1 "1 1 2 2"; GZNU A1 4-byte-word
+// deinterleave with offset-binary fix), redesigned for CDNA4:
+//  - each lane consumes one aligned uint32 (4 bytes) and emits 4..32 floats
+//    as float4 stores → 16 B/lane global writes, fully coalesced
+//  - wave64-sized blocks, grid-stride loops capped per G11.
+
+#include "common.h"
+#include "../include/srtb_kernels.h"
+
+namespace srtb_hip {
+
+namespace {
+
+template <bool kWindow>
+__device__ inline float wmul(const float* __restrict__ w, size_t i, float v) {
+  if constexpr (kWindow) return v * w[i];
+  return v;
+}
+
+// ---- sub-byte unpack: one uint32 (4 bytes) per lane ----
+// bits per sample B in {1,2,4}: byte b yields 8/B samples, MSB-first.
+template <int NBITS, bool kWindow>
+__global__ void k_unpack_subbyte(const uint32_t* __restrict__ in,
+                                 float* __restrict__ out, size_t n_words,
+                                 const float* __restrict__ window) {
+  constexpr int per_byte = 8 / NBITS;
+  constexpr int per_word = 4 * per_byte;
+  constexpr uint32_t mask = (1u << NBITS) - 1u;
+  const size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (size_t w = (size_t)blockIdx.x * blockDim.x + threadIdx.x; w < n_words;
+       w += stride) {
+    const uint32_t v = in[w];
+    const size_t base = w * per_word;
+    float vals[per_word];
+#pragma unroll
+    for (int byte = 0; byte < 4; ++byte) {
+      const uint32_t bv = (v >> (8 * byte)) & 0xffu;  // little-endian byte order
+#pragma unroll
+      for (int i = 0; i < per_byte; ++i) {
+        const uint32_t field = (bv >> ((per_byte - 1 - i) * NBITS)) & mask;
+        vals[byte * per_byte + i] =
+            wmul<kWindow>(window, base + byte * per_byte + i, (float)field);
+      }
+    }
+    float4* o4 = reinterpret_cast<float4*>(out + base);
+#pragma unroll
+    for (int q = 0; q < per_word / 4; ++q)
+      o4[q] = make_float4(vals[4 * q], vals[4 * q + 1], vals[4 * q + 2],
+                          vals[4 * q + 3]);
+  }
+}
+
+// ---- byte casts: one uint32 = 4 samples per lane ----
+template <bool kSigned, bool kWindow>
+__global__ void k_unpack_cast8(const uint32_t* __restrict__ in,
+                               float* __restrict__ out, size_t n_words,
+                               const float* __restrict__ window) {
+  const size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (size_t w = (size_t)blockIdx.x * blockDim.x + threadIdx.x; w < n_words;
+       w += stride) {
+    const uint32_t v = in[w];
+    const size_t base = w * 4;
+    float4 o;
+    float* op = &o.x;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const uint32_t b = (v >> (8 * i)) & 0xffu;
+      const float f = kSigned ? (float)(int8_t)b : (float)b;
+      op[i] = wmul<kWindow>(window, base + i, f);
+    }
+    reinterpret_cast<float4*>(out + base)[0] = o;
+  }
+}
+
+template <typename T, bool kWindow>
+__global__ void k_unpack_cast(const T* __restrict__ in, float* __restrict__ out,
+                              size_t n, const float* __restrict__ window) {
+  const size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride)
+    out[i] = wmul<kWindow>(window, i, (float)in[i]);
+}
+
+// ---- 2-pol per-sample interleave: [p0 p1 p0 p1 ...] int8 ----
+// one lane consumes 8 bytes (uint2) = 4 samples per pol, float4 stores.
+template <bool kWindow>
+__global__ void k_unpack_2pol(const uint2* __restrict__ in,
+                              float* __restrict__ out0,
+                              float* __restrict__ out1, size_t n_groups,
+                              const float* __restrict__ window) {
+  const size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (size_t g = (size_t)blockIdx.x * blockDim.x + threadIdx.x; g < n_groups;
+       g += stride) {
+    const uint2 v = in[g];
+    const size_t base = g * 4;
+    float4 a, b;
+    float* ap = &a.x;
+    float* bp = &b.x;
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      const uint32_t word = (i == 0) ? v.x : v.y;
+      ap[2 * i + 0] = (float)(int8_t)(word & 0xff);
+      bp[2 * i + 0] = (float)(int8_t)((word >> 8) & 0xff);
+      ap[2 * i + 1] = (float)(int8_t)((word >> 16) & 0xff);
+      bp[2 * i + 1] = (float)(int8_t)((word >> 24) & 0xff);
+    }
+    if constexpr (kWindow) {
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        ap[i] *= window[base + i];
+        bp[i] *= window[base + i];
+      }
+    }
+    reinterpret_cast<float4*>(out0 + base)[0] = a;
+    reinterpret_cast<float4*>(out1 + base)[0] = b;
+  }
+}
+
+// ---- SNAP-1 "1 1 2 2": uint32 = [s0p0 s1p0 s0p1 s1p1] ----
+// one lane consumes 8 bytes = 4 samples per pol.
+template <bool kWindow>
+__global__ void k_unpack_snap1(const uint2* __restrict__ in,
+                               float* __restrict__ out0,
+                               float* __restrict__ out1, size_t n_groups,
+                               const float* __restrict__ window) {
+  const size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (size_t g = (size_t)blockIdx.x * blockDim.x + threadIdx.x; g < n_groups;
+       g += stride) {
+    const uint2 v = in[g];
+    const size_t base = g * 4;
+    float4 a, b;
+    a.x = (float)(int8_t)(v.x & 0xff);
+    a.y = (float)(int8_t)((v.x >> 8) & 0xff);
+    b.x = (float)(int8_t)((v.x >> 16) & 0xff);
+    b.y = (float)(int8_t)((v.x >> 24) & 0xff);
+    a.z = (float)(int8_t)(v.y & 0xff);
+    a.w = (float)(int8_t)((v.y >> 8) & 0xff);
+    b.z = (float)(int8_t)((v.y >> 16) & 0xff);
+    b.w = (float)(int8_t)((v.y >> 24) & 0xff);
+    if constexpr (kWindow) {
+      float* ap = &a.x;
+      float* bp = &b.x;
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        ap[i] *= window[base + i];
+        bp[i] *= window[base + i];
+      }
+    }
+    reinterpret_cast<float4*>(out0 + base)[0] = a;
+    reinterpret_cast<float4*>(out1 + base)[0] = b;
+  }
+}
+
+// ---- GZNU A1: 4-byte words cycling over NS streams ----
+// one lane consumes NS consecutive words (uint32 each) = 4 samples per stream.
+template <int NS, bool kXor80, bool kWindow>
+__global__ void k_unpack_gznupsr(const uint32_t* __restrict__ in,
+                                 float* __restrict__ o0, float* __restrict__ o1,
+                                 float* __restrict__ o2, float* __restrict__ o3,
+                                 size_t n_groups,
+                                 const float* __restrict__ window) {
+  float* outs[4] = {o0, o1, o2, o3};
+  const size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (size_t g = (size_t)blockIdx.x * blockDim.x + threadIdx.x; g < n_groups;
+       g += stride) {
+    const size_t base = g * 4;
+#pragma unroll
+    for (int s = 0; s < NS; ++s) {
+      uint32_t w = in[g * NS + s];
+      if constexpr (kXor80) w ^= 0x80808080u;
+      float4 o;
+      o.x = (float)(int8_t)(w & 0xff);
+      o.y = (float)(int8_t)((w >> 8) & 0xff);
+      o.z = (float)(int8_t)((w >> 16) & 0xff);
+      o.w = (float)(int8_t)((w >> 24) & 0xff);
+      if constexpr (kWindow) {
+        o.x *= window[base];
+        o.y *= window[base + 1];
+        o.z *= window[base + 2];
+        o.w *= window[base + 3];
+      }
+      if (outs[s]) reinterpret_cast<float4*>(outs[s] + base)[0] = o;
+    }
+  }
+}
+
+}  // namespace
+
+hipError_t unpack(const uint8_t* in, float* out, size_t out_count, int nbits,
+                  const float* window, hipStream_t stream) {
+  const bool w = window != nullptr;
+  switch (nbits) {
+    case 1:
+    case 2:
+    case 4: {
+      const size_t n_words = out_count * nbits / 32;
+      const dim3 g = grid_for(n_words);
+      auto* in32 = reinterpret_cast<const uint32_t*>(in);
+#define CASE(B)                                                              \
+  if (nbits == B) {                                                          \
+    if (w)                                                                   \
+      hipLaunchKernelGGL((k_unpack_subbyte<B, true>), g, dim3(kBlock), 0,    \
+                         stream, in32, out, n_words, window);                \
+    else                                                                     \
+      hipLaunchKernelGGL((k_unpack_subbyte<B, false>), g, dim3(kBlock), 0,   \
+                         stream, in32, out, n_words, window);                \
+  }
+      CASE(1) CASE(2) CASE(4)
+#undef CASE
+      break;
+    }
+    case 8:
+    case -8: {
+      const size_t n_words = out_count / 4;
+      const dim3 g = grid_for(n_words);
+      auto* in32 = reinterpret_cast<const uint32_t*>(in);
+      if (nbits < 0) {
+        if (w)
+          hipLaunchKernelGGL((k_unpack_cast8<true, true>), g, dim3(kBlock), 0,
+                             stream, in32, out, n_words, window);
+        else
+          hipLaunchKernelGGL((k_unpack_cast8<true, false>), g, dim3(kBlock), 0,
+                             stream, in32, out, n_words, window);
+      } else {
+        if (w)
+          hipLaunchKernelGGL((k_unpack_cast8<false, true>), g, dim3(kBlock), 0,
+                             stream, in32, out, n_words, window);
+        else
+          hipLaunchKernelGGL((k_unpack_cast8<false, false>), g, dim3(kBlock),
+                             0, stream, in32, out, n_words, window);
+      }
+      break;
+    }
+    case 16:
+      hipLaunchKernelGGL((k_unpack_cast<uint16_t, false>),
+                         grid_for(out_count), dim3(kBlock), 0, stream,
+                         reinterpret_cast<const uint16_t*>(in), out, out_count,
+                         window);
+      break;
+    case -16:
+      hipLaunchKernelGGL((k_unpack_cast<int16_t, false>), grid_for(out_count),
+                         dim3(kBlock), 0, stream,
+                         reinterpret_cast<const int16_t*>(in), out, out_count,
+                         window);
+      break;
+    case 32:
+      hipLaunchKernelGGL((k_unpack_cast<uint32_t, false>),
+                         grid_for(out_count), dim3(kBlock), 0, stream,
+                         reinterpret_cast<const uint32_t*>(in), out, out_count,
+                         window);
+      break;
+    case -32:
+      hipLaunchKernelGGL((k_unpack_cast<int32_t, false>), grid_for(out_count),
+                         dim3(kBlock), 0, stream,
+                         reinterpret_cast<const int32_t*>(in), out, out_count,
+                         window);
+      break;
+    default:
+      return hipErrorInvalidValue;
+  }
+  SRTB_CHECK_LAUNCH();
+  return hipSuccess;
+}
+
+hipError_t unpack_interleaved_2pol(const int8_t* in, float* out0, float* out1,
+                                   size_t count_per_pol, const float* window,
+                                   hipStream_t stream) {
+  const size_t n_groups = count_per_pol / 4;
+  const dim3 g = grid_for(n_groups);
+  auto* in2 = reinterpret_cast<const uint2*>(in);
+  if (window)
+    hipLaunchKernelGGL((k_unpack_2pol<true>), g, dim3(kBlock), 0, stream, in2,
+                       out0, out1, n_groups, window);
+  else
+    hipLaunchKernelGGL((k_unpack_2pol<false>), g, dim3(kBlock), 0, stream, in2,
+                       out0, out1, n_groups, window);
+  SRTB_CHECK_LAUNCH();
+  return hipSuccess;
+}
+
+hipError_t unpack_naocpsr_snap1(const int8_t* in, float* out0, float* out1,
+                                size_t count_per_pol, const float* window,
+                                hipStream_t stream) {
+  const size_t n_groups = count_per_pol / 4;
+  const dim3 g = grid_for(n_groups);
+  auto* in2 = reinterpret_cast<const uint2*>(in);
+  if (window)
+    hipLaunchKernelGGL((k_unpack_snap1<true>), g, dim3(kBlock), 0, stream, in2,
+                       out0, out1, n_groups, window);
+  else
+    hipLaunchKernelGGL((k_unpack_snap1<false>), g, dim3(kBlock), 0, stream,
+                       in2, out0, out1, n_groups, window);
+  SRTB_CHECK_LAUNCH();
+  return hipSuccess;
+}
+
+hipError_t unpack_gznupsr_a1(const uint8_t* in, float* out0, float* out1,
+                             float* out2, float* out3, int n_streams,
+                             size_t count_per_stream, const float* window,
+                             hipStream_t stream) {
+  const size_t n_groups = count_per_stream / 4;
+  const dim3 g = grid_for(n_groups);
+  auto* in32 = reinterpret_cast<const uint32_t*>(in);
+  const bool w = window != nullptr;
+  if (n_streams == 2) {
+    if (w)
+      hipLaunchKernelGGL((k_unpack_gznupsr<2, false, true>), g, dim3(kBlock),
+                         0, stream, in32, out0, out1, out2, out3, n_groups,
+                         window);
+    else
+      hipLaunchKernelGGL((k_unpack_gznupsr<2, false, false>), g, dim3(kBlock),
+                         0, stream, in32, out0, out1, out2, out3, n_groups,
+                         window);
+  } else if (n_streams == 4) {
+    if (w)
+      hipLaunchKernelGGL((k_unpack_gznupsr<4, true, true>), g, dim3(kBlock), 0,
+                         stream, in32, out0, out1, out2, out3, n_groups,
+                         window);
+    else
+      hipLaunchKernelGGL((k_unpack_gznupsr<4, true, false>), g, dim3(kBlock),
+                         0, stream, in32, out0, out1, out2, out3, n_groups,
+                         window);
+  } else {
+    return hipErrorInvalidValue;
+  }
+  SRTB_CHECK_LAUNCH();
+  return hipSuccess;
+}
+
+}  // namespace srtb_hip
