@@ -225,8 +225,12 @@ torch::Tensor t_time_series(torch::Tensor wf,
   const size_t rows = wf.size(0), len = wf.size(1);
   auto out = torch::empty({ts_count}, wf.options().dtype(torch::kFloat32));
   const uint8_t* f = flags.has_value() ? flags->data_ptr<uint8_t>() : nullptr;
-  check(time_series(cptr(wf), f, rows, len, ts_count, out.data_ptr<float>(),
-                    cur_stream()),
+  auto scratch = torch::empty(
+      {(int64_t)time_series_chunks(ts_count) * ts_count},
+      wf.options().dtype(torch::kFloat32));
+  check(time_series_2stage(cptr(wf), f, rows, len, ts_count,
+                           out.data_ptr<float>(), scratch.data_ptr<float>(),
+                           cur_stream()),
         "time_series");
   return out;
 }

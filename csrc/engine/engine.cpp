@@ -58,6 +58,10 @@ PipelineEngine::PipelineEngine(const EngineConfig& cfg, int n_slots)
     check_hip(hipMalloc(&s.s2s4, s_ * sizeof(float2)), "s2s4 alloc");
     check_hip(hipMalloc(&s.flags, s_), "flags alloc");
     check_hip(hipMalloc(&s.ts, ts_count_ * sizeof(float)), "ts alloc");
+    check_hip(hipMalloc(&s.ts_partial,
+                        (size_t)time_series_chunks(ts_count_) * ts_count_ *
+                            sizeof(float)),
+              "ts partial alloc");
     check_hip(hipMalloc(&s.cumsum, ts_count_ * sizeof(float)), "cumsum alloc");
     check_hip(hipMalloc(&s.box, ts_count_ * sizeof(float)), "box alloc");
     check_hip(hipMalloc(&s.scan_scratch, 4096 * sizeof(float)), "scan alloc");
@@ -97,6 +101,7 @@ PipelineEngine::~PipelineEngine() {
     hipFree(s.s2s4);
     hipFree(s.flags);
     hipFree(s.ts);
+    hipFree(s.ts_partial);
     hipFree(s.cumsum);
     hipFree(s.box);
     hipFree(s.scan_scratch);
@@ -152,7 +157,9 @@ void PipelineEngine::enqueue_chain(Slot& s, const void* dev_raw) {
     ts_flags = s.flags;
   }
   // 7. time series over non-zapped rows
-  check_hip(time_series(wf, ts_flags, s_, l_, ts_count_, s.ts, st), "ts");
+  check_hip(time_series_2stage(wf, ts_flags, s_, l_, ts_count_, s.ts,
+                               s.ts_partial, st),
+            "ts");
   // 8. baseline subtract
   check_hip(sum_sumsq(s.ts, ts_count_, s.partials, s.sums, st), "ts sum");
   check_hip(subtract_mean(s.ts, ts_count_, s.sums, st), "ts sub");

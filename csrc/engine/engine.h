@@ -124,6 +124,7 @@ class PipelineEngine {
     float2* s2s4 = nullptr;         // [S]
     uint8_t* flags = nullptr;       // [S]
     float* ts = nullptr;            // [ts_count]
+    float* ts_partial = nullptr;    // [chunks][ts_count] two-stage scratch
     float* cumsum = nullptr;        // [ts_count]
     float* box = nullptr;           // [ts_count]
     float* scan_scratch = nullptr;  // [4096]

