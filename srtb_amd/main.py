@@ -1,0 +1,257 @@
+"""srtb_amd main application — the reference `simple-radio-telescope-backend`
+executable equivalent (reference src/main.cpp:61-333).
+
+Reads a srtb_config.cfg-compatible config (cmd > cfg-file > defaults), sets up
+the per-stream pipelines (native GPU engine, or the NumPy path on CPU-only
+hosts), consumes baseband from a recorded file (with dedispersion-overlap
+seek-back) or UDP, writes detection products (.bin/.npy/.tim) with the
+polarization-coincidence scheduler, and optionally dumps live waterfall
+frames (headless PPM — the Qt GUI equivalent).
+
+Multi-GPU: launch under torchrun, one rank per GPU; streams are sharded
+round-robin (parallel.sharding) and detection stats all-reduced over
+RCCL/xGMI.
+
+Usage:
+  python -m srtb_amd.main --config_file_name srtb_config.cfg [--key value ...]
+  torchrun --nproc-per-node 8 -m srtb_amd.main --config_file_name ...
+"""
+
+from __future__ import annotations
+
+import sys
+import time
+
+import numpy as np
+
+from . import ref
+from .config import Config, parse_args
+from .io import backends as bk
+from .io.file_input import FileBlockReader
+from .io.writers import BlockProducts, SignalWriteScheduler
+from .pipeline.cpu import CpuPipeline
+from .parallel.sharding import (DetectionAggregator, broadcast_config,
+                                init_distributed, shard_streams)
+
+# extra, non-reference options handled by the runner itself
+RUNNER_FLAGS = {"--max-blocks", "--device", "--waterfall-ppm"}
+
+
+def write_ppm(path: str, argb: np.ndarray) -> None:
+    """Dump an ARGB32 [H][W] image as binary PPM (P6)."""
+    h, w = argb.shape
+    rgb = np.empty((h, w, 3), dtype=np.uint8)
+    rgb[..., 0] = (argb >> 16) & 0xFF
+    rgb[..., 1] = (argb >> 8) & 0xFF
+    rgb[..., 2] = argb & 0xFF
+    with open(path, "wb") as f:
+        f.write(b"P6\n%d %d\n255\n" % (w, h))
+        f.write(rgb.tobytes())
+
+
+class GpuStreamPipeline:
+    """One data stream on one GPU via the native engine."""
+
+    def __init__(self, cfg: Config, nsamps_reserved: int):
+        import torch
+        from .ops import native
+        self.torch = torch
+        self.C = native()
+        self.cfg = cfg
+        ranges = ref.parse_rfi_freq_list(cfg.mitigate_rfi_freq_list)
+        nc = cfg.baseband_input_count // 2
+        bins = ref.rfi_ranges_to_bins(cfg.baseband_freq_low,
+                                      cfg.baseband_bandwidth, nc, ranges)
+        self.eng = self.C.PipelineEngine(
+            n=cfg.baseband_input_count, nbits=cfg.baseband_input_bits,
+            channels=cfg.spectrum_channel_count,
+            freq_low=cfg.baseband_freq_low, bandwidth=cfg.baseband_bandwidth,
+            sample_rate=cfg.baseband_sample_rate, dm=cfg.dm,
+            rfi_threshold=cfg.mitigate_rfi_average_method_threshold,
+            sk_threshold=cfg.mitigate_rfi_spectral_kurtosis_threshold,
+            snr_threshold=cfg.signal_detect_signal_noise_threshold,
+            max_boxcar=cfg.signal_detect_max_boxcar_length,
+            nsamps_reserved=nsamps_reserved,
+            zap_ranges=[[int(a), int(b)] for a, b in bins],
+            use_phase_table=False, enable_rfi_s1=True, enable_sk=True,
+            n_slots=2)
+
+    def process_block(self, raw: np.ndarray, counter: int) -> BlockProducts:
+        torch = self.torch
+        cfg = self.cfg
+        t = torch.from_numpy(np.ascontiguousarray(raw))
+        slot = self.eng.submit(t)
+        res = self.eng.wait(slot)
+        products = BlockProducts(counter=counter, timestamp=counter)
+        gate = res["zero_count"] < (cfg.signal_detect_channel_threshold *
+                                    cfg.spectrum_channel_count)
+        detected = [(L, c) for L, c in res["counts"] if c > 0]
+        self.last_result = res
+        self.last_slot = slot
+        if gate and detected:
+            products.raw = raw
+            products.waterfall = self.eng.waterfall(slot).cpu().numpy()
+            ts = self.eng.time_series(slot).cpu().numpy()
+            for L, _count in detected:
+                if L == 1:
+                    products.time_series.append((1, ts.copy()))
+                else:
+                    products.time_series.append(
+                        (L, self.eng.boxcar_series(slot, L).cpu().numpy()))
+        return products
+
+    def waterfall_frame(self, width: int, height: int) -> np.ndarray:
+        wf = self.eng.waterfall(self.last_slot)
+        img = self.C.resample_power(wf, height, width)
+        self.C.normalize_by_mean(img)
+        pix = self.C.generate_pixmap(img, ref.COLOR_0, ref.COLOR_1,
+                                     ref.COLOR_OVERFLOW)
+        return pix.cpu().numpy().view(np.uint32)
+
+
+class CpuStreamPipeline:
+    """CPU/NumPy path (plumbing runs, no GPU)."""
+
+    def __init__(self, cfg: Config, nsamps_reserved: int):
+        self.cfg = cfg
+        self.pipe = CpuPipeline(cfg)
+
+    def process_block(self, raw: np.ndarray, counter: int) -> BlockProducts:
+        cfg = self.cfg
+        res = self.pipe.process_block(raw)
+        self.last_result = res
+        products = BlockProducts(counter=counter, timestamp=counter)
+        gate = res["zero_count"] < (cfg.signal_detect_channel_threshold *
+                                    cfg.spectrum_channel_count)
+        if gate and res["detections"]:
+            products.raw = raw
+            products.waterfall = res["waterfall"]
+            for L, _cnt, series in res["detections"]:
+                products.time_series.append((L, series))
+        return products
+
+    def waterfall_frame(self, width: int, height: int) -> np.ndarray:
+        wf = self.last_result["waterfall"]
+        p = np.abs(wf.astype(np.complex64)) ** 2
+        img = ref.resample_power_2d(p, height, width)
+        img = ref.normalize_by_mean(img)
+        return ref.generate_pixmap(img)
+
+
+def main(argv=None) -> int:
+    argv = list(sys.argv[1:] if argv is None else argv)
+    # split runner-only flags from reference-style config options
+    max_blocks = None
+    device = None
+    waterfall_every = 0
+    cfg_argv = []
+    i = 0
+    while i < len(argv):
+        a = argv[i]
+        if a.startswith("--max-blocks"):
+            max_blocks = int(a.split("=", 1)[1] if "=" in a else argv[i + 1])
+            i += 1 if "=" in a else 2
+        elif a.startswith("--device"):
+            device = a.split("=", 1)[1] if "=" in a else argv[i + 1]
+            i += 1 if "=" in a else 2
+        elif a.startswith("--waterfall-ppm"):
+            waterfall_every = int(a.split("=", 1)[1] if "=" in a else argv[i + 1])
+            i += 1 if "=" in a else 2
+        else:
+            cfg_argv.append(a)
+            i += 1
+    rank, world, local_rank = init_distributed()
+    cfg = parse_args(cfg_argv) if rank == 0 else Config()
+    cfg = broadcast_config(cfg if rank == 0 else None)
+
+    import importlib
+    torch_spec = importlib.util.find_spec("torch")
+    use_gpu = False
+    if device != "cpu" and torch_spec is not None:
+        import torch
+        use_gpu = torch.cuda.is_available()
+
+    reserved = ref.nsamps_reserved(
+        cfg.baseband_input_count, cfg.spectrum_channel_count,
+        cfg.baseband_freq_low, cfg.baseband_bandwidth,
+        cfg.baseband_sample_rate, cfg.dm, cfg.baseband_reserve_sample)
+
+    make = GpuStreamPipeline if use_gpu else CpuStreamPipeline
+    agg = DetectionAggregator()
+    writer = SignalWriteScheduler(
+        cfg.baseband_output_file_prefix, cfg.baseband_input_count,
+        cfg.baseband_sample_rate, real_time=(cfg.input_file_path == ""))
+
+    t0 = time.time()
+    n_blocks = 0
+    if cfg.input_file_path:
+        # file replay: single stream (the reference also replays one stream)
+        pipe = make(cfg, reserved)
+        reader = FileBlockReader(cfg.input_file_path,
+                                 cfg.baseband_input_count,
+                                 cfg.baseband_input_bits, reserved,
+                                 cfg.input_file_offset_bytes)
+        for sample_index, raw in reader:
+            products = pipe.process_block(raw, sample_index)
+            writer.push(products)
+            res = getattr(pipe, "last_result", None)
+            if isinstance(res, dict) and "zero_count" in res:
+                zc = res["zero_count"]
+                counts = (res.get("counts")
+                          or [(L, c) for L, c, _ in res.get("detections", [])])
+            else:
+                zc, counts = 0, []
+            agg.update(int(zc), [(int(a), int(b)) for a, b in counts])
+            n_blocks += 1
+            if waterfall_every and n_blocks % waterfall_every == 0:
+                frame = pipe.waterfall_frame(cfg.gui_pixmap_width,
+                                             cfg.gui_pixmap_height)
+                write_ppm(f"{cfg.baseband_output_file_prefix}"
+                          f"waterfall_r{rank}_{n_blocks}.ppm", frame)
+            if max_blocks is not None and n_blocks >= max_blocks:
+                break
+    else:
+        # UDP ingest: shard endpoints across ranks
+        from .io.udp import BlockAssembler, UdpPacketProvider, run_receiver
+        backend = bk.get_backend(cfg.baseband_format_type)
+        n_endpoints = len(cfg.udp_receiver_address)
+        my = shard_streams(n_endpoints, world, rank)
+        if not my:
+            print(f"rank {rank}: no UDP endpoints assigned")
+            return 0
+        ep = my[0]  # one endpoint per rank in this runner
+        pipe = make(cfg, reserved)
+        bits = abs(cfg.baseband_input_bits)
+        block_bytes = cfg.baseband_input_count * bits // 8 * \
+            bk.get_data_stream_count(cfg.baseband_format_type)
+        assembler = BlockAssembler(backend, block_bytes)
+        provider = UdpPacketProvider(cfg.udp_receiver_address[ep],
+                                     cfg.udp_receiver_port[ep])
+        state = {"n": 0}
+
+        def on_block(blk, ts):
+            products = pipe.process_block(blk, ts)
+            writer.push(products)
+            state["n"] += 1
+
+        def stop():
+            return max_blocks is not None and state["n"] >= max_blocks
+
+        run_receiver(provider, assembler, on_block, stop)
+        n_blocks = state["n"]
+
+    elapsed = time.time() - t0
+    stats = agg.reduce()
+    if rank == 0:
+        sps = stats.blocks * cfg.baseband_input_count / max(elapsed, 1e-9)
+        print(f"[srtb_amd] blocks={stats.blocks} detections={stats.detections} "
+              f"signal_counts={stats.signal_counts} "
+              f"zapped_channels={stats.zapped_channels} "
+              f"written={len(writer.written)} "
+              f"throughput={sps / 1e6:.1f} Msamples/s "
+              f"real_time_ratio={sps / cfg.baseband_sample_rate:.2f}")
+    return 0
+
+
+if __name__ == "__main__":
+    raise SystemExit(main())
