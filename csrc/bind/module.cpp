@@ -374,13 +374,21 @@ class PyEngine {
     eng_ = std::make_unique<PipelineEngine>(c, (int)n_slots);
   }
 
-  int64_t submit(torch::Tensor raw) {
+  int64_t submit(torch::Tensor raw, double dm_override) {
     TORCH_CHECK(raw.is_contiguous());
     TORCH_CHECK((size_t)raw.numel() * raw.element_size() == eng_->raw_bytes(),
                 "raw block has wrong byte size");
-    if (raw.is_cuda()) return eng_->submit_device(raw.data_ptr(),
-                                                  eng_->raw_bytes());
-    return eng_->submit(raw.data_ptr(), eng_->raw_bytes());
+    if (raw.is_cuda())
+      return eng_->submit_device(raw.data_ptr(), eng_->raw_bytes(),
+                                 dm_override);
+    return eng_->submit(raw.data_ptr(), eng_->raw_bytes(), dm_override);
+  }
+
+  int64_t submit_samples(torch::Tensor samples, double dm_override) {
+    TORCH_CHECK(samples.is_cuda() && samples.is_contiguous());
+    TORCH_CHECK(samples.scalar_type() == torch::kFloat32);
+    return eng_->submit_samples_device(samples.data_ptr<float>(),
+                                       samples.numel(), dm_override);
   }
 
   py::dict wait(int64_t slot) {
@@ -481,7 +489,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
            py::arg("use_phase_table") = false,
            py::arg("enable_rfi_s1") = true, py::arg("enable_sk") = true,
            py::arg("n_slots") = 2)
-      .def("submit", &PyEngine::submit)
+      .def("submit", &PyEngine::submit, py::arg("raw"),
+           py::arg("dm_override") = std::nan(""))
+      .def("submit_samples", &PyEngine::submit_samples, py::arg("samples"),
+           py::arg("dm_override") = std::nan(""))
       .def("wait", &PyEngine::wait)
       .def("synchronize", &PyEngine::synchronize)
       .def("waterfall", &PyEngine::waterfall)

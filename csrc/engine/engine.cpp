@@ -118,16 +118,26 @@ PipelineEngine::~PipelineEngine() {
   if (phase_table_) hipFree(phase_table_);
 }
 
-void PipelineEngine::enqueue_chain(Slot& s, const void* dev_raw) {
+void PipelineEngine::enqueue_chain(Slot& s, const uint8_t* dev_raw,
+                                   const float* dev_samples, double dm) {
   hipStream_t st = s.stream;
-  const uint8_t* raw = static_cast<const uint8_t*>(dev_raw);
+  if (std::isnan(dm)) dm = cfg_.dm;
+  const float2* table = phase_table_;
+  if (table && dm != cfg_.dm)
+    throw std::runtime_error("dm override requires use_phase_table=false");
 
-  // 1. unpack (+ window fused; default rectangle → none)
-  check_hip(unpack(raw, s.samples, n_, cfg_.baseband_input_bits, nullptr, st),
-            "unpack");
+  const float* fft_in = s.samples;
+  if (dev_raw) {
+    // 1. unpack (+ window fused; default rectangle → none)
+    check_hip(unpack(dev_raw, s.samples, n_, cfg_.baseband_input_bits, nullptr,
+                     st),
+              "unpack");
+  } else {
+    fft_in = dev_samples;
+  }
   // 2. R2C forward (out-of-place; Nyquist bin written but ignored: the
   //    downstream count is Nc — reference drops it, fft_pipe.hpp:77)
-  s.plans.exec_r2c(s.samples, s.spec);
+  s.plans.exec_r2c(const_cast<float*>(fft_in), s.spec);
   // 3. mean |X|^2 over Nc
   if (cfg_.enable_rfi_s1)
     check_hip(mean_power(s.spec, nc_, s.partials, s.mean_power, st), "meanp");
@@ -135,8 +145,7 @@ void PipelineEngine::enqueue_chain(Slot& s, const void* dev_raw) {
   check_hip(rfi_dedisperse_fused(
                 s.spec, nc_, cfg_.enable_rfi_s1 ? s.mean_power : nullptr,
                 cfg_.rfi_threshold, norm_coeff_, cfg_.zap_ranges,
-                cfg_.n_zap_ranges, f_min_, f_c_, df_, cfg_.dm, phase_table_,
-                st),
+                cfg_.n_zap_ranges, f_min_, f_c_, df_, dm, table, st),
             "rfi+dedisp");
   // 5. waterfall: batched backward C2C in place → [S][L]
   s.plans.exec_c2c_backward(s.spec);
@@ -194,7 +203,8 @@ void PipelineEngine::enqueue_chain(Slot& s, const void* dev_raw) {
   s.busy = true;
 }
 
-int PipelineEngine::submit(const void* host_bytes, size_t nbytes) {
+int PipelineEngine::submit(const void* host_bytes, size_t nbytes,
+                           double dm_override) {
   if (nbytes != raw_bytes_) throw std::runtime_error("submit: wrong size");
   const int id = next_slot_;
   next_slot_ = (next_slot_ + 1) % n_slots_;
@@ -206,11 +216,12 @@ int PipelineEngine::submit(const void* host_bytes, size_t nbytes) {
   check_hip(hipMemcpyAsync(s.raw, host_bytes, nbytes, hipMemcpyHostToDevice,
                            s.stream),
             "raw h2d");
-  enqueue_chain(s, s.raw);
+  enqueue_chain(s, s.raw, nullptr, dm_override);
   return id;
 }
 
-int PipelineEngine::submit_device(const void* dev_bytes, size_t nbytes) {
+int PipelineEngine::submit_device(const void* dev_bytes, size_t nbytes,
+                                  double dm_override) {
   if (nbytes != raw_bytes_) throw std::runtime_error("submit: wrong size");
   const int id = next_slot_;
   next_slot_ = (next_slot_ + 1) % n_slots_;
@@ -219,7 +230,22 @@ int PipelineEngine::submit_device(const void* dev_bytes, size_t nbytes) {
     check_hip(hipEventSynchronize(s.done), "slot wait");
     s.busy = false;
   }
-  enqueue_chain(s, dev_bytes);
+  enqueue_chain(s, static_cast<const uint8_t*>(dev_bytes), nullptr,
+                dm_override);
+  return id;
+}
+
+int PipelineEngine::submit_samples_device(const float* dev_samples,
+                                          size_t count, double dm_override) {
+  if (count != n_) throw std::runtime_error("submit_samples: wrong count");
+  const int id = next_slot_;
+  next_slot_ = (next_slot_ + 1) % n_slots_;
+  Slot& s = *slots_[id];
+  if (s.busy) {
+    check_hip(hipEventSynchronize(s.done), "slot wait");
+    s.busy = false;
+  }
+  enqueue_chain(s, nullptr, dev_samples, dm_override);
   return id;
 }
 

@@ -66,11 +66,21 @@ class PipelineEngine {
   // Upload raw bytes (host, ideally pinned) and enqueue the whole chain on
   // the slot's stream.  Blocks only if the slot's previous block is still in
   // flight.  Returns the slot used.
-  int submit(const void* host_bytes, size_t nbytes);
+  // dm_override: NAN = use the configured DM; a finite value re-runs the
+  // (on-the-fly fp64) dedispersion at that DM — the DM-trial sweep path.
+  int submit(const void* host_bytes, size_t nbytes, double dm_override = NAN);
 
   // Enqueue the chain reading raw bytes already on the device (e.g. from a
   // torch tensor); caller guarantees lifetime until wait().
-  int submit_device(const void* dev_bytes, size_t nbytes);
+  int submit_device(const void* dev_bytes, size_t nbytes,
+                    double dm_override = NAN);
+
+  // Enqueue the chain starting at the R2C FFT from already-unpacked float
+  // samples on the device (count = baseband_input_count).  Used for
+  // multi-polarization formats whose unpack fans one packet stream out into
+  // several sample streams (reference unpack_pipe.hpp:146-390).
+  int submit_samples_device(const float* dev_samples, size_t count,
+                            double dm_override = NAN);
 
   // Wait for a slot's chain and return its detection counters.
   BlockResult wait(int slot);
@@ -101,7 +111,8 @@ class PipelineEngine {
 
  private:
   struct Slot;
-  void enqueue_chain(Slot& s, const void* dev_raw);
+  void enqueue_chain(Slot& s, const uint8_t* dev_raw,
+                     const float* dev_samples, double dm);
 
   EngineConfig cfg_;
   size_t n_, nc_, s_, l_, ts_count_, raw_bytes_;
