@@ -11,6 +11,7 @@
 #include <vector>
 
 #include "../engine/engine.h"
+#include "../fft/native_fft.h"
 #include "../include/srtb_kernels.h"
 
 namespace {
@@ -339,6 +340,50 @@ std::vector<torch::Tensor> t_correlate(torch::Tensor f1, torch::Tensor f2,
   return {corr, mag};
 }
 
+// ---------------- hand-written FFT (test/bench surface) ----------------
+
+torch::Tensor t_native_fft(torch::Tensor x, int64_t sign) {
+  CHECK_CUDA(x);
+  CHECK_CONTIG(x);
+  TORCH_CHECK(x.scalar_type() == torch::kComplexFloat);
+  const size_t len = x.size(-1);
+  const size_t batch = x.numel() / len;
+  auto stream = cur_stream();
+  NativeFft plan;
+  plan.plan(len, batch, (int)sign, stream);
+  auto out = torch::empty_like(x);
+  if (plan.n_passes() == 1) {
+    plan.exec(cptr(x), cptr(out), stream);
+  } else {
+    auto work = x.clone();  // passes 0..k-2 run in place on the input copy
+    plan.exec(cptr(work), cptr(out), stream);
+  }
+  check(hipStreamSynchronize(stream), "native_fft sync");  // plan is local
+  return out;
+}
+
+torch::Tensor t_native_rfft(torch::Tensor x) {
+  // real forward via packed-complex trick + r2c post; returns n/2 bins
+  CHECK_CUDA(x);
+  CHECK_CONTIG(x);
+  TORCH_CHECK(x.scalar_type() == torch::kFloat32);
+  const size_t n = x.numel();
+  const size_t m = n / 2;
+  auto stream = cur_stream();
+  NativeFft plan;
+  plan.plan(m, 1, -1, stream);
+  auto z = torch::empty({(int64_t)m},
+                        x.options().dtype(torch::kComplexFloat));
+  auto packed = x.view({(int64_t)m, 2});  // reinterpret as complex pairs
+  auto work = torch::empty_like(z);
+  work.copy_(torch::view_as_complex(packed));
+  plan.exec(cptr(work), cptr(z), stream);
+  check(r2c_post_process(cptr(z), cptr(z), m, nullptr, nullptr, stream),
+        "r2c post");
+  check(hipStreamSynchronize(stream), "native_rfft sync");
+  return z;
+}
+
 // ---------------- engine binding ----------------
 
 class PyEngine {
@@ -474,6 +519,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("generate_pixmap", &t_generate_pixmap);
   m.def("running_mean", &t_running_mean);
   m.def("correlate", &t_correlate);
+  m.def("native_fft", &t_native_fft, py::arg("x"), py::arg("sign"));
+  m.def("native_rfft", &t_native_rfft, py::arg("x"));
 
   py::class_<PyEngine>(m, "PipelineEngine")
       .def(py::init<int64_t, int64_t, int64_t, double, double, double, double,

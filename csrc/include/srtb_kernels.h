@@ -143,6 +143,40 @@ hipError_t inclusive_scan(const float* in, float* out, size_t n,
 hipError_t boxcar(const float* cumsum, float* out, size_t n_out, size_t L,
                   hipStream_t stream);
 
+// ---------------- hand-written Stockham FFT ----------------
+// One generic pass of the LDS-staged Stockham FFT (csrc/kernels/fft.hip).
+// Index-math oracle: srtb_amd/fftref.py.  Host-side planning:
+// csrc/fft/native_fft.h.
+struct FftPassDesc {
+  uint32_t n;            // pow2 FFT length of this pass (<= 4096)
+  uint32_t d0 = 1, d1 = 1;  // instance id -> digits q0, q1, q2
+  unsigned long long in_c0 = 0, in_c1 = 0, in_c2 = 0;
+  unsigned long long in_stride = 1;
+  unsigned long long out_c0 = 0, out_c1 = 0, out_c2 = 0;
+  unsigned long long out_stride = 1;
+  unsigned long long tw_f0 = 0, tw_f1 = 0;
+  unsigned long long tw_mod = 0;  // 0 = no inter-pass twiddle
+  int tw_lo_bits = 0;
+};
+
+// build table[j] = exp(sign * 2*pi*i * j / m), j in [0, count)
+hipError_t fft_build_twiddle(float2* table, size_t count, double m, int sign,
+                             hipStream_t stream);
+
+hipError_t fft_stockham_pass(const float2* in, float2* out,
+                             const FftPassDesc& d, size_t n_ffts, int F,
+                             bool load_ffast, bool store_ffast,
+                             const float2* tw_n, const float2* tw_hi,
+                             const float2* tw_lo, hipStream_t stream);
+
+// packed-real R2C finish: Z = C2C(x_even + i*x_odd) of length m -> true
+// spectrum X[0..m) (Nyquist dropped).  In-place safe (x may alias z).
+// If mean_partials != null (>= 1024 doubles), also writes mean|X|^2 to
+// out_mean (fused RFI-s1 statistic).
+hipError_t r2c_post_process(const float2* z, float2* x, size_t m,
+                            double* mean_partials, double* out_mean,
+                            hipStream_t stream);
+
 // ---------------- display / spectrum simplification ----------------
 // Area-averaged power resample of [rows][len] complex waterfall to [H][W]
 // floats; one workgroup (64 lanes) per output pixel with LDS tree reduce

@@ -1,0 +1,140 @@
+"""Reference implementation of the hand-written FFT path (index-math oracle).
+
+This file defines, in NumPy, EXACTLY the algorithm the HIP kernels in
+csrc/kernels/fft.hip implement:
+
+ 1. ``fft_small``  — power-of-two DIT FFT: bit-reversed gather at load time,
+    then in-place butterflies (this is what one workgroup does in LDS).
+ 2. ``fft_four_step`` — composite length L = L1 * L2 (L1, L2 small):
+    pass 1: for each column n2 of the [L1][L2] row-major view, an L1-point
+            FFT over stride-L2 elements, multiplied by the inter-pass
+            twiddle w(sign * n2 * k1 / L), stored back in place;
+    pass 2: for each row k1, an L2-point FFT over the contiguous row,
+            scattered to out[k1 + L1 * k2] — the "transpose" is folded into
+            pass 2's store addressing, so there is no separate transpose
+            kernel (rocFFT's plan for these shapes runs 2 transposes).
+ 3. ``fft_six_step_deep`` — length L1 * L2 * L3 via recursion (2^29 forward).
+ 4. ``r2c_post`` — real-input FFT via the packed-complex trick: N reals are
+    viewed as N/2 complex, C2C-transformed, then split into the true R2C
+    spectrum (reference fft/fft_1d_r2c_post_process.hpp:33-82 capability).
+
+Also serves as the "naive FFT" debug fallback of the reference
+(fft/naive_fft.hpp K8-K11 in SURVEY.md §2b).
+
+Conventions follow cuFFT/hipFFT: forward sign = -1, backward sign = +1, both
+unnormalized.
+"""
+
+from __future__ import annotations
+
+import numpy as np
+
+
+def bit_reverse_indices(n: int) -> np.ndarray:
+    t = n.bit_length() - 1
+    idx = np.arange(n)
+    rev = np.zeros(n, dtype=np.int64)
+    for b in range(t):
+        rev |= ((idx >> b) & 1) << (t - 1 - b)
+    return rev
+
+
+def twiddle_table(n: int, sign: int) -> np.ndarray:
+    """tw[j] = exp(sign * 2πi * j / n), j in [0, n/2) — all stages index into
+    this single table: stage with butterfly length ``len`` uses
+    tw[j * (n // len)] for j in [0, len/2)."""
+    j = np.arange(n // 2)
+    return np.exp(sign * 2j * np.pi * j / n).astype(np.complex64)
+
+
+def fft_small(x: np.ndarray, sign: int, dtype=np.complex64) -> np.ndarray:
+    """Power-of-two DIT FFT: bit-reversed load + in-place butterflies.
+
+    Mirrors the workgroup-level LDS algorithm of the HIP kernel.
+    """
+    n = x.size
+    assert n & (n - 1) == 0
+    tw = twiddle_table(n, sign).astype(np.complex128)
+    y = np.asarray(x, dtype=np.complex128)[bit_reverse_indices(n)].copy()
+    length = 2
+    while length <= n:
+        half = length // 2
+        tstep = n // length
+        # butterflies: for each group g, element j
+        for b in range(n // 2):
+            g, j = divmod(b, half)
+            i0 = g * length + j
+            i1 = i0 + half
+            w = tw[j * tstep]
+            a, c = y[i0], y[i1] * w
+            y[i0] = a + c
+            y[i1] = a - c
+        length *= 2
+    return y.astype(dtype)
+
+
+def fft_four_step(x: np.ndarray, l1: int, l2: int, sign: int,
+                  fft1=None, fft2=None) -> np.ndarray:
+    """Composite FFT of length l1*l2 per the docstring above."""
+    n = l1 * l2
+    assert x.size == n
+    fft1 = fft1 or (lambda v: fft_small(v, sign, np.complex128))
+    fft2 = fft2 or (lambda v: fft_small(v, sign, np.complex128))
+    m = np.asarray(x, dtype=np.complex128).reshape(l1, l2).copy()
+    # pass 1: columns (stride l2), then inter-pass twiddle
+    for n2 in range(l2):
+        col = fft1(m[:, n2])
+        k1 = np.arange(l1)
+        col = col * np.exp(sign * 2j * np.pi * (k1 * n2) / n)
+        m[:, n2] = col
+    # pass 2: rows (contiguous), scatter to out[k1 + l1*k2]
+    out = np.empty(n, dtype=np.complex128)
+    for k1 in range(l1):
+        row = fft2(m[k1, :])
+        out[k1 + l1 * np.arange(l2)] = row
+    return out.astype(np.complex64)
+
+
+def fft_deep(x: np.ndarray, factors: list[int], sign: int) -> np.ndarray:
+    """Arbitrary-depth composite: factors [f0, f1, ..., fk]; recursion
+    fft(len=f0 * rest) = four_step with l1=f0, l2=rest (pass-1 FFTs of f0,
+    pass-2 = recursive composite of the rest)."""
+    if len(factors) == 1:
+        return fft_small(x, sign, np.complex128).astype(np.complex64)
+    l1 = factors[0]
+    l2 = int(np.prod(factors[1:]))
+    return fft_four_step(
+        x, l1, l2, sign,
+        fft1=lambda v: fft_small(v, sign, np.complex128),
+        fft2=lambda v: fft_deep(v, factors[1:], sign).astype(np.complex128))
+
+
+def r2c_post(z: np.ndarray, sign: int = -1) -> np.ndarray:
+    """Packed-real R2C: given Z = C2C_fft(x_even + i*x_odd) of length M=N/2,
+    recover the true R2C spectrum X[0..M-1] (Nyquist dropped, matching the
+    pipeline's spectrum count Nc = N/2).
+
+    X[k] = (Z[k] + conj(Z[M-k]))/2 + w(k) * (Z[k] - conj(Z[M-k]))/(2i),
+    w(k) = exp(sign*2πi*k/N); Z[M] := Z[0].
+    """
+    m = z.size
+    zf = np.asarray(z, dtype=np.complex128)
+    zk = zf
+    zmk = np.conj(np.roll(zf[::-1], 1))  # conj(Z[M-k]), k=0..M-1
+    k = np.arange(m)
+    w = np.exp(sign * 2j * np.pi * k / (2 * m))
+    even = 0.5 * (zk + zmk)
+    odd = -0.5j * (zk - zmk)
+    return (even + w * odd).astype(np.complex64)
+
+
+def rfft_packed(x: np.ndarray, factors: list[int] | None = None) -> np.ndarray:
+    """Full real-input forward FFT via the packed trick; returns Nc bins."""
+    x = np.asarray(x, dtype=np.float64)
+    n = x.size
+    z = x[0::2] + 1j * x[1::2]
+    if factors is None:
+        zf = np.fft.fft(z)
+    else:
+        zf = fft_deep(z.astype(np.complex64), factors, -1).astype(np.complex128)
+    return r2c_post(zf, -1)

@@ -1,0 +1,83 @@
+"""GPU tests: hand-written Stockham FFT vs torch.fft (rocFFT oracle)."""
+
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def C():
+    from srtb_amd.ops import native
+    torch.cuda.set_device(0)
+    return native()
+
+
+def rand_c64(shape, seed):
+    rng = np.random.default_rng(seed)
+    return (rng.normal(size=shape) + 1j * rng.normal(size=shape)
+            ).astype(np.complex64)
+
+
+def rel_err(a, b):
+    return float(np.linalg.norm(a - b) / max(np.linalg.norm(b), 1e-30))
+
+
+# single-pass lengths
+@pytest.mark.parametrize("n", [4, 64, 256, 1024, 4096])
+@pytest.mark.parametrize("sign", [-1, 1])
+def test_fft_single_pass(C, n, sign):
+    x = rand_c64((8, n), seed=n + sign)
+    xt = torch.from_numpy(x).cuda()
+    out = C.native_fft(xt, sign).cpu().numpy()
+    ref = (np.fft.fft(x, axis=1) if sign == -1
+           else np.fft.ifft(x, axis=1) * n)
+    assert rel_err(out, ref) < 2e-6 * np.sqrt(n) + 1e-5
+
+
+# two-pass lengths (the waterfall backward shapes)
+@pytest.mark.parametrize("n", [8192, 1 << 13, 1 << 16, 1 << 18])
+@pytest.mark.parametrize("sign", [-1, 1])
+def test_fft_two_pass(C, n, sign):
+    batch = max(1, (1 << 20) // n)
+    x = rand_c64((batch, n), seed=n % 97 + sign)
+    xt = torch.from_numpy(x).cuda()
+    out = C.native_fft(xt, sign).cpu().numpy()
+    tt = torch.from_numpy(x).cuda()
+    ref = (torch.fft.fft(tt, dim=1) if sign == -1
+           else torch.fft.ifft(tt, dim=1) * n).cpu().numpy()
+    assert rel_err(out, ref) < 1e-4
+
+
+# three-pass lengths (forward 2^25..2^29 class; test smaller for speed)
+@pytest.mark.parametrize("n", [1 << 25, 1 << 27])
+def test_fft_three_pass(C, n):
+    x = rand_c64(n, seed=5)
+    xt = torch.from_numpy(x).cuda()
+    out = C.native_fft(xt, -1).cpu().numpy()
+    ref = torch.fft.fft(torch.from_numpy(x).cuda()).cpu().numpy()
+    assert rel_err(out, ref) < 2e-4
+
+
+@pytest.mark.parametrize("n", [1 << 12, 1 << 20, 1 << 24])
+def test_native_rfft(C, n):
+    rng = np.random.default_rng(3)
+    x = rng.normal(size=n).astype(np.float32)
+    xt = torch.from_numpy(x).cuda()
+    out = C.native_rfft(xt).cpu().numpy()
+    ref = torch.fft.rfft(xt).cpu().numpy()[:-1]
+    assert rel_err(out, ref) < 1e-4
+
+
+def test_native_rfft_j1644_scale(C):
+    """The flagship forward shape at reduced size: 2^26 reals."""
+    n = 1 << 26
+    rng = np.random.default_rng(4)
+    x = rng.normal(size=n).astype(np.float32)
+    xt = torch.from_numpy(x).cuda()
+    out = C.native_rfft(xt)
+    ref = torch.fft.rfft(xt)[:-1]
+    err = (out - ref).abs().max().item()
+    scale = ref.abs().max().item()
+    assert err < 2e-4 * scale
