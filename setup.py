@@ -27,6 +27,7 @@ HIP_SOURCES = [
     "csrc/kernels/unpack.hip",
     "csrc/kernels/spectrum.hip",
     "csrc/kernels/display.hip",
+    "csrc/kernels/fft.hip",
     "csrc/engine/engine.cpp",
 ]
 
@@ -50,7 +51,8 @@ def build_hip_objects():
                 os.path.join(ROOT, "csrc/include/srtb_kernels.h"),
                 os.path.join(ROOT, "csrc/kernels/common.h"),
                 os.path.join(ROOT, "csrc/engine/engine.h"),
-                os.path.join(ROOT, "csrc/fft/fft_plans.h")]
+                os.path.join(ROOT, "csrc/fft/fft_plans.h"),
+                os.path.join(ROOT, "csrc/fft/native_fft.h")]
         if os.path.exists(obj) and all(
                 os.path.getmtime(obj) >= os.path.getmtime(d) for d in deps
                 if os.path.exists(d)):

@@ -31,10 +31,10 @@ def native():
         import torch
         if torch.cuda.is_available():
             raise RuntimeError(
-                "GPU present but the srtb_amd HIP extension is not built. "
-                "Run `python setup.py build_ext --inplace` (gfx950). "
-                "Refusing to fall back to a non-HIP path on a GPU box."
-            ) from e
+                "GPU present but the srtb_amd HIP extension failed to "
+                f"import ({e!r}). Run `python setup.py build_ext --inplace` "
+                "(gfx950). Refusing to fall back to a non-HIP path on a GPU "
+                "box.") from e
         raise
 
 
