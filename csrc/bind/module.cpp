@@ -8,6 +8,7 @@
 #include <ATen/cuda/CUDAContext.h>
 #include <c10/cuda/CUDAGuard.h>
 
+#include <functional>
 #include <vector>
 
 #include "../engine/engine.h"
@@ -384,6 +385,92 @@ torch::Tensor t_native_rfft(torch::Tensor x) {
   return z;
 }
 
+// ---------------- FFT microbenchmarks (plan once, event-timed) ----------------
+
+double time_iters(hipStream_t stream, int iters, const std::function<void()>& f) {
+  hipEvent_t e0, e1;
+  check(hipEventCreate(&e0), "ev");
+  check(hipEventCreate(&e1), "ev");
+  f();  // warmup
+  check(hipStreamSynchronize(stream), "warm sync");
+  check(hipEventRecord(e0, stream), "rec");
+  for (int i = 0; i < iters; ++i) f();
+  check(hipEventRecord(e1, stream), "rec");
+  check(hipEventSynchronize(e1), "sync");
+  float ms = 0;
+  check(hipEventElapsedTime(&ms, e0, e1), "elapsed");
+  hipEventDestroy(e0);
+  hipEventDestroy(e1);
+  return ms / iters;
+}
+
+double t_bench_fft(int64_t len, int64_t batch, int64_t sign, int64_t iters,
+                   std::string backend) {
+  auto stream = cur_stream();
+  auto x = torch::randn({batch, len, 2},
+                        torch::TensorOptions().dtype(torch::kFloat32)
+                            .device(torch::kCUDA));
+  auto y = torch::empty_like(x);
+  float2* xp = reinterpret_cast<float2*>(x.data_ptr());
+  float2* yp = reinterpret_cast<float2*>(y.data_ptr());
+  double ms = 0;
+  if (backend == "native") {
+    NativeFft plan;
+    plan.plan(len, batch, (int)sign, stream);
+    ms = time_iters(stream, (int)iters, [&] { plan.exec(xp, yp, stream); });
+  } else {
+    hipfftHandle h;
+    check_fft(hipfftCreate(&h), "create");
+    long long n1[1] = {(long long)len};
+    size_t ws = 0;
+    check_fft(hipfftMakePlanMany64(h, 1, n1, nullptr, 1, len, nullptr, 1, len,
+                                   HIPFFT_C2C, batch, &ws),
+              "plan");
+    check_fft(hipfftSetStream(h, stream), "stream");
+    ms = time_iters(stream, (int)iters, [&] {
+      hipfftExecC2C(h, reinterpret_cast<hipfftComplex*>(xp),
+                    reinterpret_cast<hipfftComplex*>(yp),
+                    sign < 0 ? HIPFFT_FORWARD : HIPFFT_BACKWARD);
+    });
+    hipfftDestroy(h);
+  }
+  return ms;
+}
+
+double t_bench_rfft(int64_t n, int64_t iters, std::string backend) {
+  auto stream = cur_stream();
+  auto x = torch::randn({n}, torch::TensorOptions()
+                                 .dtype(torch::kFloat32)
+                                 .device(torch::kCUDA));
+  auto y = torch::empty({n / 2 + 1, 2}, x.options());
+  float* xp = x.data_ptr<float>();
+  float2* yp = reinterpret_cast<float2*>(y.data_ptr());
+  double ms = 0;
+  if (backend == "native") {
+    const size_t m = n / 2;
+    NativeFft plan;
+    plan.plan(m, 1, -1, stream);
+    ms = time_iters(stream, (int)iters, [&] {
+      plan.exec(reinterpret_cast<float2*>(xp), yp, stream);
+      check(r2c_post_process(yp, yp, m, nullptr, nullptr, stream), "post");
+    });
+  } else {
+    hipfftHandle h;
+    check_fft(hipfftCreate(&h), "create");
+    long long n1[1] = {(long long)n};
+    size_t ws = 0;
+    check_fft(hipfftMakePlanMany64(h, 1, n1, nullptr, 1, 0, nullptr, 1, 0,
+                                   HIPFFT_R2C, 1, &ws),
+              "plan");
+    check_fft(hipfftSetStream(h, stream), "stream");
+    ms = time_iters(stream, (int)iters, [&] {
+      hipfftExecR2C(h, xp, reinterpret_cast<hipfftComplex*>(yp));
+    });
+    hipfftDestroy(h);
+  }
+  return ms;
+}
+
 // ---------------- engine binding ----------------
 
 class PyEngine {
@@ -521,6 +608,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("correlate", &t_correlate);
   m.def("native_fft", &t_native_fft, py::arg("x"), py::arg("sign"));
   m.def("native_rfft", &t_native_rfft, py::arg("x"));
+  m.def("bench_fft", &t_bench_fft, py::arg("len"), py::arg("batch"),
+        py::arg("sign"), py::arg("iters") = 20, py::arg("backend") = "native");
+  m.def("bench_rfft", &t_bench_rfft, py::arg("n"), py::arg("iters") = 20,
+        py::arg("backend") = "native");
 
   py::class_<PyEngine>(m, "PipelineEngine")
       .def(py::init<int64_t, int64_t, int64_t, double, double, double, double,
