@@ -59,7 +59,7 @@ class NativeFft {
     if (f.size() == 1) {
       Pass p;
       p.d.n = f[0];
-      p.d.d0 = 0;  // set at exec: n_ffts
+      p.d.d0 = 0;  // sentinel: q0 = id (contiguous rows; batch not pow2-safe)
       p.d.in_c0 = L;  // row stride per instance (q0 = global row)
       p.d.in_stride = 1;
       p.d.out_c0 = L;
@@ -159,11 +159,10 @@ class NativeFft {
       float2* dst = last ? out : cur;
       FftPassDesc d = p.d;
       size_t n_ffts = p.n_ffts;
-      if (p.contig_rows) d.d0 = (uint32_t)std::max<size_t>(n_ffts, 1);
       const int F = pick_f(d.n, p, n_ffts);
       check_hip(fft_stockham_pass(cur, dst, d, n_ffts, F, p.load_ffast,
-                                  p.store_ffast, p.tw_n, p.tw_hi, p.tw_lo,
-                                  stream),
+                                  p.store_ffast, sign_, p.tw_n, p.tw_hi,
+                                  p.tw_lo, stream),
                 "fft_stockham_pass");
       cur = dst;
     }
@@ -220,8 +219,10 @@ class NativeFft {
   int pick_f(uint32_t n, const Pass& p, size_t n_ffts) const {
     size_t f = (size_t)kElemsPerWg / n;
     if (f < 1) f = 1;
-    // keep LDS under 160 KiB: 2*F*(n+2)*8
-    while (f > 1 && 2ull * f * (n + 2) * sizeof(float2) > 160 * 1024) f >>= 1;
+    // keep LDS under 160 KiB: (n + 2*F*(n+2)) * 8  (tw table + ping-pong)
+    while (f > 1 &&
+           ((size_t)n + 2ull * f * (n + 2)) * sizeof(float2) > 160 * 1024)
+      f >>= 1;
     while (f > 1 && n_ffts % f != 0) f >>= 1;
     // instances in a workgroup must share q1/q2 digits only if... they need
     // not; addressing is exact per instance.  But f-fast coalescing wants
@@ -234,10 +235,11 @@ class NativeFft {
   void ensure_len_table(uint32_t n, int sign, hipStream_t stream) {
     if (len_table(n)) return;
     float2* t = nullptr;
-    check_hip(hipMalloc(&t, std::max<size_t>(n / 2, 1) * sizeof(float2)),
+    // FULL circle: radix-4 stages index up to 3n/4
+    check_hip(hipMalloc(&t, std::max<size_t>(n, 1) * sizeof(float2)),
               "tw_n alloc");
-    check_hip(fft_build_twiddle(t, std::max<size_t>(n / 2, 1), (double)n,
-                                sign, stream),
+    check_hip(fft_build_twiddle(t, std::max<size_t>(n, 1), (double)n, sign,
+                                stream),
               "tw_n build");
     tables_.push_back({0, n, t});
   }

@@ -149,7 +149,8 @@ hipError_t boxcar(const float* cumsum, float* out, size_t n_out, size_t L,
 // csrc/fft/native_fft.h.
 struct FftPassDesc {
   uint32_t n;            // pow2 FFT length of this pass (<= 4096)
-  uint32_t d0 = 1, d1 = 1;  // instance id -> digits q0, q1, q2
+  uint32_t d0 = 1, d1 = 1;  // instance id -> digits q0, q1, q2 (pow2;
+                            // d0 == 0 means "q0 = id" for contiguous rows)
   unsigned long long in_c0 = 0, in_c1 = 0, in_c2 = 0;
   unsigned long long in_stride = 1;
   unsigned long long out_c0 = 0, out_c1 = 0, out_c2 = 0;
@@ -163,9 +164,10 @@ struct FftPassDesc {
 hipError_t fft_build_twiddle(float2* table, size_t count, double m, int sign,
                              hipStream_t stream);
 
+// tw_n: FULL-circle per-length table (n entries, staged into LDS)
 hipError_t fft_stockham_pass(const float2* in, float2* out,
                              const FftPassDesc& d, size_t n_ffts, int F,
-                             bool load_ffast, bool store_ffast,
+                             bool load_ffast, bool store_ffast, int sign,
                              const float2* tw_n, const float2* tw_hi,
                              const float2* tw_lo, hipStream_t stream);
 

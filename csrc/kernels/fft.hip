@@ -1,27 +1,31 @@
 // Hand-written LDS-staged Stockham FFT for CDNA4 (gfx950).
 //
 // One generic pass kernel covers every shape the pipeline needs:
-//  - batched contiguous C2C (any pow2 length ≤ kFftMaxLen per pass),
+//  - batched contiguous C2C (any pow2 length ≤ 4096 per pass),
 //  - four-step composite passes: strided column FFTs with fused inter-pass
 //    twiddles (split-table exact), and row FFTs whose digit-reversal output
 //    scatter replaces rocFFT's separate transpose kernels entirely,
 //  - the r2c post-process (packed-real trick) as a separate small kernel.
 //
-// Index math is the NumPy oracle in srtb_amd/fftref.py (fft_small /
+// Index math is the NumPy oracle in srtb_amd/fftref.py (fft_small_r4 /
 // fft_four_step / fft_deep / r2c_post); conventions are cuFFT's (forward
 // sign -1, backward +1, unnormalized).
 //
 // Design notes (MI355X):
 //  - Stockham auto-sort ping-pong in LDS: no bit-reversal permutation, so
 //    global loads/stores stay linear per instance group and LDS access is
-//    stride-regular; rows padded by one float2 to break power-of-2 bank
-//    strides (G4).
+//    stride-regular; rows padded by one float2 to break pow2 bank strides.
+//  - mixed radix-4 stages (one final radix-2 for odd log2) halve LDS traffic
+//    and barrier count vs radix-2.
+//  - ALL addressing is shift/mask (every radix, digit and group count is a
+//    power of two): per-element 64-bit divisions cost ~10x on these loops.
+//  - the per-length butterfly twiddle table (full circle, n entries) is
+//    staged into LDS once per workgroup; inter-pass twiddles
+//    e^{s*2πi*m/M} are exact via split tables THi[m>>b]*TLo[m&(2^b-1)]
+//    built in fp64 at plan time (L2-resident).
 //  - each workgroup (256 threads) processes F = ELEMS/n FFT instances; for
 //    strided passes instances are q0-consecutive so a row of F elements is
-//    contiguous in HBM (F*8 bytes per transaction run).
-//  - butterfly twiddles come from a per-length global table (L2-resident,
-//    broadcast reads); inter-pass twiddles e^{s*2πi*m/M} are exact via
-//    split tables THi[m>>b]*TLo[m&(2^b-1)] computed in fp64 at plan build.
+//    contiguous in HBM.
 
 #include "common.h"
 #include "../include/srtb_kernels.h"
@@ -36,85 +40,131 @@ __device__ inline float2 cmulf(float2 a, float2 b) {
 
 }  // namespace
 
-// descriptor for one FFT pass (see srtb_kernels.h for the field contract)
+// device-side pass descriptor; counts as log2 (see srtb_kernels.h contract)
 struct FftPassDescDev {
-  uint32_t n;            // pow2 FFT length of this pass
-  uint32_t d0, d1;       // instance id -> q0 = id%d0, q1=(id/d0)%d1, q2=rest
-  unsigned long long in_c0, in_c1, in_c2;    // input base = Σ q_i * c_i
+  uint32_t n;
+  int n_log2, f_log2;
+  int d0_log2, d1_log2;
+  unsigned long long in_c0, in_c1, in_c2;
   unsigned long long in_stride;
   unsigned long long out_c0, out_c1, out_c2;
   unsigned long long out_stride;
-  unsigned long long tw_f0, tw_f1;  // twiddle factor = q0*tw_f0 + q1*tw_f1
-  unsigned long long tw_mask;       // modulus-1 (modulus = pow2), 0 = off
+  unsigned long long tw_f0, tw_f1;
+  unsigned long long tw_mask;  // modulus-1; twiddle enabled via template
   int tw_lo_bits;
 };
 
 namespace {
 
-template <bool LOAD_FFAST, bool STORE_FFAST, bool TWIDDLE>
+__device__ inline void digits(unsigned long long id, const FftPassDescDev& d,
+                              unsigned long long& q0, unsigned long long& q1,
+                              unsigned long long& q2) {
+  q0 = id & ((1ull << d.d0_log2) - 1);
+  const unsigned long long r = id >> d.d0_log2;
+  q1 = r & ((1ull << d.d1_log2) - 1);
+  q2 = r >> d.d1_log2;
+}
+
+// LDS layout: [tw: n][X: F*(n+2)][Y: F*(n+2)] float2s.
+template <bool LOAD_FFAST, bool STORE_FFAST, bool TWIDDLE, int SIGN>
 __global__ void __launch_bounds__(256)
     k_fft_stockham(const float2* __restrict__ in, float2* __restrict__ out,
-                   FftPassDescDev d, int F,
-                   const float2* __restrict__ tw_n,
+                   FftPassDescDev d, const float2* __restrict__ tw_n,
                    const float2* __restrict__ tw_hi,
                    const float2* __restrict__ tw_lo) {
   extern __shared__ float2 lds[];
   const int n = d.n;
-  const int ldst = n + 2;  // padded row stride (breaks pow2 bank conflicts)
-  float2* X = lds;
-  float2* Y = lds + (size_t)F * ldst;
-  const unsigned long long fft0 = (unsigned long long)blockIdx.x * F;
-  const int total = F * n;
+  const int nl = d.n_log2;
+  const int F = 1 << d.f_log2;
+  const int ldst = n + 2;
+  float2* ltw = lds;
+  float2* X = lds + n;
+  float2* Y = X + (size_t)F * ldst;
+  const unsigned long long fft0 = (unsigned long long)blockIdx.x << d.f_log2;
+  const int total = F << nl;
+
+  for (int j = threadIdx.x; j < n; j += blockDim.x) ltw[j] = tw_n[j];
 
   // ---- load ----
   for (int e = threadIdx.x; e < total; e += blockDim.x) {
     int f, i;
-    if (LOAD_FFAST) { f = e % F; i = e / F; }
-    else            { f = e / n; i = e % n; }
-    const unsigned long long id = fft0 + f;
-    const unsigned long long q0 = id % d.d0;
-    const unsigned long long r = id / d.d0;
-    const unsigned long long q1 = r % d.d1;
-    const unsigned long long q2 = r / d.d1;
+    if (LOAD_FFAST) { f = e & (F - 1); i = e >> d.f_log2; }
+    else            { f = e >> nl; i = e & (n - 1); }
+    unsigned long long q0, q1, q2;
+    digits(fft0 + f, d, q0, q1, q2);
     const unsigned long long base =
         q0 * d.in_c0 + q1 * d.in_c1 + q2 * d.in_c2;
     X[f * ldst + i] = in[base + (unsigned long long)i * d.in_stride];
   }
   __syncthreads();
 
-  // ---- Stockham stages (radix-2 ping-pong; fftref.fft_small/fft0 math) ----
-  const int half = n >> 1;
-  int tstep = 1;  // n / ncur
-  for (int ncur = n, s = 1; ncur > 1; ncur >>= 1, s <<= 1, tstep <<= 1) {
-    const int m = ncur >> 1;
-    for (int b = threadIdx.x; b < F * half; b += blockDim.x) {
-      const int f = b / half;
-      const int bb = b - f * half;
-      const int p = bb / s;
-      const int q = bb - p * s;
-      const float2 a = X[f * ldst + q + s * p];
-      const float2 c = X[f * ldst + q + s * (p + m)];
-      const float2 w = tw_n[(size_t)p * tstep];
-      const float2 diff = make_float2(a.x - c.x, a.y - c.y);
-      Y[f * ldst + q + s * (2 * p)] = make_float2(a.x + c.x, a.y + c.y);
-      Y[f * ldst + q + s * (2 * p + 1)] = cmulf(diff, w);
+  // ---- radix-4 stages (fftref.fft_small_r4 math) ----
+  int ncur = n, s = 1, tstep_log2 = 0;
+  const int quarter = n >> 2;
+  while ((ncur & 3) == 0 && ncur > 1) {
+    const int m = ncur >> 2;
+    const int s_log2 = tstep_log2;  // s == 1 << tstep_log2 here
+    for (int b = threadIdx.x; b < F * quarter; b += blockDim.x) {
+      const int f = b >> (nl - 2);
+      const int bb = b & (quarter - 1);
+      const int p = bb >> s_log2;
+      const int q = bb & (s - 1);
+      const float2* row = X + f * ldst;
+      const float2 a = row[q + s * p];
+      const float2 bv = row[q + s * (p + m)];
+      const float2 c = row[q + s * (p + 2 * m)];
+      const float2 dv = row[q + s * (p + 3 * m)];
+      const float2 apc = make_float2(a.x + c.x, a.y + c.y);
+      const float2 amc = make_float2(a.x - c.x, a.y - c.y);
+      const float2 bpd = make_float2(bv.x + dv.x, bv.y + dv.y);
+      const float2 bmd = make_float2(bv.x - dv.x, bv.y - dv.y);
+      // si*(b-d) with si = SIGN*i: i*(x,y) = (-y, x)
+      const float2 sibmd = (SIGN > 0) ? make_float2(-bmd.y, bmd.x)
+                                      : make_float2(bmd.y, -bmd.x);
+      const float2 u0 = make_float2(apc.x + bpd.x, apc.y + bpd.y);
+      const float2 u2 = make_float2(apc.x - bpd.x, apc.y - bpd.y);
+      const float2 u1 = make_float2(amc.x + sibmd.x, amc.y + sibmd.y);
+      const float2 u3 = make_float2(amc.x - sibmd.x, amc.y - sibmd.y);
+      const int i1 = p << tstep_log2;
+      const float2 w1 = ltw[i1];
+      const float2 w2 = ltw[i1 * 2];
+      const float2 w3 = ltw[i1 * 3];
+      float2* orow = Y + f * ldst;
+      const int ob = q + ((p * s) << 2);
+      orow[ob] = u0;
+      orow[ob + s] = cmulf(u1, w1);
+      orow[ob + 2 * s] = cmulf(u2, w2);
+      orow[ob + 3 * s] = cmulf(u3, w3);
     }
     __syncthreads();
-    float2* t = X;
-    X = Y;
-    Y = t;
+    float2* t = X; X = Y; Y = t;
+    ncur >>= 2;
+    s <<= 2;
+    tstep_log2 += 2;
+  }
+  // ---- final radix-2 stage (odd log2(n)) ----
+  if (ncur == 2) {
+    const int half = n >> 1;  // == s at this point
+    for (int b = threadIdx.x; b < F * half; b += blockDim.x) {
+      const int f = b >> (nl - 1);
+      const int q = b & (half - 1);
+      const float2* row = X + f * ldst;
+      const float2 a = row[q];
+      const float2 c = row[q + half];
+      Y[f * ldst + q] = make_float2(a.x + c.x, a.y + c.y);
+      Y[f * ldst + q + half] = make_float2(a.x - c.x, a.y - c.y);
+    }
+    __syncthreads();
+    float2* t = X; X = Y; Y = t;
   }
 
   // ---- store (+ inter-pass twiddle) ----
   for (int e = threadIdx.x; e < total; e += blockDim.x) {
     int f, k;
-    if (STORE_FFAST) { f = e % F; k = e / F; }
-    else             { f = e / n; k = e % n; }
-    const unsigned long long id = fft0 + f;
-    const unsigned long long q0 = id % d.d0;
-    const unsigned long long r = id / d.d0;
-    const unsigned long long q1 = r % d.d1;
-    const unsigned long long q2 = r / d.d1;
+    if (STORE_FFAST) { f = e & (F - 1); k = e >> d.f_log2; }
+    else             { f = e >> nl; k = e & (n - 1); }
+    unsigned long long q0, q1, q2;
+    digits(fft0 + f, d, q0, q1, q2);
     float2 v = X[f * ldst + k];
     if constexpr (TWIDDLE) {
       const unsigned long long tf = q0 * d.tw_f0 + q1 * d.tw_f1;
@@ -129,7 +179,7 @@ __global__ void __launch_bounds__(256)
   }
 }
 
-// twiddle-table builders (fp64 on device)
+// twiddle-table builder (fp64 on device)
 __global__ void k_build_twiddle(float2* __restrict__ t, size_t count,
                                 double sign_two_pi_over_m) {
   const size_t stride = (size_t)gridDim.x * blockDim.x;
@@ -145,11 +195,10 @@ __global__ void k_build_twiddle(float2* __restrict__ t, size_t count,
 //   E = (Z[k]+conj(Z[M-k]))/2,  O = -i/2*(Z[k]-conj(Z[M-k])),
 //   w(k) = exp(-2πi k/(2M)),
 //   X[k] = E + w*O  and  X[M-k] = conj(E - w*O),
-// so each thread handles the (k, M-k) PAIR — this makes the kernel
-// in-place-safe (x may alias z).  k=0 gives X[0]=Re(Z0)+Im(Z0) (the would-be
-// X[M] Nyquist bin is dropped, matching the pipeline count Nc).  Optionally
-// accumulates Σ|X|² partials (fused mean-power for RFI s1 — saves a full
-// 4 GB spectrum read).
+// each thread handles the (k, M-k) PAIR so the kernel is in-place safe
+// (x may alias z).  k=0: X[0]=Re(Z0)+Im(Z0) (Nyquist dropped, count = Nc).
+// Optionally accumulates Σ|X|² partials (fused RFI-s1 mean-power — saves a
+// full 4 GB spectrum read).
 template <bool MEANP>
 __global__ void k_r2c_post(const float2* __restrict__ z,
                            float2* __restrict__ x, size_t m,
@@ -164,7 +213,6 @@ __global__ void k_r2c_post(const float2* __restrict__ z,
     const float2 zmc = make_float2(zm.x, -zm.y);
     const float2 even = make_float2(0.5f * (zk.x + zmc.x),
                                     0.5f * (zk.y + zmc.y));
-    // odd = -0.5i * (zk - conj(z[m-k]))
     const float2 dif = make_float2(zk.x - zmc.x, zk.y - zmc.y);
     const float2 odd = make_float2(0.5f * dif.y, -0.5f * dif.x);
     double sw, cw;
@@ -198,6 +246,12 @@ __global__ void k_r2c_post_finish_mean(const double* __restrict__ partials,
 
 constexpr int kR2cPostBlocks = 1024;
 
+inline int ilog2(unsigned long long v) {
+  int t = 0;
+  while ((1ull << t) < v) ++t;
+  return t;
+}
+
 }  // namespace
 
 hipError_t fft_build_twiddle(float2* table, size_t count, double m, int sign,
@@ -210,13 +264,19 @@ hipError_t fft_build_twiddle(float2* table, size_t count, double m, int sign,
 
 hipError_t fft_stockham_pass(const float2* in, float2* out,
                              const FftPassDesc& hd, size_t n_ffts, int F,
-                             bool load_ffast, bool store_ffast,
+                             bool load_ffast, bool store_ffast, int sign,
                              const float2* tw_n, const float2* tw_hi,
                              const float2* tw_lo, hipStream_t stream) {
+  if ((hd.n & (hd.n - 1)) || (F & (F - 1))) return hipErrorInvalidValue;
+  if (hd.d0 & (hd.d0 - 1)) return hipErrorInvalidValue;
+  if (hd.d1 & (hd.d1 - 1)) return hipErrorInvalidValue;
   FftPassDescDev d;
   d.n = hd.n;
-  d.d0 = hd.d0;
-  d.d1 = hd.d1;
+  d.n_log2 = ilog2(hd.n);
+  d.f_log2 = ilog2((unsigned)F);
+  // d0 == 0: contiguous-rows sentinel -> q0 = id (48-bit mask), q1 = q2 = 0
+  d.d0_log2 = hd.d0 ? ilog2(hd.d0) : 48;
+  d.d1_log2 = ilog2(hd.d1 ? hd.d1 : 1);
   d.in_c0 = hd.in_c0; d.in_c1 = hd.in_c1; d.in_c2 = hd.in_c2;
   d.in_stride = hd.in_stride;
   d.out_c0 = hd.out_c0; d.out_c1 = hd.out_c1; d.out_c2 = hd.out_c2;
@@ -227,20 +287,33 @@ hipError_t fft_stockham_pass(const float2* in, float2* out,
   const bool twiddle = hd.tw_mod != 0;
   if (n_ffts % F != 0) return hipErrorInvalidValue;
   const uint32_t grid = (uint32_t)(n_ffts / F);
-  const size_t lds_bytes = 2ull * F * (hd.n + 2) * sizeof(float2);
+  const size_t lds_bytes =
+      ((size_t)hd.n + 2ull * F * (hd.n + 2)) * sizeof(float2);
   if (lds_bytes > 160 * 1024) return hipErrorInvalidValue;
 
-#define DISPATCH(LF, SF, TW)                                                \
-  hipLaunchKernelGGL((k_fft_stockham<LF, SF, TW>), dim3(grid), dim3(256),   \
-                     lds_bytes, stream, in, out, d, F, tw_n, tw_hi, tw_lo)
+#define DISPATCH4(LF, SF, TW, SG)                                           \
+  hipLaunchKernelGGL((k_fft_stockham<LF, SF, TW, SG>), dim3(grid),          \
+                     dim3(256), lds_bytes, stream, in, out, d, tw_n, tw_hi, \
+                     tw_lo)
+#define DISPATCH2(LF, SF)                                                   \
+  do {                                                                      \
+    if (twiddle) {                                                          \
+      if (sign < 0) DISPATCH4(LF, SF, true, -1);                            \
+      else          DISPATCH4(LF, SF, true, 1);                             \
+    } else {                                                                \
+      if (sign < 0) DISPATCH4(LF, SF, false, -1);                           \
+      else          DISPATCH4(LF, SF, false, 1);                            \
+    }                                                                       \
+  } while (0)
   if (load_ffast) {
-    if (store_ffast) { if (twiddle) DISPATCH(true, true, true); else DISPATCH(true, true, false); }
-    else             { if (twiddle) DISPATCH(true, false, true); else DISPATCH(true, false, false); }
+    if (store_ffast) DISPATCH2(true, true);
+    else             DISPATCH2(true, false);
   } else {
-    if (store_ffast) { if (twiddle) DISPATCH(false, true, true); else DISPATCH(false, true, false); }
-    else             { if (twiddle) DISPATCH(false, false, true); else DISPATCH(false, false, false); }
+    if (store_ffast) DISPATCH2(false, true);
+    else             DISPATCH2(false, false);
   }
-#undef DISPATCH
+#undef DISPATCH2
+#undef DISPATCH4
   SRTB_CHECK_LAUNCH();
   return hipSuccess;
 }

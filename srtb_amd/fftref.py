@@ -73,6 +73,66 @@ def fft_small(x: np.ndarray, sign: int, dtype=np.complex64) -> np.ndarray:
     return y.astype(dtype)
 
 
+def twiddle_table_full(n: int, sign: int) -> np.ndarray:
+    """Full-circle table tw[j] = exp(sign*2πi*j/n), j in [0, n) — the radix-4
+    stages need indices up to 3n/4."""
+    j = np.arange(n)
+    return np.exp(sign * 2j * np.pi * j / n)
+
+
+def fft_small_r4(x: np.ndarray, sign: int, dtype=np.complex64) -> np.ndarray:
+    """Mixed radix-4/radix-2 Stockham (ping-pong, auto-sort) — EXACTLY the
+    HIP kernel's stage structure: radix-4 stages while n_cur % 4 == 0, one
+    final radix-2 stage when log2(n) is odd.
+
+    Stage (radix 4), n_cur, s: m = n_cur/4; for p in [0,m), q in [0,s):
+      a,b,c,d = x[q+s(p+km)] k=0..3;  si = sign*1j
+      u0 = a+b+c+d; u1 = a+si*b-c-si*d; u2 = a-b+c-d; u3 = a-si*b-c+si*d
+      y[q+s(4p+j)] = u_j * w^(j*p),  w = exp(sign*2πi/n_cur)
+    then n_cur /= 4, s *= 4.
+    """
+    n = x.size
+    assert n & (n - 1) == 0
+    tw = twiddle_table_full(n, sign)
+    X = np.asarray(x, dtype=np.complex128).copy()
+    Y = np.empty_like(X)
+    si = sign * 1j
+    n_cur, s = n, 1
+    while n_cur % 4 == 0 and n_cur > 1:
+        m = n_cur // 4
+        tstep = n // n_cur
+        for p in range(m):
+            w1 = tw[p * tstep]
+            w2 = tw[2 * p * tstep]
+            w3 = tw[3 * p * tstep]
+            for q in range(s):
+                a = X[q + s * p]
+                b = X[q + s * (p + m)]
+                c = X[q + s * (p + 2 * m)]
+                d = X[q + s * (p + 3 * m)]
+                u0 = a + b + c + d
+                u1 = a + si * b - c - si * d
+                u2 = a - b + c - d
+                u3 = a - si * b - c + si * d
+                Y[q + s * (4 * p + 0)] = u0
+                Y[q + s * (4 * p + 1)] = u1 * w1
+                Y[q + s * (4 * p + 2)] = u2 * w2
+                Y[q + s * (4 * p + 3)] = u3 * w3
+        X, Y = Y, X
+        n_cur //= 4
+        s *= 4
+    if n_cur == 2:
+        m = 1
+        tstep = n // 2
+        for q in range(s):
+            a = X[q]
+            b = X[q + s]
+            Y[q] = a + b
+            Y[q + s] = a - b
+        X, Y = Y, X
+    return X.astype(dtype)
+
+
 def fft_four_step(x: np.ndarray, l1: int, l2: int, sign: int,
                   fft1=None, fft2=None) -> np.ndarray:
     """Composite FFT of length l1*l2 per the docstring above."""
