@@ -15,6 +15,7 @@
 
 #include <algorithm>
 #include <cstdint>
+#include <cstdlib>
 #include <stdexcept>
 #include <vector>
 
@@ -26,7 +27,15 @@ namespace srtb_hip {
 class NativeFft {
  public:
   static constexpr uint32_t kMaxPassLen = 4096;
-  static constexpr int kElemsPerWg = 4096;
+  // elements per workgroup: smaller -> more blocks/CU (latency hiding),
+  // larger -> longer coalesced runs on strided passes.  Sweepable via env.
+  static int elems_per_wg() {
+    const char* e = std::getenv("SRTB_FFT_ELEMS");
+    int v = e ? std::atoi(e) : 4096;
+    if (v < 512) v = 512;
+    if (v > 8192) v = 8192;
+    return v;
+  }
 
   static bool is_pow2(size_t v) { return v && (v & (v - 1)) == 0; }
 
@@ -217,7 +226,7 @@ class NativeFft {
   }
 
   int pick_f(uint32_t n, const Pass& p, size_t n_ffts) const {
-    size_t f = (size_t)kElemsPerWg / n;
+    size_t f = (size_t)elems_per_wg() / n;
     if (f < 1) f = 1;
     // keep LDS under 160 KiB: (n + 2*F*(n+2)) * 8  (tw table + ping-pong)
     while (f > 1 &&

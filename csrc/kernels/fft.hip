@@ -85,16 +85,40 @@ __global__ void __launch_bounds__(256)
 
   for (int j = threadIdx.x; j < n; j += blockDim.x) ltw[j] = tw_n[j];
 
-  // ---- load ----
-  for (int e = threadIdx.x; e < total; e += blockDim.x) {
-    int f, i;
-    if (LOAD_FFAST) { f = e & (F - 1); i = e >> d.f_log2; }
-    else            { f = e >> nl; i = e & (n - 1); }
-    unsigned long long q0, q1, q2;
-    digits(fft0 + f, d, q0, q1, q2);
-    const unsigned long long base =
-        q0 * d.in_c0 + q1 * d.in_c1 + q2 * d.in_c2;
-    X[f * ldst + i] = in[base + (unsigned long long)i * d.in_stride];
+  // ---- load (register-staged in chunks of 8 so ≥8 global loads stay in
+  // flight per thread; a naive load→ds_write loop serializes on vmcnt) ----
+  {
+    const int iters = total >> 8;  // blockDim == 256
+    int it = 0;
+    for (; it + 8 <= iters; it += 8) {
+      float2 tmp[8];
+      int lidx[8];
+#pragma unroll
+      for (int k = 0; k < 8; ++k) {
+        const int e = ((it + k) << 8) + threadIdx.x;
+        int f, i;
+        if (LOAD_FFAST) { f = e & (F - 1); i = e >> d.f_log2; }
+        else            { f = e >> nl; i = e & (n - 1); }
+        unsigned long long q0, q1, q2;
+        digits(fft0 + f, d, q0, q1, q2);
+        const unsigned long long base =
+            q0 * d.in_c0 + q1 * d.in_c1 + q2 * d.in_c2;
+        tmp[k] = in[base + (unsigned long long)i * d.in_stride];
+        lidx[k] = f * ldst + i;
+      }
+#pragma unroll
+      for (int k = 0; k < 8; ++k) X[lidx[k]] = tmp[k];
+    }
+    for (int e = (it << 8) + threadIdx.x; e < total; e += blockDim.x) {
+      int f, i;
+      if (LOAD_FFAST) { f = e & (F - 1); i = e >> d.f_log2; }
+      else            { f = e >> nl; i = e & (n - 1); }
+      unsigned long long q0, q1, q2;
+      digits(fft0 + f, d, q0, q1, q2);
+      const unsigned long long base =
+          q0 * d.in_c0 + q1 * d.in_c1 + q2 * d.in_c2;
+      X[f * ldst + i] = in[base + (unsigned long long)i * d.in_stride];
+    }
   }
   __syncthreads();
 
