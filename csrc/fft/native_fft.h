@@ -58,99 +58,88 @@ class NativeFft {
     batch_ = batch;
     sign_ = sign;
 
-    std::vector<uint32_t> f = factorize(len);
     const size_t L = len;
-
-    // twiddle tables per distinct pass length
-    for (uint32_t fl : f) ensure_len_table(fl, sign, stream);
-
+    // Decomposition (see csrc/kernels/fft.hip):
+    //   t = log2(len) <= 12           -> single LDS Stockham pass
+    //   t > 12: column factors (register kernel, N in {2..64}) + final 256
+    //           (in-place DIF in LDS, digit-reversal folded into the store)
+    const int t = ilog2z(len);
     passes_.clear();
-    if (f.size() == 1) {
+    if (t <= 12) {
+      ensure_len_table((uint32_t)len, sign, stream);
       Pass p;
-      p.d.n = f[0];
-      p.d.d0 = 0;  // sentinel: q0 = id (contiguous rows; batch not pow2-safe)
-      p.d.in_c0 = L;  // row stride per instance (q0 = global row)
+      p.kind = PassKind::kStockham;
+      p.d.n = (uint32_t)len;
+      p.d.d0 = 0;  // sentinel: q0 = id (contiguous rows)
+      p.d.in_c0 = L;
       p.d.in_stride = 1;
       p.d.out_c0 = L;
       p.d.out_stride = 1;
       p.n_ffts = batch;
-      p.load_ffast = false;
-      p.store_ffast = false;
-      p.contig_rows = true;
-      p.tw_n = len_table(f[0]);
+      p.load_ffast = p.store_ffast = false;
+      p.tw_n = len_table((uint32_t)len);
       passes_.push_back(p);
-    } else if (f.size() == 2) {
-      const uint32_t f0 = f[0], f1 = f[1];
-      ensure_mod_table((size_t)f0 * f1, sign, stream);
-      {  // pass 1: columns (len f0, stride f1), twiddle mod L
-        Pass p;
-        p.d.n = f0;
-        p.d.d0 = f1; p.d.d1 = 1;
-        p.d.in_c0 = 1; p.d.in_c2 = L; p.d.in_stride = f1;
-        p.d.out_c0 = 1; p.d.out_c2 = L; p.d.out_stride = f1;
-        p.d.tw_f0 = 1; p.d.tw_mod = L; p.d.tw_lo_bits = mod_lo_bits(L);
-        p.n_ffts = batch * f1;
-        p.load_ffast = p.store_ffast = true;
-        p.tw_n = len_table(f0);
-        p.tw_hi = mod_hi(L); p.tw_lo = mod_lo(L);
-        passes_.push_back(p);
+      return;
+    }
+    std::vector<uint32_t> f;
+    {
+      int rest = t - 8;
+      while (rest > 6) {
+        f.push_back(64);
+        rest -= 6;
       }
-      {  // pass 2: rows (len f1, contiguous), scatter out[k1 + f0*k2]
-        Pass p;
-        p.d.n = f1;
-        p.d.d0 = f0; p.d.d1 = 1;
-        p.d.in_c0 = f1; p.d.in_c2 = L; p.d.in_stride = 1;
-        p.d.out_c0 = 1; p.d.out_c2 = L; p.d.out_stride = f0;
-        p.n_ffts = batch * f0;
-        p.load_ffast = false;
-        p.store_ffast = true;
-        p.tw_n = len_table(f1);
-        passes_.push_back(p);
+      f.push_back(1u << rest);
+      f.push_back(256);
+    }
+    const int m = (int)f.size();
+    if (m - 1 > 4) throw std::runtime_error("NativeFft: too many factors");
+
+    // suffix products M[j] = prod(f[j..m-1]); M[m] = 1
+    std::vector<size_t> M(m + 1, 1);
+    for (int j = m - 1; j >= 0; --j) M[j] = M[j + 1] * f[j];
+    // prefix products C[i] = prod(f[0..i-1])
+    std::vector<size_t> Cp(m + 1, 1);
+    for (int i = 1; i <= m; ++i) Cp[i] = Cp[i - 1] * f[i - 1];
+
+    for (int j = 0; j + 1 < m; ++j) {  // column passes
+      ensure_len_table(f[j], sign, stream);
+      ensure_mod_table(M[j], sign, stream);
+      Pass p;
+      p.kind = PassKind::kCol;
+      p.d.n = f[j];
+      p.d.d0 = (uint32_t)M[j + 1];
+      p.d.d1 = (uint32_t)Cp[j];
+      p.d.in_c0 = 1;
+      p.d.in_c1 = M[j];
+      p.d.in_c2 = L;
+      p.d.in_stride = M[j + 1];
+      p.d.tw_f0 = 1;
+      p.d.tw_mod = M[j];
+      p.d.tw_lo_bits = mod_lo_bits(M[j]);
+      p.n_ffts = batch * (L / f[j]);
+      p.tw_n = len_table(f[j]);
+      p.tw_hi = mod_hi(M[j]);
+      p.tw_lo = mod_lo(M[j]);
+      passes_.push_back(p);
+    }
+    {  // final DIF pass
+      const uint32_t fn = f[m - 1];
+      ensure_len_table(fn, sign, stream);
+      Pass p;
+      p.kind = PassKind::kDif;
+      p.d.n = fn;
+      p.dif.n = fn;
+      p.dif.out_c2 = L;
+      p.dif.out_elem_coef = Cp[m - 1];  // prod of all prefix factors
+      p.dif.n_prefix = m - 1;
+      for (int w = 0; w < m - 1; ++w) {
+        const int src = m - 2 - w;  // extraction order: last col digit first
+        p.dif.pf_bits[w] = ilog2z(f[src]);
+        p.dif.pf_coef[w] = Cp[src];
       }
-    } else {  // 3 factors
-      const uint32_t f0 = f[0], f1 = f[1], f2 = f[2];
-      const size_t f12 = (size_t)f1 * f2;
-      ensure_mod_table(L, sign, stream);
-      ensure_mod_table(f12, sign, stream);
-      {  // pass A: len f0, stride f1*f2, twiddle (q0=n_hat)*k0 mod L
-        Pass p;
-        p.d.n = f0;
-        p.d.d0 = (uint32_t)f12; p.d.d1 = 1;
-        p.d.in_c0 = 1; p.d.in_c2 = L; p.d.in_stride = f12;
-        p.d.out_c0 = 1; p.d.out_c2 = L; p.d.out_stride = f12;
-        p.d.tw_f0 = 1; p.d.tw_mod = L; p.d.tw_lo_bits = mod_lo_bits(L);
-        p.n_ffts = batch * f12;
-        p.load_ffast = p.store_ffast = true;
-        p.tw_n = len_table(f0);
-        p.tw_hi = mod_hi(L); p.tw_lo = mod_lo(L);
-        passes_.push_back(p);
-      }
-      {  // pass B: len f1, stride f2, within each [k0] chunk; tw mod f1*f2
-        Pass p;
-        p.d.n = f1;
-        p.d.d0 = f2; p.d.d1 = f0;
-        p.d.in_c0 = 1; p.d.in_c1 = f12; p.d.in_c2 = L; p.d.in_stride = f2;
-        p.d.out_c0 = 1; p.d.out_c1 = f12; p.d.out_c2 = L; p.d.out_stride = f2;
-        p.d.tw_f0 = 1; p.d.tw_mod = f12; p.d.tw_lo_bits = mod_lo_bits(f12);
-        p.n_ffts = batch * f0 * f2;
-        p.load_ffast = p.store_ffast = true;
-        p.tw_n = len_table(f1);
-        p.tw_hi = mod_hi(f12); p.tw_lo = mod_lo(f12);
-        passes_.push_back(p);
-      }
-      {  // pass C: len f2, contiguous rows (k0,k1); scatter k0 + f0*k1 + f0f1*k2
-        Pass p;
-        p.d.n = f2;
-        p.d.d0 = f0; p.d.d1 = f1;
-        p.d.in_c0 = f12; p.d.in_c1 = f2; p.d.in_c2 = L; p.d.in_stride = 1;
-        p.d.out_c0 = 1; p.d.out_c1 = f0; p.d.out_c2 = L;
-        p.d.out_stride = (size_t)f0 * f1;
-        p.n_ffts = batch * f0 * f1;
-        p.load_ffast = false;
-        p.store_ffast = true;
-        p.tw_n = len_table(f2);
-        passes_.push_back(p);
-      }
+      p.n_ffts = batch * (L / fn);
+      p.tw_n = len_table(fn);
+      passes_.push_back(p);
     }
   }
 
@@ -166,13 +155,30 @@ class NativeFft {
       Pass& p = passes_[i];
       const bool last = (i + 1 == passes_.size());
       float2* dst = last ? out : cur;
-      FftPassDesc d = p.d;
-      size_t n_ffts = p.n_ffts;
-      const int F = pick_f(d.n, p, n_ffts);
-      check_hip(fft_stockham_pass(cur, dst, d, n_ffts, F, p.load_ffast,
-                                  p.store_ffast, sign_, p.tw_n, p.tw_hi,
-                                  p.tw_lo, stream),
-                "fft_stockham_pass");
+      switch (p.kind) {
+        case PassKind::kStockham: {
+          const int F = pick_f(p.d.n, p, p.n_ffts);
+          check_hip(fft_stockham_pass(cur, dst, p.d, p.n_ffts, F,
+                                      p.load_ffast, p.store_ffast, sign_,
+                                      p.tw_n, p.tw_hi, p.tw_lo, stream),
+                    "fft_stockham_pass");
+          break;
+        }
+        case PassKind::kCol:
+          check_hip(fft_col_pass(cur, cur, p.d, p.n_ffts, sign_, p.tw_n,
+                                 p.tw_hi, p.tw_lo, stream),
+                    "fft_col_pass");
+          dst = cur;  // in place
+          break;
+        case PassKind::kDif: {
+          int F = 32;
+          while (F > 1 && p.n_ffts % F != 0) F >>= 1;
+          check_hip(fft_dif_final(cur, dst, p.dif, p.n_ffts, F, sign_,
+                                  p.tw_n, stream),
+                    "fft_dif_final");
+          break;
+        }
+      }
       cur = dst;
     }
   }
@@ -184,11 +190,14 @@ class NativeFft {
   }
 
  private:
+  enum class PassKind { kStockham, kCol, kDif };
+
   struct Pass {
+    PassKind kind = PassKind::kStockham;
     FftPassDesc d{};
+    DifFinalDesc dif{};
     size_t n_ffts = 0;
     bool load_ffast = false, store_ffast = false;
-    bool contig_rows = false;
     const float2* tw_n = nullptr;
     const float2* tw_hi = nullptr;
     const float2* tw_lo = nullptr;
@@ -200,7 +209,13 @@ class NativeFft {
     float2* ptr;
   };
 
-  static std::vector<uint32_t> factorize(size_t len) {
+  static int ilog2z(size_t v) {
+    int t = 0;
+    while ((1ull << t) < v) ++t;
+    return t;
+  }
+
+  static std::vector<uint32_t> factorize_unused(size_t len) {
     // split into <= 3 balanced pow2 factors, each <= kMaxPassLen;
     // the LAST factor is the largest (it gets the contiguous-load pass).
     if (len <= kMaxPassLen) return {(uint32_t)len};

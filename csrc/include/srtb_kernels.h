@@ -171,6 +171,29 @@ hipError_t fft_stockham_pass(const float2* in, float2* out,
                              const float2* tw_n, const float2* tw_hi,
                              const float2* tw_lo, hipStream_t stream);
 
+// Register-resident column FFT pass (N in {2,4,8,16,32,64}; one FFT per
+// thread fully in VGPRs; in-place: out must alias layout of in addressing;
+// uses the same FftPassDesc fields; out_* ignored, stores to input layout).
+hipError_t fft_col_pass(const float2* in, float2* out, const FftPassDesc& d,
+                        size_t n_ffts, int sign, const float2* tw_n,
+                        const float2* tw_hi, const float2* tw_lo,
+                        hipStream_t stream);
+
+// Final composite pass: in-place radix-4 DIF in LDS with base-4
+// digit-reversal folded into the store, plus multi-digit output scatter.
+// Input must be fully linear (instance blocks contiguous).  n = 4^t.
+struct DifFinalDesc {
+  uint32_t n;
+  unsigned long long out_c2;         // batch-row coefficient
+  unsigned long long out_elem_coef;  // output element stride
+  int n_prefix = 0;                  // prefix digits, extraction order
+  int pf_bits[4] = {0, 0, 0, 0};
+  unsigned long long pf_coef[4] = {0, 0, 0, 0};
+};
+hipError_t fft_dif_final(const float2* in, float2* out, const DifFinalDesc& d,
+                         size_t n_ffts, int F, int sign, const float2* tw_n,
+                         hipStream_t stream);
+
 // packed-real R2C finish: Z = C2C(x_even + i*x_odd) of length m -> true
 // spectrum X[0..m) (Nyquist dropped).  In-place safe (x may alias z).
 // If mean_partials != null (>= 1024 doubles), also writes mean|X|^2 to

@@ -203,6 +203,274 @@ __global__ void __launch_bounds__(256)
   }
 }
 
+// ---------------------------------------------------------------------------
+// Register-resident column FFT (one FFT per thread, N ≤ 64 in VGPRs).
+//
+// Used for the strided passes of large composites: a wave's 64 lanes own 64
+// CONSECUTIVE columns, so every load/store instruction touches a contiguous
+// 512-byte run regardless of the column stride — the LDS kernel's F-limited
+// runs (64 B at F=8) were the bandwidth killer on these passes.
+//
+// Algorithm: mixed-radix in-place DIT (radix-4 stages + trailing radix-2),
+// input permutation σ folded into the (compile-time) register indices —
+// validated in srtb_amd/fftref.py and the dit_mixed prototype.
+// ---------------------------------------------------------------------------
+
+constexpr int col_ilog2(int v) { return v <= 1 ? 0 : 1 + col_ilog2(v / 2); }
+
+// schedule: first-to-last radices = [4]*(t/2) + [2 if t odd];
+// σ_N(n) = (n % r_last)*M + σ_M(n / r_last)
+template <int N>
+constexpr int col_sigma(int n) {
+  int res = 0, x = n, cur = N;
+  // iterate from the LAST stage radix down; last radix is 2 iff t odd
+  int t = col_ilog2(N);
+  bool odd = (t & 1) != 0;
+  // stage radices last-to-first: if odd, radices are [2, 4, 4, ...] from last?
+  // Schedule first-to-last: [4,4,...,2] (trailing 2).  Last radix = 2 if odd.
+  bool first_iter = true;
+  while (cur > 1) {
+    int r;
+    if (first_iter && odd) r = 2;
+    else r = 4;
+    first_iter = false;
+    const int M = cur / r;
+    res += (x % r) * M;
+    x /= r;
+    cur = M;
+  }
+  return res;
+}
+
+template <int SIGN>
+__device__ inline void bfly2(float2& a, float2& b, float2 w) {
+  // DIT radix-2: y1 = b*w; out: a+y1, a-y1
+  const float2 y = cmulf(b, w);
+  b = make_float2(a.x - y.x, a.y - y.y);
+  a = make_float2(a.x + y.x, a.y + y.y);
+}
+
+template <int SIGN>
+__device__ inline void bfly4(float2& a, float2& b, float2& c, float2& d,
+                             float2 w1, float2 w2, float2 w3) {
+  const float2 y1 = cmulf(b, w1);
+  const float2 y2 = cmulf(c, w2);
+  const float2 y3 = cmulf(d, w3);
+  const float2 t0 = make_float2(a.x + y2.x, a.y + y2.y);
+  const float2 t1 = make_float2(a.x - y2.x, a.y - y2.y);
+  const float2 t2 = make_float2(y1.x + y3.x, y1.y + y3.y);
+  const float2 dmy = make_float2(y1.x - y3.x, y1.y - y3.y);
+  const float2 t3 = (SIGN > 0) ? make_float2(-dmy.y, dmy.x)
+                               : make_float2(dmy.y, -dmy.x);
+  a = make_float2(t0.x + t2.x, t0.y + t2.y);
+  b = make_float2(t1.x + t3.x, t1.y + t3.y);
+  c = make_float2(t0.x - t2.x, t0.y - t2.y);
+  d = make_float2(t1.x - t3.x, t1.y - t3.y);
+}
+
+// one DIT stage over the register array; L = output block size, radix r
+template <int N, int L, int R, int SIGN>
+__device__ inline void col_stage(float2 (&v)[N],
+                                 const float2* __restrict__ tw_n) {
+  constexpr int M = L / R;
+  constexpr int TS = N / L;  // tw index scale: W_L^x = tw_n[x*TS]
+#pragma unroll
+  for (int g = 0; g < N; g += L) {
+#pragma unroll
+    for (int j = 0; j < M; ++j) {
+      if constexpr (R == 2) {
+        const float2 w = (j == 0) ? make_float2(1.f, 0.f) : tw_n[j * TS];
+        bfly2<SIGN>(v[g + j], v[g + j + M], w);
+      } else {
+        float2 w1, w2, w3;
+        if (j == 0) {
+          w1 = w2 = w3 = make_float2(1.f, 0.f);
+        } else {
+          w1 = tw_n[j * TS];
+          w2 = tw_n[2 * j * TS];
+          w3 = tw_n[3 * j * TS];
+        }
+        bfly4<SIGN>(v[g + j], v[g + j + M], v[g + j + 2 * M],
+                    v[g + j + 3 * M], w1, w2, w3);
+      }
+    }
+  }
+}
+
+template <int N, int SIGN>
+__device__ inline void col_fft(float2 (&v)[N],
+                               const float2* __restrict__ tw_n) {
+  constexpr int T = col_ilog2(N);
+  // first-to-last: radix-4 stages, trailing radix-2 when T is odd.
+  if constexpr (T >= 2) col_stage<N, 4, 4, SIGN>(v, tw_n);
+  if constexpr (T >= 4) col_stage<N, 16, 4, SIGN>(v, tw_n);
+  if constexpr (T >= 6) col_stage<N, 64, 4, SIGN>(v, tw_n);
+  if constexpr (T == 1) col_stage<N, 2, 2, SIGN>(v, tw_n);
+  if constexpr (T == 3) col_stage<N, 8, 2, SIGN>(v, tw_n);
+  if constexpr (T == 5) col_stage<N, 32, 2, SIGN>(v, tw_n);
+}
+
+template <int N, bool TWIDDLE, int SIGN>
+__global__ void __launch_bounds__(256)
+    k_fft_col(const float2* __restrict__ in, float2* __restrict__ out,
+              FftPassDescDev d, unsigned long long n_ffts,
+              const float2* __restrict__ tw_n,
+              const float2* __restrict__ tw_hi,
+              const float2* __restrict__ tw_lo) {
+  const unsigned long long id =
+      (unsigned long long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (id >= n_ffts) return;
+  unsigned long long q0, q1, q2;
+  digits(id, d, q0, q1, q2);
+  const unsigned long long base = q0 * d.in_c0 + q1 * d.in_c1 + q2 * d.in_c2;
+  float2 v[N];
+#pragma unroll
+  for (int i = 0; i < N; ++i)
+    v[col_sigma<N>(i)] = in[base + (unsigned long long)i * d.in_stride];
+  col_fft<N, SIGN>(v, tw_n);
+#pragma unroll
+  for (int k = 0; k < N; ++k) {
+    float2 r = v[k];
+    if constexpr (TWIDDLE) {
+      const unsigned long long m_ =
+          (q0 * d.tw_f0 * (unsigned long long)k) & d.tw_mask;
+      const float2 w = cmulf(tw_hi[m_ >> d.tw_lo_bits],
+                             tw_lo[m_ & ((1ull << d.tw_lo_bits) - 1)]);
+      r = cmulf(r, w);
+    }
+    out[base + (unsigned long long)k * d.in_stride] = r;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Final pass: in-place radix-4 DIF in LDS (single buffer — twice the F of the
+// ping-pong kernel) with the base-4 digit reversal folded into the STORE
+// index and the multi-digit output scatter (validated: dif_r4 prototype).
+// ---------------------------------------------------------------------------
+
+struct DifFinalDescDev {
+  uint32_t n;  // pure 4^t
+  int n_log2, f_log2;
+  unsigned long long in_c2;   // batch row stride (in: base = q0*n + q2*in_c2)
+  unsigned long long out_c2;
+  unsigned long long out_elem_coef;  // out elem k: += k * coef
+  int n_prefix;                      // prefix digits (extraction order)
+  int pf_bits[4];
+  unsigned long long pf_coef[4];
+};
+
+__device__ inline unsigned rev4_bits(unsigned k, int t2 /* log2(n) */) {
+  // base-4 digit reversal = bit reversal then swap adjacent bit pairs
+  unsigned r = __brev(k) >> (32 - t2);
+  return ((r & 0x55555555u) << 1) | ((r & 0xAAAAAAAAu) >> 1);
+}
+
+template <int SIGN>
+__global__ void __launch_bounds__(256)
+    k_fft_dif_final(const float2* __restrict__ in, float2* __restrict__ out,
+                    DifFinalDescDev d, const float2* __restrict__ tw_n) {
+  extern __shared__ float2 lds[];
+  const int n = d.n;
+  const int nl = d.n_log2;
+  const int F = 1 << d.f_log2;
+  const int ldst = n + 2;
+  float2* ltw = lds;
+  float2* X = lds + n;
+  const unsigned long long fft0 = (unsigned long long)blockIdx.x << d.f_log2;
+  const int total = F << nl;
+
+  for (int j = threadIdx.x; j < n; j += blockDim.x) ltw[j] = tw_n[j];
+
+  // ---- contiguous load (register-staged chunks of 8) ----
+  {
+    const int iters = total >> 8;
+    int it = 0;
+    for (; it + 8 <= iters; it += 8) {
+      float2 tmp[8];
+      int lidx[8];
+#pragma unroll
+      for (int k = 0; k < 8; ++k) {
+        const int e = ((it + k) << 8) + threadIdx.x;
+        const int f = e >> nl;
+        const int i = e & (n - 1);
+        // input is fully linear: instance blocks (prefix within row, rows
+        // within batch) are contiguous, so block index == instance id
+        tmp[k] = in[(fft0 + f) * (unsigned long long)n + i];
+        lidx[k] = f * ldst + i;
+      }
+#pragma unroll
+      for (int k = 0; k < 8; ++k) X[lidx[k]] = tmp[k];
+    }
+    for (int e = (it << 8) + threadIdx.x; e < total; e += blockDim.x) {
+      const int f = e >> nl;
+      const int i = e & (n - 1);
+      X[f * ldst + i] = in[(fft0 + f) * (unsigned long long)n + i];
+    }
+  }
+  __syncthreads();
+
+  // ---- in-place radix-4 DIF stages ----
+  const int quarter = n >> 2;
+  for (int L = n, tl = 0; L >= 4; L >>= 2, tl += 2) {
+    const int M = L >> 2;
+    const int m_log2 = (nl - tl) - 2;  // log2(M)
+    for (int b = threadIdx.x; b < F * quarter; b += blockDim.x) {
+      const int f = b >> (nl - 2);
+      const int bb = b & (quarter - 1);
+      // bb -> (block g, j): j = bb & (M-1); g = (bb >> m_log2) * L
+      const int j = bb & (M - 1);
+      const int g = (bb >> m_log2) << (m_log2 + 2);
+      float2* row = X + f * ldst;
+      const float2 a = row[g + j];
+      const float2 bv = row[g + j + M];
+      const float2 c = row[g + j + 2 * M];
+      const float2 dv = row[g + j + 3 * M];
+      // DFT over t then post-twiddle W_L^{m*j} (dif_r4 prototype)
+      const float2 t0 = make_float2(a.x + c.x, a.y + c.y);
+      const float2 t1 = make_float2(a.x - c.x, a.y - c.y);
+      const float2 t2 = make_float2(bv.x + dv.x, bv.y + dv.y);
+      const float2 dmy = make_float2(bv.x - dv.x, bv.y - dv.y);
+      const float2 t3 = (SIGN > 0) ? make_float2(-dmy.y, dmy.x)
+                                   : make_float2(dmy.y, -dmy.x);
+      const float2 x0 = make_float2(t0.x + t2.x, t0.y + t2.y);
+      const float2 x1 = make_float2(t1.x + t3.x, t1.y + t3.y);
+      const float2 x2 = make_float2(t0.x - t2.x, t0.y - t2.y);
+      const float2 x3 = make_float2(t1.x - t3.x, t1.y - t3.y);
+      const int ts = tl;  // W_L^x = ltw[x << tl]
+      row[g + j] = x0;
+      row[g + j + M] = (j == 0) ? x1 : cmulf(x1, ltw[(j << ts)]);
+      row[g + j + 2 * M] = (j == 0) ? x2 : cmulf(x2, ltw[(2 * j) << ts]);
+      row[g + j + 3 * M] = (j == 0) ? x3 : cmulf(x3, ltw[(3 * j) << ts]);
+    }
+    __syncthreads();
+  }
+
+  // ---- scatter store with digit-reversed LDS read ----
+  for (int e = threadIdx.x; e < total; e += blockDim.x) {
+    const int f = e & (F - 1);
+    const int k = e >> d.f_log2;
+    const unsigned long long id = fft0 + f;
+    // id = q2 * n_prefix_total + P handled via pf digits; here id = global
+    // instance: decompose: P = low digits, q2 = rest
+    unsigned long long P = id;
+    unsigned long long obase = 0;
+    int pbits = 0;
+#pragma unroll
+    for (int w = 0; w < 4; ++w) {
+      if (w < d.n_prefix) {
+        const unsigned long long dig = P & ((1ull << d.pf_bits[w]) - 1);
+        obase += dig * d.pf_coef[w];
+        P >>= d.pf_bits[w];
+        pbits += d.pf_bits[w];
+      }
+    }
+    obase += P * d.out_c2;  // remaining digits = batch row
+    const unsigned rk = rev4_bits((unsigned)k, nl);
+    out[obase + (unsigned long long)k * d.out_elem_coef] =
+        X[f * ldst + (int)rk];
+  }
+}
+
 // twiddle-table builder (fp64 on device)
 __global__ void k_build_twiddle(float2* __restrict__ t, size_t count,
                                 double sign_two_pi_over_m) {
@@ -338,6 +606,95 @@ hipError_t fft_stockham_pass(const float2* in, float2* out,
   }
 #undef DISPATCH2
 #undef DISPATCH4
+  SRTB_CHECK_LAUNCH();
+  return hipSuccess;
+}
+
+hipError_t fft_col_pass(const float2* in, float2* out, const FftPassDesc& hd,
+                        size_t n_ffts, int sign, const float2* tw_n,
+                        const float2* tw_hi, const float2* tw_lo,
+                        hipStream_t stream) {
+  FftPassDescDev d;
+  d.n = hd.n;
+  d.n_log2 = ilog2(hd.n);
+  d.f_log2 = 0;
+  d.d0_log2 = hd.d0 ? ilog2(hd.d0) : 48;
+  d.d1_log2 = ilog2(hd.d1 ? hd.d1 : 1);
+  d.in_c0 = hd.in_c0; d.in_c1 = hd.in_c1; d.in_c2 = hd.in_c2;
+  d.in_stride = hd.in_stride;
+  d.out_c0 = hd.out_c0; d.out_c1 = hd.out_c1; d.out_c2 = hd.out_c2;
+  d.out_stride = hd.out_stride;
+  d.tw_f0 = hd.tw_f0; d.tw_f1 = hd.tw_f1;
+  d.tw_mask = hd.tw_mod ? hd.tw_mod - 1 : 0;
+  d.tw_lo_bits = hd.tw_lo_bits;
+  const bool twiddle = hd.tw_mod != 0;
+  const uint32_t grid = (uint32_t)((n_ffts + 255) / 256);
+
+#define COL_DISPATCH(N)                                                      \
+  case N:                                                                    \
+    if (twiddle) {                                                           \
+      if (sign < 0)                                                          \
+        hipLaunchKernelGGL((k_fft_col<N, true, -1>), dim3(grid), dim3(256),  \
+                           0, stream, in, out, d, n_ffts, tw_n, tw_hi,       \
+                           tw_lo);                                           \
+      else                                                                   \
+        hipLaunchKernelGGL((k_fft_col<N, true, 1>), dim3(grid), dim3(256),   \
+                           0, stream, in, out, d, n_ffts, tw_n, tw_hi,       \
+                           tw_lo);                                           \
+    } else {                                                                 \
+      if (sign < 0)                                                          \
+        hipLaunchKernelGGL((k_fft_col<N, false, -1>), dim3(grid), dim3(256), \
+                           0, stream, in, out, d, n_ffts, tw_n, tw_hi,       \
+                           tw_lo);                                           \
+      else                                                                   \
+        hipLaunchKernelGGL((k_fft_col<N, false, 1>), dim3(grid), dim3(256),  \
+                           0, stream, in, out, d, n_ffts, tw_n, tw_hi,       \
+                           tw_lo);                                           \
+    }                                                                        \
+    break;
+  switch (hd.n) {
+    COL_DISPATCH(2)
+    COL_DISPATCH(4)
+    COL_DISPATCH(8)
+    COL_DISPATCH(16)
+    COL_DISPATCH(32)
+    COL_DISPATCH(64)
+    default:
+      return hipErrorInvalidValue;
+  }
+#undef COL_DISPATCH
+  SRTB_CHECK_LAUNCH();
+  return hipSuccess;
+}
+
+hipError_t fft_dif_final(const float2* in, float2* out,
+                         const DifFinalDesc& hd, size_t n_ffts, int F,
+                         int sign, const float2* tw_n, hipStream_t stream) {
+  if (hd.n & (hd.n - 1)) return hipErrorInvalidValue;
+  DifFinalDescDev d;
+  d.n = hd.n;
+  d.n_log2 = ilog2(hd.n);
+  if (d.n_log2 & 1) return hipErrorInvalidValue;  // pure radix-4 lengths only
+  d.f_log2 = ilog2((unsigned)F);
+  d.in_c2 = 0;
+  d.out_c2 = hd.out_c2;
+  d.out_elem_coef = hd.out_elem_coef;
+  d.n_prefix = hd.n_prefix;
+  for (int i = 0; i < 4; ++i) {
+    d.pf_bits[i] = hd.pf_bits[i];
+    d.pf_coef[i] = hd.pf_coef[i];
+  }
+  if (n_ffts % F != 0) return hipErrorInvalidValue;
+  const uint32_t grid = (uint32_t)(n_ffts / F);
+  const size_t lds_bytes =
+      ((size_t)hd.n + (size_t)F * (hd.n + 2)) * sizeof(float2);
+  if (lds_bytes > 160 * 1024) return hipErrorInvalidValue;
+  if (sign < 0)
+    hipLaunchKernelGGL((k_fft_dif_final<-1>), dim3(grid), dim3(256),
+                       lds_bytes, stream, in, out, d, tw_n);
+  else
+    hipLaunchKernelGGL((k_fft_dif_final<1>), dim3(grid), dim3(256), lds_bytes,
+                       stream, in, out, d, tw_n);
   SRTB_CHECK_LAUNCH();
   return hipSuccess;
 }

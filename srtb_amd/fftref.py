@@ -133,6 +133,60 @@ def fft_small_r4(x: np.ndarray, sign: int, dtype=np.complex64) -> np.ndarray:
     return X.astype(dtype)
 
 
+def digit_reverse4(n: int) -> np.ndarray:
+    """Base-4 digit reversal permutation for n = 4^t."""
+    t = 0
+    while 4**t < n:
+        t += 1
+    assert 4**t == n
+    idx = np.arange(n)
+    out = np.zeros(n, dtype=np.int64)
+    for d in range(t):
+        out = out * 4 + (idx >> (2 * d)) % 4
+    return out
+
+
+def fft_small_dit_r4(x: np.ndarray, sign: int, dtype=np.complex64) -> np.ndarray:
+    """Pure radix-4 in-place DIT FFT (n = 4^t): digit-reversed load, then
+    in-place butterflies — the register-resident column kernel's algorithm
+    (all data stays in one array; no ping-pong).
+
+    Stage with output sub-size len (4, 16, ..., n), quarter q = len/4:
+      for each group g (step len), j in [0, q):
+        w = exp(sign*2πi*j/len);  b*=w; c*=w²; d*=w³
+        t0 = a + c; t1 = a - c; t2 = b + d; t3 = si*(b - d)
+        v[g+j]      = t0 + t2
+        v[g+j+q]    = t1 + t3
+        v[g+j+2q]   = t0 - t2
+        v[g+j+3q]   = t1 - t3
+    """
+    n = x.size
+    v = np.asarray(x, dtype=np.complex128)[digit_reverse4(n)].copy()
+    si = sign * 1j
+    ln = 4
+    while ln <= n:
+        q = ln // 4
+        for g in range(0, n, ln):
+            for j in range(q):
+                w1 = np.exp(sign * 2j * np.pi * j / ln)
+                w2 = w1 * w1
+                w3 = w2 * w1
+                a = v[g + j]
+                b = v[g + j + q] * w1
+                c = v[g + j + 2 * q] * w2
+                d = v[g + j + 3 * q] * w3
+                t0 = a + c
+                t1 = a - c
+                t2 = b + d
+                t3 = si * (b - d)
+                v[g + j] = t0 + t2
+                v[g + j + q] = t1 + t3
+                v[g + j + 2 * q] = t0 - t2
+                v[g + j + 3 * q] = t1 - t3
+        ln *= 4
+    return v.astype(dtype)
+
+
 def fft_four_step(x: np.ndarray, l1: int, l2: int, sign: int,
                   fft1=None, fft2=None) -> np.ndarray:
     """Composite FFT of length l1*l2 per the docstring above."""
