@@ -16,6 +16,7 @@
 #include <algorithm>
 #include <cstdint>
 #include <cstdlib>
+#include <cmath>
 #include <stdexcept>
 #include <vector>
 
@@ -83,10 +84,12 @@ class NativeFft {
     }
     std::vector<uint32_t> f;
     {
+      const char* mc = std::getenv("SRTB_FFT_MAXCOL");
+      const int maxcol_log2 = mc ? ilog2z(std::atoi(mc)) : 5;  // default 32
       int rest = t - 8;
-      while (rest > 6) {
-        f.push_back(64);
-        rest -= 6;
+      while (rest > maxcol_log2) {
+        f.push_back(1u << maxcol_log2);
+        rest -= maxcol_log2;
       }
       f.push_back(1u << rest);
       f.push_back(256);
@@ -116,6 +119,7 @@ class NativeFft {
       p.d.tw_f0 = 1;
       p.d.tw_mod = M[j];
       p.d.tw_lo_bits = mod_lo_bits(M[j]);
+      p.d.tw_angle = sign * 2.0 * M_PI / (double)M[j];
       p.n_ffts = batch * (L / f[j]);
       p.tw_n = len_table(f[j]);
       p.tw_hi = mod_hi(M[j]);

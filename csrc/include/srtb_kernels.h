@@ -158,6 +158,7 @@ struct FftPassDesc {
   unsigned long long tw_f0 = 0, tw_f1 = 0;
   unsigned long long tw_mod = 0;  // 0 = no inter-pass twiddle
   int tw_lo_bits = 0;
+  double tw_angle = 0.0;  // sign * 2*pi / tw_mod (for computed twiddles)
 };
 
 // build table[j] = exp(sign * 2*pi*i * j / m), j in [0, count)

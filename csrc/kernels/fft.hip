@@ -52,9 +52,20 @@ struct FftPassDescDev {
   unsigned long long tw_f0, tw_f1;
   unsigned long long tw_mask;  // modulus-1; twiddle enabled via template
   int tw_lo_bits;
+  double tw_angle;  // sign * 2*pi / modulus
 };
 
 namespace {
+
+// inter-pass twiddle e^{i * m * tw_angle}: exact integer reduction (m < M),
+// fp64 angle, fast f32 sincos — table gathers spread over up to 64 cache
+// lines per wave instruction and were the column kernel's bottleneck.
+__device__ inline float2 tw_eval(unsigned long long m, double angle) {
+  const float th = (float)((double)m * angle);
+  float s, c;
+  __sincosf(th, &s, &c);
+  return make_float2(c, s);
+}
 
 __device__ inline void digits(unsigned long long id, const FftPassDescDev& d,
                               unsigned long long& q0, unsigned long long& q1,
@@ -193,9 +204,7 @@ __global__ void __launch_bounds__(256)
     if constexpr (TWIDDLE) {
       const unsigned long long tf = q0 * d.tw_f0 + q1 * d.tw_f1;
       const unsigned long long m_ = (tf * (unsigned long long)k) & d.tw_mask;
-      const float2 w = cmulf(tw_hi[m_ >> d.tw_lo_bits],
-                             tw_lo[m_ & ((1ull << d.tw_lo_bits) - 1)]);
-      v = cmulf(v, w);
+      v = cmulf(v, tw_eval(m_, d.tw_angle));
     }
     const unsigned long long base =
         q0 * d.out_c0 + q1 * d.out_c1 + q2 * d.out_c2;
@@ -334,9 +343,7 @@ __global__ void __launch_bounds__(256)
     if constexpr (TWIDDLE) {
       const unsigned long long m_ =
           (q0 * d.tw_f0 * (unsigned long long)k) & d.tw_mask;
-      const float2 w = cmulf(tw_hi[m_ >> d.tw_lo_bits],
-                             tw_lo[m_ & ((1ull << d.tw_lo_bits) - 1)]);
-      r = cmulf(r, w);
+      r = cmulf(r, tw_eval(m_, d.tw_angle));
     }
     out[base + (unsigned long long)k * d.in_stride] = r;
   }
@@ -576,6 +583,7 @@ hipError_t fft_stockham_pass(const float2* in, float2* out,
   d.tw_f0 = hd.tw_f0; d.tw_f1 = hd.tw_f1;
   d.tw_mask = hd.tw_mod ? hd.tw_mod - 1 : 0;
   d.tw_lo_bits = hd.tw_lo_bits;
+  d.tw_angle = hd.tw_angle;
   const bool twiddle = hd.tw_mod != 0;
   if (n_ffts % F != 0) return hipErrorInvalidValue;
   const uint32_t grid = (uint32_t)(n_ffts / F);
@@ -627,6 +635,7 @@ hipError_t fft_col_pass(const float2* in, float2* out, const FftPassDesc& hd,
   d.tw_f0 = hd.tw_f0; d.tw_f1 = hd.tw_f1;
   d.tw_mask = hd.tw_mod ? hd.tw_mod - 1 : 0;
   d.tw_lo_bits = hd.tw_lo_bits;
+  d.tw_angle = hd.tw_angle;
   const bool twiddle = hd.tw_mod != 0;
   const uint32_t grid = (uint32_t)((n_ffts + 255) / 256);
 
