@@ -137,9 +137,10 @@ class NativeFft {
       p.dif.out_elem_coef = Cp[m - 1];  // prod of all prefix factors
       p.dif.n_prefix = m - 1;
       for (int w = 0; w < m - 1; ++w) {
-        const int src = m - 2 - w;  // extraction order: last col digit first
-        p.dif.pf_bits[w] = ilog2z(f[src]);
-        p.dif.pf_coef[w] = Cp[src];
+        // id low bits = k_0 (output-order digits); stored-prefix coefficient
+        // S_w = prod(f_{w+1} .. f_{m-2}) = M[w+1] / fn
+        p.dif.pf_bits[w] = ilog2z(f[w]);
+        p.dif.pf_coef[w] = M[w + 1] / fn;
       }
       p.n_ffts = batch * (L / fn);
       p.tw_n = len_table(fn);

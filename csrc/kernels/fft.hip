@@ -358,13 +358,34 @@ __global__ void __launch_bounds__(256)
 struct DifFinalDescDev {
   uint32_t n;  // pure 4^t
   int n_log2, f_log2;
-  unsigned long long in_c2;   // batch row stride (in: base = q0*n + q2*in_c2)
-  unsigned long long out_c2;
-  unsigned long long out_elem_coef;  // out elem k: += k * coef
-  int n_prefix;                      // prefix digits (extraction order)
-  int pf_bits[4];
-  unsigned long long pf_coef[4];
+  int j_bits;                        // log2(L/n) = total prefix bits
+  unsigned long long row_len;        // L (batch row stride, elements)
+  unsigned long long out_elem_coef;  // = L/n (output digit coefficient)
+  int n_prefix;                      // prefix digit count (id low bits = k_0)
+  int pf_bits[4];                    // log2 f_w, w = 0..n_prefix-1
+  unsigned long long pf_coef[4];     // S_w = prod(f_{w+1}..f_{m-2})
 };
+
+// instance id = row * 2^j_bits + J, J = sum k_w * prod(f_0..f_{w-1});
+// input block index P = sum k_w * S_w (stored layout); output base = row*L + J
+__device__ inline void dif_addr(unsigned long long id,
+                                const DifFinalDescDev& d,
+                                unsigned long long& in_blk,
+                                unsigned long long& obase) {
+  const unsigned long long J = id & ((1ull << d.j_bits) - 1);
+  const unsigned long long row = id >> d.j_bits;
+  unsigned long long x = J, P = 0;
+#pragma unroll
+  for (int w = 0; w < 4; ++w) {
+    if (w < d.n_prefix) {
+      const unsigned long long dig = x & ((1ull << d.pf_bits[w]) - 1);
+      P += dig * d.pf_coef[w];
+      x >>= d.pf_bits[w];
+    }
+  }
+  in_blk = row * d.row_len + P * d.n;  // element offset of the input block
+  obase = row * d.row_len + J;
+}
 
 __device__ inline unsigned rev4_bits(unsigned k, int t2 /* log2(n) */) {
   // base-4 digit reversal = bit reversal then swap adjacent bit pairs
@@ -400,9 +421,9 @@ __global__ void __launch_bounds__(256)
         const int e = ((it + k) << 8) + threadIdx.x;
         const int f = e >> nl;
         const int i = e & (n - 1);
-        // input is fully linear: instance blocks (prefix within row, rows
-        // within batch) are contiguous, so block index == instance id
-        tmp[k] = in[(fft0 + f) * (unsigned long long)n + i];
+        unsigned long long in_blk, obase;
+        dif_addr(fft0 + f, d, in_blk, obase);
+        tmp[k] = in[in_blk + i];
         lidx[k] = f * ldst + i;
       }
 #pragma unroll
@@ -411,7 +432,9 @@ __global__ void __launch_bounds__(256)
     for (int e = (it << 8) + threadIdx.x; e < total; e += blockDim.x) {
       const int f = e >> nl;
       const int i = e & (n - 1);
-      X[f * ldst + i] = in[(fft0 + f) * (unsigned long long)n + i];
+      unsigned long long in_blk, obase;
+      dif_addr(fft0 + f, d, in_blk, obase);
+      X[f * ldst + i] = in[in_blk + i];
     }
   }
   __syncthreads();
@@ -452,26 +475,12 @@ __global__ void __launch_bounds__(256)
     __syncthreads();
   }
 
-  // ---- scatter store with digit-reversed LDS read ----
+  // ---- store: digit-reversed LDS read, contiguous-lane output runs ----
   for (int e = threadIdx.x; e < total; e += blockDim.x) {
     const int f = e & (F - 1);
     const int k = e >> d.f_log2;
-    const unsigned long long id = fft0 + f;
-    // id = q2 * n_prefix_total + P handled via pf digits; here id = global
-    // instance: decompose: P = low digits, q2 = rest
-    unsigned long long P = id;
-    unsigned long long obase = 0;
-    int pbits = 0;
-#pragma unroll
-    for (int w = 0; w < 4; ++w) {
-      if (w < d.n_prefix) {
-        const unsigned long long dig = P & ((1ull << d.pf_bits[w]) - 1);
-        obase += dig * d.pf_coef[w];
-        P >>= d.pf_bits[w];
-        pbits += d.pf_bits[w];
-      }
-    }
-    obase += P * d.out_c2;  // remaining digits = batch row
+    unsigned long long in_blk, obase;
+    dif_addr(fft0 + f, d, in_blk, obase);
     const unsigned rk = rev4_bits((unsigned)k, nl);
     out[obase + (unsigned long long)k * d.out_elem_coef] =
         X[f * ldst + (int)rk];
@@ -685,14 +694,16 @@ hipError_t fft_dif_final(const float2* in, float2* out,
   d.n_log2 = ilog2(hd.n);
   if (d.n_log2 & 1) return hipErrorInvalidValue;  // pure radix-4 lengths only
   d.f_log2 = ilog2((unsigned)F);
-  d.in_c2 = 0;
-  d.out_c2 = hd.out_c2;
+  d.row_len = hd.out_c2;  // L
   d.out_elem_coef = hd.out_elem_coef;
   d.n_prefix = hd.n_prefix;
+  int jb = 0;
   for (int i = 0; i < 4; ++i) {
     d.pf_bits[i] = hd.pf_bits[i];
     d.pf_coef[i] = hd.pf_coef[i];
+    if (i < hd.n_prefix) jb += hd.pf_bits[i];
   }
+  d.j_bits = jb;
   if (n_ffts % F != 0) return hipErrorInvalidValue;
   const uint32_t grid = (uint32_t)(n_ffts / F);
   const size_t lds_bytes =
