@@ -85,8 +85,12 @@ class NativeFft {
     std::vector<uint32_t> f;
     {
       const char* mc = std::getenv("SRTB_FFT_MAXCOL");
-      const int maxcol_log2 = mc ? ilog2z(std::atoi(mc)) : 5;  // default 32
+      int maxcol_log2 = mc ? ilog2z(std::atoi(mc)) : 5;  // default 32
       int rest = t - 8;
+      // at most 4 column passes fit the scatter descriptor: widen if needed
+      while ((rest + maxcol_log2 - 1) / maxcol_log2 > 4) ++maxcol_log2;
+      if (maxcol_log2 > 6)
+        throw std::runtime_error("NativeFft: length too large");
       while (rest > maxcol_log2) {
         f.push_back(1u << maxcol_log2);
         rest -= maxcol_log2;
