@@ -85,17 +85,19 @@ class NativeFft {
     std::vector<uint32_t> f;
     {
       const char* mc = std::getenv("SRTB_FFT_MAXCOL");
-      int maxcol_log2 = mc ? ilog2z(std::atoi(mc)) : 5;  // default 32
-      int rest = t - 8;
-      // at most 4 column passes fit the scatter descriptor: widen if needed
-      while ((rest + maxcol_log2 - 1) / maxcol_log2 > 4) ++maxcol_log2;
-      if (maxcol_log2 > 6)
-        throw std::runtime_error("NativeFft: length too large");
-      while (rest > maxcol_log2) {
-        f.push_back(1u << maxcol_log2);
-        rest -= maxcol_log2;
+      const int maxcol_log2 = mc ? ilog2z(std::atoi(mc)) : 5;  // default 32
+      const int rest = t - 8;
+      // at most 4 column passes fit the scatter descriptor; distribute the
+      // bits evenly so the wide (register-hungry) factors are as small as
+      // possible: e.g. 21 bits -> [64,32,32,32], 10 -> [32,32]
+      int ncols = (rest + maxcol_log2 - 1) / maxcol_log2;
+      if (ncols > 4) throw std::runtime_error("NativeFft: length too large");
+      const int base_b = rest / ncols, extra = rest % ncols;
+      for (int i = 0; i < ncols; ++i) {
+        const int b = base_b + (i < extra ? 1 : 0);
+        if (b > 6) throw std::runtime_error("NativeFft: factor too large");
+        f.push_back(1u << b);
       }
-      f.push_back(1u << rest);
       f.push_back(256);
     }
     const int m = (int)f.size();
