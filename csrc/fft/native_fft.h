@@ -91,7 +91,7 @@ class NativeFft {
       // bits evenly so the wide (register-hungry) factors are as small as
       // possible: e.g. 21 bits -> [64,32,32,32], 10 -> [32,32]
       int ncols = (rest + maxcol_log2 - 1) / maxcol_log2;
-      if (ncols > 4) throw std::runtime_error("NativeFft: length too large");
+      if (ncols > 4) ncols = 4;  // widen factors instead (up to 64 each)
       const int base_b = rest / ncols, extra = rest % ncols;
       for (int i = 0; i < ncols; ++i) {
         const int b = base_b + (i < extra ? 1 : 0);
