@@ -44,6 +44,8 @@ def main():
                     help="cache dedispersion phase factors (fixed DM); "
                     "default recomputes fp64 phase per block like the reference")
     ap.add_argument("--no-rfi", action="store_true")
+    ap.add_argument("--fft", choices=["native", "hipfft"], default="native",
+                    help="FFT backend (hand-written Stockham vs hipFFT)")
     args = ap.parse_args()
 
     import torch
@@ -86,7 +88,8 @@ def main():
         rfi_threshold=rfi_threshold, sk_threshold=sk_threshold,
         snr_threshold=snr, max_boxcar=max_boxcar, nsamps_reserved=0,
         zap_ranges=zap_ranges, use_phase_table=args.phase_table,
-        enable_rfi_s1=not args.no_rfi, enable_sk=True, n_slots=2)
+        enable_rfi_s1=not args.no_rfi, enable_sk=True, n_slots=2,
+        fft_backend=0 if args.fft == "native" else 1)
 
     # synthetic 2-bit baseband noise, pinned, one buffer per slot
     rng = np.random.default_rng(1234 + rank)
@@ -169,6 +172,7 @@ def main():
                 "sample_rate": sample_rate,
                 "rfi": (not args.no_rfi),
                 "phase_table": bool(args.phase_table),
+                "fft_backend": args.fft,
                 "real_time_ratio_per_gpu": round(real_time_ratio, 1),
                 "parallelism": f"stream-sharded dp{n_gpus}",
                 "global_batch": n_gpus,

@@ -76,7 +76,17 @@ PipelineEngine::PipelineEngine(const EngineConfig& cfg, int n_slots)
               "pinned cnt");
     check_hip(hipHostMalloc(&s.h_thresholds, (1 + n_boxcars_) * sizeof(float)),
               "pinned thr");
-    s.plans.create(n_, l_, s_, s.stream);
+    // hand-written FFT when both shapes are pow2 (fwd: packed C2C of n/2;
+    // bwd: batched C2C of l_); hipFFT otherwise
+    native_fft_ = (cfg.fft_backend == 0) && NativeFft::supported(nc_) &&
+                  NativeFft::supported(l_);
+    if (native_fft_) {
+      s.nfwd.plan(nc_, 1, -1, s.stream);
+      s.nbwd.plan(l_, s_, +1, s.stream);
+      check_hip(hipStreamSynchronize(s.stream), "fft table sync");
+    } else {
+      s.plans.create(n_, l_, s_, s.stream);
+    }
   }
 
   if (cfg.use_phase_table) {
@@ -132,24 +142,51 @@ void PipelineEngine::enqueue_chain(Slot& s, const uint8_t* dev_raw,
     check_hip(unpack(dev_raw, s.samples, n_, cfg_.baseband_input_bits, nullptr,
                      st),
               "unpack");
+  } else if (native_fft_) {
+    // native fwd runs column passes in place on its input: copy the caller's
+    // samples into the slot buffer first (D2D, overlapped on stream)
+    check_hip(hipMemcpyAsync(s.samples, dev_samples, n_ * sizeof(float),
+                             hipMemcpyDeviceToDevice, st),
+              "samples d2d");
   } else {
     fft_in = dev_samples;
   }
-  // 2. R2C forward (out-of-place; Nyquist bin written but ignored: the
-  //    downstream count is Nc — reference drops it, fft_pipe.hpp:77)
-  s.plans.exec_r2c(const_cast<float*>(fft_in), s.spec);
-  // 3. mean |X|^2 over Nc
-  if (cfg_.enable_rfi_s1)
-    check_hip(mean_power(s.spec, nc_, s.partials, s.mean_power, st), "meanp");
+  if (native_fft_) {
+    // 2. forward C2C of the packed-real view + r2c post-process with FUSED
+    //    mean-|X|^2 (saves the separate 4 GB mean_power pass)
+    s.nfwd.exec(reinterpret_cast<float2*>(s.samples), s.spec, st);
+    check_hip(r2c_post_process(
+                  s.spec, s.spec, nc_,
+                  cfg_.enable_rfi_s1 ? s.partials : nullptr,
+                  cfg_.enable_rfi_s1 ? s.mean_power : nullptr, st),
+              "r2c post");
+  } else {
+    // 2. R2C forward (out-of-place; Nyquist bin written but ignored: the
+    //    downstream count is Nc — reference drops it, fft_pipe.hpp:77)
+    s.plans.exec_r2c(const_cast<float*>(fft_in), s.spec);
+    // 3. mean |X|^2 over Nc
+    if (cfg_.enable_rfi_s1)
+      check_hip(mean_power(s.spec, nc_, s.partials, s.mean_power, st),
+                "meanp");
+  }
   // 4. fused RFI s1 + manual zap + dedispersion (single pass over 8·Nc bytes)
   check_hip(rfi_dedisperse_fused(
                 s.spec, nc_, cfg_.enable_rfi_s1 ? s.mean_power : nullptr,
                 cfg_.rfi_threshold, norm_coeff_, cfg_.zap_ranges,
                 cfg_.n_zap_ranges, f_min_, f_c_, df_, dm, table, st),
             "rfi+dedisp");
-  // 5. waterfall: batched backward C2C in place → [S][L]
-  s.plans.exec_c2c_backward(s.spec);
-  float2* wf = s.spec;
+  // 5. waterfall: batched backward C2C → [S][L]
+  float2* wf;
+  if (native_fft_) {
+    // column passes in place on spec; final DIF pass scatters into the
+    // samples buffer (same 4 GB footprint, no longer needed this block)
+    s.nbwd.exec(s.spec, reinterpret_cast<float2*>(s.samples), st);
+    wf = reinterpret_cast<float2*>(s.samples);
+  } else {
+    s.plans.exec_c2c_backward(s.spec);
+    wf = s.spec;
+  }
+  s.wf = wf;
 
   const int ncnt = 2 + n_boxcars_;
   check_hip(hipMemsetAsync(s.counters, 0, ncnt * sizeof(unsigned), st),
@@ -269,7 +306,10 @@ void PipelineEngine::synchronize() {
   }
 }
 
-float2* PipelineEngine::waterfall_ptr(int slot) { return slots_.at(slot)->spec; }
+float2* PipelineEngine::waterfall_ptr(int slot) {
+  Slot& s = *slots_.at(slot);
+  return s.wf ? s.wf : s.spec;
+}
 float* PipelineEngine::time_series_ptr(int slot) { return slots_.at(slot)->ts; }
 float* PipelineEngine::cumsum_ptr(int slot) { return slots_.at(slot)->cumsum; }
 hipStream_t PipelineEngine::stream(int slot) { return slots_.at(slot)->stream; }
