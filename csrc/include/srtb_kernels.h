@@ -232,6 +232,10 @@ hipError_t running_mean(const float* data, size_t nsamp, size_t nchan,
                         uint8_t* out, size_t windowsize, float* ave,
                         hipStream_t stream);
 
+// mag[i] = |x[i]| for complex input (correlator backward-FFT epilogue)
+hipError_t complex_abs(const float2* x, float* mag, size_t n,
+                       hipStream_t stream);
+
 // correlator pointwise: corr[i] = scale * f1[i]*conj(f2[i]); mag[i] = |corr[i]|
 // (reference src/correlator.cpp:116-140; mag may be null).
 hipError_t correlate_pointwise(const float2* f1, const float2* f2,

@@ -107,6 +107,14 @@ __global__ void k_running_mean(const float* __restrict__ data, size_t nsamp,
   ave[j] = a;
 }
 
+__global__ void k_complex_abs(const float2* __restrict__ x,
+                              float* __restrict__ mag, size_t n) {
+  const size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride)
+    mag[i] = sqrtf(norm2(x[i]));
+}
+
 __global__ void k_correlate(const float2* __restrict__ f1,
                             const float2* __restrict__ f2,
                             float2* __restrict__ corr, float* __restrict__ mag,
@@ -167,6 +175,14 @@ hipError_t running_mean(const float* data, size_t nsamp, size_t nchan,
                         hipStream_t stream) {
   hipLaunchKernelGGL(k_running_mean, grid_for(nchan), dim3(kBlock), 0, stream,
                      data, nsamp, nchan, out, windowsize, ave);
+  SRTB_CHECK_LAUNCH();
+  return hipSuccess;
+}
+
+hipError_t complex_abs(const float2* x, float* mag, size_t n,
+                       hipStream_t stream) {
+  hipLaunchKernelGGL(k_complex_abs, grid_for(n), dim3(kBlock), 0, stream, x,
+                     mag, n);
   SRTB_CHECK_LAUNCH();
   return hipSuccess;
 }

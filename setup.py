@@ -64,8 +64,39 @@ def build_hip_objects():
     return objects
 
 
+APP_BINARIES = {
+    "srtb-backend": "csrc/app/srtb_backend.cpp",
+    "srtb-correlator": "csrc/app/correlator.cpp",
+    "srtb-baseband-receiver": "csrc/app/baseband_receiver.cpp",
+}
+
+
+def build_apps(objects):
+    """Link the native executables (reference src/*.cpp equivalents)."""
+    bindir = os.path.join(ROOT, "bin")
+    os.makedirs(bindir, exist_ok=True)
+    for name, src in APP_BINARIES.items():
+        out = os.path.join(bindir, name)
+        src_path = os.path.join(ROOT, src)
+        deps = [src_path] + objects + [
+            os.path.join(ROOT, p) for p in (
+                "csrc/app/config.h", "csrc/app/runtime.h",
+                "csrc/app/udp_receiver.h", "csrc/app/writers.h",
+                "csrc/engine/engine.h", "csrc/fft/native_fft.h")]
+        if os.path.exists(out) and all(
+                os.path.getmtime(out) >= os.path.getmtime(d) for d in deps
+                if os.path.exists(d)):
+            continue
+        cmd = [HIPCC, *HIPCC_FLAGS, "-x", "hip", src_path, "-x", "none", *objects,
+               f"-I{ROOT}/csrc/include", f"-L{ROCM}/lib", "-lhipfft",
+               "-o", out]
+        print("+", " ".join(cmd), flush=True)
+        subprocess.check_call(cmd)
+
+
 def main():
     objects = build_hip_objects()
+    build_apps(objects)
 
     from torch.utils.cpp_extension import BuildExtension, CUDAExtension
 
