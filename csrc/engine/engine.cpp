@@ -169,22 +169,38 @@ void PipelineEngine::enqueue_chain(Slot& s, const uint8_t* dev_raw,
       check_hip(mean_power(s.spec, nc_, s.partials, s.mean_power, st),
                 "meanp");
   }
-  // 4. fused RFI s1 + manual zap + dedispersion (single pass over 8·Nc bytes)
-  check_hip(rfi_dedisperse_fused(
-                s.spec, nc_, cfg_.enable_rfi_s1 ? s.mean_power : nullptr,
-                cfg_.rfi_threshold, norm_coeff_, cfg_.zap_ranges,
-                cfg_.n_zap_ranges, f_min_, f_c_, df_, dm, table, st),
-            "rfi+dedisp");
-  // 5. waterfall: batched backward C2C → [S][L]
+  // 4.+5. RFI s1 + manual zap + dedispersion fused into the backward FFT's
+  // first column pass when the native planner allows it (saves a full
+  // read+write of the 4 GB spectrum); otherwise the standalone fused kernel.
   float2* wf;
-  if (native_fft_) {
-    // column passes in place on spec; final DIF pass scatters into the
-    // samples buffer (same 4 GB footprint, no longer needed this block)
-    s.nbwd.exec(s.spec, reinterpret_cast<float2*>(s.samples), st);
+  const bool fuse_into_bwd =
+      native_fft_ && !table && s.nbwd.first_pass_fusable();
+  if (fuse_into_bwd) {
+    FftPreop pre;
+    pre.mean_power = cfg_.enable_rfi_s1 ? s.mean_power : nullptr;
+    pre.threshold = cfg_.rfi_threshold;
+    pre.norm_coeff = norm_coeff_;
+    pre.n_zap = cfg_.n_zap_ranges;
+    for (int i = 0; i < cfg_.n_zap_ranges; ++i) pre.zap[i] = cfg_.zap_ranges[i];
+    pre.f_min = f_min_;
+    pre.f_c = f_c_;
+    pre.df = df_;
+    pre.dm = dm;
+    s.nbwd.exec(s.spec, reinterpret_cast<float2*>(s.samples), st, &pre);
     wf = reinterpret_cast<float2*>(s.samples);
   } else {
-    s.plans.exec_c2c_backward(s.spec);
-    wf = s.spec;
+    check_hip(rfi_dedisperse_fused(
+                  s.spec, nc_, cfg_.enable_rfi_s1 ? s.mean_power : nullptr,
+                  cfg_.rfi_threshold, norm_coeff_, cfg_.zap_ranges,
+                  cfg_.n_zap_ranges, f_min_, f_c_, df_, dm, table, st),
+              "rfi+dedisp");
+    if (native_fft_) {
+      s.nbwd.exec(s.spec, reinterpret_cast<float2*>(s.samples), st);
+      wf = reinterpret_cast<float2*>(s.samples);
+    } else {
+      s.plans.exec_c2c_backward(s.spec);
+      wf = s.spec;
+    }
   }
   s.wf = wf;
 

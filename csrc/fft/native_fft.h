@@ -156,16 +156,28 @@ class NativeFft {
 
   int n_passes() const { return (int)passes_.size(); }
 
+  // The engine can fuse the RFI+dedispersion elementwise stage into the
+  // first pass when that pass is a register column pass.
+  bool first_pass_fusable() const {
+    return !passes_.empty() && passes_[0].kind == PassKind::kCol;
+  }
+
   // Execute the planned transform.  out may equal in only for 1-pass plans.
-  void exec(const float2* in, float2* out, hipStream_t stream) {
+  // preop (optional) is applied to the FIRST pass's loads (requires
+  // first_pass_fusable()).
+  void exec(const float2* in, float2* out, hipStream_t stream,
+            const FftPreop* preop = nullptr) {
     if (passes_.empty()) throw std::runtime_error("NativeFft: not planned");
     if (passes_.size() > 1 && in == out)
       throw std::runtime_error("NativeFft: multi-pass needs out != in");
+    if (preop && !first_pass_fusable())
+      throw std::runtime_error("NativeFft: preop needs a column first pass");
     float2* cur = const_cast<float2*>(in);
     for (size_t i = 0; i < passes_.size(); ++i) {
       Pass& p = passes_[i];
       const bool last = (i + 1 == passes_.size());
       float2* dst = last ? out : cur;
+      const FftPreop* pre = (i == 0) ? preop : nullptr;
       switch (p.kind) {
         case PassKind::kStockham: {
           const int F = pick_f(p.d.n, p, p.n_ffts);
@@ -177,7 +189,7 @@ class NativeFft {
         }
         case PassKind::kCol:
           check_hip(fft_col_pass(cur, cur, p.d, p.n_ffts, sign_, p.tw_n,
-                                 p.tw_hi, p.tw_lo, stream),
+                                 p.tw_hi, p.tw_lo, stream, pre),
                     "fft_col_pass");
           dst = cur;  // in place
           break;

@@ -172,13 +172,26 @@ hipError_t fft_stockham_pass(const float2* in, float2* out,
                              const float2* tw_n, const float2* tw_hi,
                              const float2* tw_lo, hipStream_t stream);
 
+// Optional per-element pre-op fused into a column pass's LOADS: the full
+// RFI-s1 zap + normalize + manual zap + coherent dedispersion of
+// rfi_dedisperse_fused, applied at the element's flat spectrum index
+// (= its load address offset).  Fusing it into the backward FFT's first
+// pass saves a whole read+write of the 4 GB spectrum per block.
+struct FftPreop {
+  const double* mean_power = nullptr;  // null = no RFI zap/normalize
+  float threshold = 0.f, norm_coeff = 1.f;
+  int n_zap = 0;
+  ZapRange zap[16] = {};
+  double f_min = 0, f_c = 0, df = 0, dm = 0;
+};
+
 // Register-resident column FFT pass (N in {2,4,8,16,32,64}; one FFT per
 // thread fully in VGPRs; in-place: out must alias layout of in addressing;
 // uses the same FftPassDesc fields; out_* ignored, stores to input layout).
 hipError_t fft_col_pass(const float2* in, float2* out, const FftPassDesc& d,
                         size_t n_ffts, int sign, const float2* tw_n,
                         const float2* tw_hi, const float2* tw_lo,
-                        hipStream_t stream);
+                        hipStream_t stream, const FftPreop* preop = nullptr);
 
 // Final composite pass: in-place radix-4 DIF in LDS with base-4
 // digit-reversal folded into the store, plus multi-digit output scatter.

@@ -57,4 +57,23 @@ __device__ inline T block_reduce_sum(T v) {
 
 __device__ inline float norm2(float2 c) { return c.x * c.x + c.y * c.y; }
 
+// Coherent-dedispersion phase factor (reference phase_factor_v3,
+// coherent_dedispersion.hpp:133-150): k = D*1e6*dm/f*((f-f_c)/f_c)^2 with
+// |k| up to ~1e9, so everything before the wrap stays in fp64.
+constexpr double kDispersionConstantMHz = 4.148808e3;
+
+__device__ inline float2 srtb_dedisp_factor(size_t i, double f_min,
+                                            double f_c, double df,
+                                            double dm) {
+  const double f = f_min + df * (double)i;
+  const double r = (f - f_c) / f_c;
+  const double k = (kDispersionConstantMHz * 1e6) * dm / f * (r * r);
+  double k_int;
+  const double k_frac = modf(k, &k_int);
+  const double phi = -2.0 * M_PI * k_frac;
+  double s, c;
+  sincos(phi, &s, &c);
+  return make_float2((float)c, (float)s);
+}
+
 }  // namespace srtb_hip

@@ -319,23 +319,56 @@ __device__ inline void col_fft(float2 (&v)[N],
   if constexpr (T == 5) col_stage<N, 32, 2, SIGN>(v, tw_n);
 }
 
-template <int N, bool TWIDDLE, int SIGN>
+struct FftPreopDev {
+  const double* mean_power;
+  float threshold, norm_coeff;
+  int n_zap;
+  ZapRange zap[16];
+  double f_min, f_c, df, dm;
+};
+
+template <int N, bool TWIDDLE, int SIGN, bool PREOP>
 __global__ void __launch_bounds__(256)
     k_fft_col(const float2* __restrict__ in, float2* __restrict__ out,
               FftPassDescDev d, unsigned long long n_ffts,
               const float2* __restrict__ tw_n,
               const float2* __restrict__ tw_hi,
-              const float2* __restrict__ tw_lo) {
+              const float2* __restrict__ tw_lo, FftPreopDev pre) {
   const unsigned long long id =
       (unsigned long long)blockIdx.x * blockDim.x + threadIdx.x;
   if (id >= n_ffts) return;
   unsigned long long q0, q1, q2;
   digits(id, d, q0, q1, q2);
   const unsigned long long base = q0 * d.in_c0 + q1 * d.in_c1 + q2 * d.in_c2;
+  float thr_mean = 0.f;
+  if constexpr (PREOP) {
+    if (pre.mean_power) thr_mean = pre.threshold * (float)(*pre.mean_power);
+  }
   float2 v[N];
 #pragma unroll
-  for (int i = 0; i < N; ++i)
-    v[col_sigma<N>(i)] = in[base + (unsigned long long)i * d.in_stride];
+  for (int i = 0; i < N; ++i) {
+    const unsigned long long flat =
+        base + (unsigned long long)i * d.in_stride;
+    float2 x = in[flat];
+    if constexpr (PREOP) {
+      // fused rfi_dedisperse (spectrum.hip k_rfi_dedisp_fused semantics);
+      // `flat` IS the spectrum bin index for in-place column passes
+      bool zap = pre.mean_power && (norm2(x) > thr_mean);
+      for (int z = 0; z < pre.n_zap; ++z)
+        zap |= (flat >= pre.zap[z].lo) & (flat <= pre.zap[z].hi);
+      if (zap) {
+        x = make_float2(0.f, 0.f);
+      } else {
+        if (pre.mean_power) {
+          x.x *= pre.norm_coeff;
+          x.y *= pre.norm_coeff;
+        }
+        x = cmulf(x, srtb_dedisp_factor(flat, pre.f_min, pre.f_c, pre.df,
+                                        pre.dm));
+      }
+    }
+    v[col_sigma<N>(i)] = x;
+  }
   col_fft<N, SIGN>(v, tw_n);
 #pragma unroll
   for (int k = 0; k < N; ++k) {
@@ -630,7 +663,19 @@ hipError_t fft_stockham_pass(const float2* in, float2* out,
 hipError_t fft_col_pass(const float2* in, float2* out, const FftPassDesc& hd,
                         size_t n_ffts, int sign, const float2* tw_n,
                         const float2* tw_hi, const float2* tw_lo,
-                        hipStream_t stream) {
+                        hipStream_t stream, const FftPreop* preop) {
+  FftPreopDev pre{};
+  if (preop) {
+    pre.mean_power = preop->mean_power;
+    pre.threshold = preop->threshold;
+    pre.norm_coeff = preop->norm_coeff;
+    pre.n_zap = preop->n_zap;
+    for (int i = 0; i < preop->n_zap; ++i) pre.zap[i] = preop->zap[i];
+    pre.f_min = preop->f_min;
+    pre.f_c = preop->f_c;
+    pre.df = preop->df;
+    pre.dm = preop->dm;
+  }
   FftPassDescDev d;
   d.n = hd.n;
   d.n_log2 = ilog2(hd.n);
@@ -648,26 +693,21 @@ hipError_t fft_col_pass(const float2* in, float2* out, const FftPassDesc& hd,
   const bool twiddle = hd.tw_mod != 0;
   const uint32_t grid = (uint32_t)((n_ffts + 255) / 256);
 
+#define COL_LAUNCH(N, TW, SG, PR)                                           \
+  hipLaunchKernelGGL((k_fft_col<N, TW, SG, PR>), dim3(grid), dim3(256), 0,   \
+                     stream, in, out, d, n_ffts, tw_n, tw_hi, tw_lo, pre)
 #define COL_DISPATCH(N)                                                      \
   case N:                                                                    \
     if (twiddle) {                                                           \
-      if (sign < 0)                                                          \
-        hipLaunchKernelGGL((k_fft_col<N, true, -1>), dim3(grid), dim3(256),  \
-                           0, stream, in, out, d, n_ffts, tw_n, tw_hi,       \
-                           tw_lo);                                           \
-      else                                                                   \
-        hipLaunchKernelGGL((k_fft_col<N, true, 1>), dim3(grid), dim3(256),   \
-                           0, stream, in, out, d, n_ffts, tw_n, tw_hi,       \
-                           tw_lo);                                           \
+      if (sign < 0) { if (preop) COL_LAUNCH(N, true, -1, true);              \
+                      else COL_LAUNCH(N, true, -1, false); }                 \
+      else          { if (preop) COL_LAUNCH(N, true, 1, true);               \
+                      else COL_LAUNCH(N, true, 1, false); }                  \
     } else {                                                                 \
-      if (sign < 0)                                                          \
-        hipLaunchKernelGGL((k_fft_col<N, false, -1>), dim3(grid), dim3(256), \
-                           0, stream, in, out, d, n_ffts, tw_n, tw_hi,       \
-                           tw_lo);                                           \
-      else                                                                   \
-        hipLaunchKernelGGL((k_fft_col<N, false, 1>), dim3(grid), dim3(256),  \
-                           0, stream, in, out, d, n_ffts, tw_n, tw_hi,       \
-                           tw_lo);                                           \
+      if (sign < 0) { if (preop) COL_LAUNCH(N, false, -1, true);             \
+                      else COL_LAUNCH(N, false, -1, false); }                \
+      else          { if (preop) COL_LAUNCH(N, false, 1, true);              \
+                      else COL_LAUNCH(N, false, 1, false); }                 \
     }                                                                        \
     break;
   switch (hd.n) {
@@ -681,6 +721,7 @@ hipError_t fft_col_pass(const float2* in, float2* out, const FftPassDesc& hd,
       return hipErrorInvalidValue;
   }
 #undef COL_DISPATCH
+#undef COL_LAUNCH
   SRTB_CHECK_LAUNCH();
   return hipSuccess;
 }

@@ -24,7 +24,6 @@ namespace srtb_hip {
 namespace {
 
 constexpr int kReducePartials = 1024;
-constexpr double kDispersionConstant = 4.148808e3;  // MHz^2 pc^-1 cm^3 s
 
 // ---------------- two-pass reductions ----------------
 
@@ -108,18 +107,7 @@ __global__ void k_reduce_partials_2(const double* __restrict__ partials,
 
 __device__ inline float2 dedisp_factor(size_t i, double f_min, double f_c,
                                        double df, double dm) {
-  // k = D*1e6 * dm / f * ((f-f_c)/f_c)^2 ; factor = exp(-2*pi*i*frac(k))
-  // (reference phase_factor_v3, coherent_dedispersion.hpp:133-150; |k| can
-  // reach ~1e9 so everything up to the wrap stays in fp64)
-  const double f = f_min + df * (double)i;
-  const double r = (f - f_c) / f_c;
-  const double k = (kDispersionConstant * 1e6) * dm / f * (r * r);
-  double k_int;
-  const double k_frac = modf(k, &k_int);
-  const double phi = -2.0 * M_PI * k_frac;
-  double s, c;
-  sincos(phi, &s, &c);
-  return make_float2((float)c, (float)s);
+  return srtb_dedisp_factor(i, f_min, f_c, df, dm);
 }
 
 __device__ inline float2 cmul(float2 a, float2 b) {

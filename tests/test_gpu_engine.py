@@ -50,8 +50,14 @@ def make_engine(C, cfg, **kw):
     return C.PipelineEngine(**args)
 
 
-def test_engine_matches_cpu_oracle(C):
+@pytest.mark.parametrize("n,s", [(1 << 18, 1 << 6), (1 << 20, 1 << 6)])
+def test_engine_matches_cpu_oracle(C, n, s):
+    """Second shape has waterfall len 2^13 > 4096 → exercises the multi-pass
+    native FFT with the RFI+dedispersion preop fused into its first column
+    pass."""
     cfg = small_cfg()
+    cfg.baseband_input_count = n
+    cfg.spectrum_channel_count = s
     # enable real thresholds so RFI/SK paths execute
     cfg.mitigate_rfi_average_method_threshold = 20.0
     cfg.mitigate_rfi_spectral_kurtosis_threshold = 1.5
