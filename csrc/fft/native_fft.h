@@ -86,7 +86,13 @@ class NativeFft {
     {
       const char* mc = std::getenv("SRTB_FFT_MAXCOL");
       const int maxcol_log2 = mc ? ilog2z(std::atoi(mc)) : 5;  // default 32
-      const int rest = t - 8;
+      const char* fe = std::getenv("SRTB_FFT_FINAL");
+      int final_log2 = fe ? ilog2z(std::atoi(fe)) : 8;  // 256/1024/4096
+      if (final_log2 & 1) ++final_log2;                 // pure 4^t only
+      if (final_log2 < 8) final_log2 = 8;
+      if (final_log2 > 12) final_log2 = 12;
+      if (final_log2 >= t) final_log2 = (t % 2) ? t - 1 : t - 2;
+      const int rest = t - final_log2;
       // at most 4 column passes fit the scatter descriptor; distribute the
       // bits evenly so the wide (register-hungry) factors are as small as
       // possible: e.g. 21 bits -> [64,32,32,32], 10 -> [32,32]
@@ -98,7 +104,7 @@ class NativeFft {
         if (b > 6) throw std::runtime_error("NativeFft: factor too large");
         f.push_back(1u << b);
       }
-      f.push_back(256);
+      f.push_back(1u << final_log2);
     }
     const int m = (int)f.size();
     if (m - 1 > 4) throw std::runtime_error("NativeFft: too many factors");
@@ -195,6 +201,10 @@ class NativeFft {
           break;
         case PassKind::kDif: {
           int F = 32;
+          while (F > 1 &&
+                 ((size_t)p.dif.n + (size_t)F * (p.dif.n + 2)) *
+                         sizeof(float2) > 160 * 1024)
+            F >>= 1;
           while (F > 1 && p.n_ffts % F != 0) F >>= 1;
           check_hip(fft_dif_final(cur, dst, p.dif, p.n_ffts, F, sign_,
                                   p.tw_n, stream),
