@@ -46,6 +46,8 @@ def main():
     ap.add_argument("--no-rfi", action="store_true")
     ap.add_argument("--fft", choices=["native", "hipfft"], default="native",
                     help="FFT backend (hand-written Stockham vs hipFFT)")
+    ap.add_argument("--slots", type=int, default=2,
+                    help="double-buffered engine slots (streams)")
     args = ap.parse_args()
 
     import torch
@@ -88,14 +90,14 @@ def main():
         rfi_threshold=rfi_threshold, sk_threshold=sk_threshold,
         snr_threshold=snr, max_boxcar=max_boxcar, nsamps_reserved=0,
         zap_ranges=zap_ranges, use_phase_table=args.phase_table,
-        enable_rfi_s1=not args.no_rfi, enable_sk=True, n_slots=2,
+        enable_rfi_s1=not args.no_rfi, enable_sk=True, n_slots=args.slots,
         fft_backend=0 if args.fft == "native" else 1)
 
     # synthetic 2-bit baseband noise, pinned, one buffer per slot
     rng = np.random.default_rng(1234 + rank)
     raw_bytes = eng.raw_bytes
     pinned = []
-    for _ in range(2):
+    for _ in range(max(args.slots, 2)):
         t = torch.from_numpy(rng.integers(0, 256, raw_bytes, dtype=np.uint8))
         pinned.append(t.pin_memory())
 
@@ -111,7 +113,7 @@ def main():
     def run_steps(k):
         prev = None
         for i in range(k):
-            slot = eng.submit(pinned[i % 2])
+            slot = eng.submit(pinned[i % len(pinned)])
             if prev is not None:
                 agg(eng.wait(prev))
             prev = slot
