@@ -46,7 +46,7 @@ def main():
     ap.add_argument("--no-rfi", action="store_true")
     ap.add_argument("--fft", choices=["native", "hipfft"], default="native",
                     help="FFT backend (hand-written Stockham vs hipFFT)")
-    ap.add_argument("--slots", type=int, default=2,
+    ap.add_argument("--slots", type=int, default=4,
                     help="double-buffered engine slots (streams)")
     args = ap.parse_args()
 

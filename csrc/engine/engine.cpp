@@ -89,6 +89,12 @@ PipelineEngine::PipelineEngine(const EngineConfig& cfg, int n_slots)
     }
   }
 
+  if (cfg.window_kind != 0) {
+    check_hip(hipMalloc(&window_, n_ * sizeof(float)), "window alloc");
+    check_hip(build_window(window_, n_, cfg.window_kind, slots_[0]->stream),
+              "window build");
+    check_hip(hipStreamSynchronize(slots_[0]->stream), "window sync");
+  }
   if (cfg.use_phase_table) {
     check_hip(hipMalloc(&phase_table_, nc_ * sizeof(float2)), "table alloc");
     check_hip(dedisp_phase_table(phase_table_, nc_, f_min_, f_c_, df_, cfg.dm,
@@ -126,6 +132,7 @@ PipelineEngine::~PipelineEngine() {
     if (s.stream) hipStreamDestroy(s.stream);
   }
   if (phase_table_) hipFree(phase_table_);
+  if (window_) hipFree(window_);
 }
 
 void PipelineEngine::enqueue_chain(Slot& s, const uint8_t* dev_raw,
@@ -139,8 +146,8 @@ void PipelineEngine::enqueue_chain(Slot& s, const uint8_t* dev_raw,
   const float* fft_in = s.samples;
   if (dev_raw) {
     // 1. unpack (+ window fused; default rectangle → none)
-    check_hip(unpack(dev_raw, s.samples, n_, cfg_.baseband_input_bits, nullptr,
-                     st),
+    check_hip(unpack(dev_raw, s.samples, n_, cfg_.baseband_input_bits,
+                     window_, st),
               "unpack");
   } else if (native_fft_) {
     // native fwd runs column passes in place on its input: copy the caller's

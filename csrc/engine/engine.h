@@ -51,6 +51,9 @@ struct EngineConfig {
   // 0 = hand-written Stockham FFT when shapes are pow2 (measured 2.2x
   // rocFFT on the J1644 waterfall, 1.1x on the 2^30 forward), 1 = hipFFT
   int fft_backend = 0;
+  // FFT window fused into unpack: 0 = rectangle (reference default),
+  // 1 = hann, 2 = hamming
+  int window_kind = 0;
 };
 
 struct BlockResult {
@@ -128,6 +131,7 @@ class PipelineEngine {
   int n_slots_ = 2;
 
   float2* phase_table_ = nullptr;  // shared across slots (read-only)
+  float* window_ = nullptr;        // fused FFT window table (null = rect)
   bool native_fft_ = false;        // hand-written FFT active
 
 

@@ -40,6 +40,10 @@ hipError_t unpack_gznupsr_a1(const uint8_t* in, float* out0, float* out1,
                              size_t count_per_stream, const float* window,
                              hipStream_t stream);
 
+// FFT window coefficient table (0 = rectangle [no-op values], 1 = hann,
+// 2 = hamming) — reference fft/fft_window.hpp.
+hipError_t build_window(float* coef, size_t n, int kind, hipStream_t stream);
+
 // ---------------- reductions ----------------
 // Deterministic two-pass mean of |x|^2 over n complex bins.
 // partials: device scratch of n_partials doubles (n_partials = reduce grid),

@@ -188,7 +188,29 @@ __global__ void k_unpack_gznupsr(const uint32_t* __restrict__ in,
   }
 }
 
+// cosine-sum window coefficient table: coef[i] = a0 - a1*cos(2*pi*i/(n-1))
+// (reference fft/fft_window.hpp:27-83; hann a0=a1=0.5, hamming 25/46, 21/46)
+__global__ void k_build_window(float* __restrict__ coef, size_t n, double a0,
+                               double a1) {
+  const size_t stride = (size_t)gridDim.x * blockDim.x;
+  const double inv = (n > 1) ? 1.0 / (double)(n - 1) : 0.0;
+  for (size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride)
+    coef[i] = (float)(a0 - a1 * cos(2.0 * M_PI * (double)i * inv));
+}
+
 }  // namespace
+
+hipError_t build_window(float* coef, size_t n, int kind, hipStream_t stream) {
+  double a0 = 1.0, a1 = 0.0;
+  if (kind == 1) { a0 = 0.5; a1 = 0.5; }                    // hann
+  else if (kind == 2) { a0 = 25.0 / 46.0; a1 = 21.0 / 46.0; }  // hamming
+  else if (kind != 0) return hipErrorInvalidValue;
+  hipLaunchKernelGGL(k_build_window, grid_for(n), dim3(kBlock), 0, stream,
+                     coef, n, a0, a1);
+  SRTB_CHECK_LAUNCH();
+  return hipSuccess;
+}
 
 hipError_t unpack(const uint8_t* in, float* out, size_t out_count, int nbits,
                   const float* window, hipStream_t stream) {

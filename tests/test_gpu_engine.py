@@ -159,3 +159,22 @@ def test_engine_boxcar_series_readback(C):
     box = eng.boxcar_series(slot, 4).cpu().numpy()
     expect = ref.boxcar_series(ts, 4)
     np.testing.assert_allclose(box, expect, rtol=1e-3, atol=2e-2)
+
+
+def test_engine_fused_window(C):
+    """window_kind=2 (hamming) fused at unpack matches the CPU oracle with
+    the same window."""
+    cfg = small_cfg()
+    rng = np.random.default_rng(9)
+    raw = np.clip(np.round(rng.normal(0, 16, cfg.baseband_input_count)),
+                  -128, 127).astype(np.int8).view(np.uint8)
+    pipe = CpuPipeline(cfg)
+    pipe.window_kind = "hamming"
+    res_cpu = pipe.process_block(raw)
+    eng = make_engine(C, cfg, window_kind=2)
+    slot = eng.submit(torch.from_numpy(raw.copy()))
+    eng.wait(slot)
+    ts_gpu = eng.time_series(slot).cpu().numpy()
+    ts_cpu = res_cpu["time_series"]
+    scale = max(np.abs(ts_cpu).max(), 1e-9)
+    np.testing.assert_allclose(ts_gpu / scale, ts_cpu / scale, atol=2e-3)
