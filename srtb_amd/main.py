@@ -184,6 +184,14 @@ def main(argv=None) -> int:
 
     t0 = time.time()
     n_blocks = 0
+    # baseband_write_all: record every block (minus the overlap tail) into
+    # one file per stream (reference write_file_pipe.hpp:41-94)
+    write_all_f = None
+    if cfg.baseband_write_all:
+        write_all_f = open(cfg.baseband_output_file_prefix +
+                           f"all_r{rank}.bin", "ab")
+        reserved_bytes = reserved * abs(cfg.baseband_input_bits) // 8
+
     if cfg.input_file_path:
         # file replay: single stream (the reference also replays one stream)
         pipe = make(cfg, reserved)
@@ -194,6 +202,9 @@ def main(argv=None) -> int:
         for sample_index, raw in reader:
             products = pipe.process_block(raw, sample_index)
             writer.push(products)
+            if write_all_f is not None:
+                end = raw.size - reserved_bytes if reserved_bytes else raw.size
+                write_all_f.write(raw[:end].tobytes())
             res = getattr(pipe, "last_result", None)
             if isinstance(res, dict) and "zero_count" in res:
                 zc = res["zero_count"]
@@ -240,6 +251,8 @@ def main(argv=None) -> int:
         run_receiver(provider, assembler, on_block, stop)
         n_blocks = state["n"]
 
+    if write_all_f is not None:
+        write_all_f.close()
     elapsed = time.time() - t0
     stats = agg.reduce()
     if rank == 0:

@@ -124,3 +124,27 @@ def test_write_ppm_roundtrip(tmp_path):
     write_ppm(p, img)
     data = open(p, "rb").read()
     assert data.endswith(bytes([0x12, 0x34, 0x56]) * 6)
+
+
+def test_main_baseband_write_all(tmp_path):
+    cfg, rec = make_recording(tmp_path, n_blocks=2)
+    cfg_file = tmp_path / "wa.cfg"
+    cfg_file.write_text(f"""
+baseband_input_count = 2 ** 17
+spectrum_channel_count = 2 ** 6
+baseband_input_bits = -8
+baseband_sample_rate = 128 * 1e6
+baseband_freq_low = 1400
+baseband_bandwidth = 64
+dm = 0
+baseband_reserve_sample = 0
+baseband_write_all = 1
+signal_detect_signal_noise_threshold = 100
+input_file_path = {rec}
+baseband_output_file_prefix = {tmp_path}/wa_
+""")
+    rc = main(["--config_file_name", str(cfg_file), "--device", "cpu"])
+    assert rc == 0
+    data = np.fromfile(tmp_path / "wa_all_r0.bin", dtype=np.uint8)
+    orig = np.fromfile(rec, dtype=np.uint8)
+    np.testing.assert_array_equal(data, orig)
