@@ -248,31 +248,6 @@ class NativeFft {
     return t;
   }
 
-  static std::vector<uint32_t> factorize_unused(size_t len) {
-    // split into <= 3 balanced pow2 factors, each <= kMaxPassLen;
-    // the LAST factor is the largest (it gets the contiguous-load pass).
-    if (len <= kMaxPassLen) return {(uint32_t)len};
-    int t = 0;
-    while ((1ull << t) < len) ++t;
-    const int tm = 12;  // log2(kMaxPassLen)
-    if (len <= (size_t)kMaxPassLen * kMaxPassLen) {
-      int t2 = std::min(tm, (t + 1) / 2);
-      int t1 = t - t2;
-      if (t1 > tm) { t1 = tm; t2 = t - tm; }
-      return {(uint32_t)(1u << std::min(t1, t2)),
-              (uint32_t)(1u << std::max(t1, t2))};
-    }
-    int t3 = std::min(tm, (t + 2) / 3);
-    int rem = t - t3;
-    int t2 = std::min(tm, (rem + 1) / 2);
-    int t1 = rem - t2;
-    if (t1 > tm) throw std::runtime_error("NativeFft: length too large");
-    std::vector<int> v{t1, t2, t3};
-    std::sort(v.begin(), v.end());
-    return {(uint32_t)(1u << v[0]), (uint32_t)(1u << v[1]),
-            (uint32_t)(1u << v[2])};
-  }
-
   int pick_f(uint32_t n, const Pass& p, size_t n_ffts) const {
     size_t f = (size_t)elems_per_wg() / n;
     if (f < 1) f = 1;
