@@ -420,6 +420,12 @@ __device__ inline void dif_addr(unsigned long long id,
   obase = row * d.row_len + J;
 }
 
+// LDS bank swizzle for the DIF data array: butterfly strides of 4^k element
+// granularity put 2-4 lanes of a 32-lane b64 group on one bank (measured
+// 4e8 conflicts/dispatch); e ^ ((e>>2)&31) is bijective within any pow2 row
+// >= 32 and brute-force conflict-free for every stage pattern of n=256/1024.
+__device__ inline int dif_swz(int i) { return i ^ ((i >> 2) & 31); }
+
 __device__ inline unsigned rev4_bits(unsigned k, int t2 /* log2(n) */) {
   // base-4 digit reversal = bit reversal then swap adjacent bit pairs
   unsigned r = __brev(k) >> (32 - t2);
@@ -457,7 +463,7 @@ __global__ void __launch_bounds__(256)
         unsigned long long in_blk, obase;
         dif_addr(fft0 + f, d, in_blk, obase);
         tmp[k] = in[in_blk + i];
-        lidx[k] = f * ldst + i;
+        lidx[k] = f * ldst + dif_swz(i);
       }
 #pragma unroll
       for (int k = 0; k < 8; ++k) X[lidx[k]] = tmp[k];
@@ -467,7 +473,7 @@ __global__ void __launch_bounds__(256)
       const int i = e & (n - 1);
       unsigned long long in_blk, obase;
       dif_addr(fft0 + f, d, in_blk, obase);
-      X[f * ldst + i] = in[in_blk + i];
+      X[f * ldst + dif_swz(i)] = in[in_blk + i];
     }
   }
   __syncthreads();
@@ -484,10 +490,14 @@ __global__ void __launch_bounds__(256)
       const int j = bb & (M - 1);
       const int g = (bb >> m_log2) << (m_log2 + 2);
       float2* row = X + f * ldst;
-      const float2 a = row[g + j];
-      const float2 bv = row[g + j + M];
-      const float2 c = row[g + j + 2 * M];
-      const float2 dv = row[g + j + 3 * M];
+      const int e0 = dif_swz(g + j);
+      const int e1 = dif_swz(g + j + M);
+      const int e2 = dif_swz(g + j + 2 * M);
+      const int e3 = dif_swz(g + j + 3 * M);
+      const float2 a = row[e0];
+      const float2 bv = row[e1];
+      const float2 c = row[e2];
+      const float2 dv = row[e3];
       // DFT over t then post-twiddle W_L^{m*j} (dif_r4 prototype)
       const float2 t0 = make_float2(a.x + c.x, a.y + c.y);
       const float2 t1 = make_float2(a.x - c.x, a.y - c.y);
@@ -500,10 +510,10 @@ __global__ void __launch_bounds__(256)
       const float2 x2 = make_float2(t0.x - t2.x, t0.y - t2.y);
       const float2 x3 = make_float2(t1.x - t3.x, t1.y - t3.y);
       const int ts = tl;  // W_L^x = ltw[x << tl]
-      row[g + j] = x0;
-      row[g + j + M] = (j == 0) ? x1 : cmulf(x1, ltw[(j << ts)]);
-      row[g + j + 2 * M] = (j == 0) ? x2 : cmulf(x2, ltw[(2 * j) << ts]);
-      row[g + j + 3 * M] = (j == 0) ? x3 : cmulf(x3, ltw[(3 * j) << ts]);
+      row[e0] = x0;
+      row[e1] = (j == 0) ? x1 : cmulf(x1, ltw[(j << ts)]);
+      row[e2] = (j == 0) ? x2 : cmulf(x2, ltw[(2 * j) << ts]);
+      row[e3] = (j == 0) ? x3 : cmulf(x3, ltw[(3 * j) << ts]);
     }
     __syncthreads();
   }
@@ -516,7 +526,7 @@ __global__ void __launch_bounds__(256)
     dif_addr(fft0 + f, d, in_blk, obase);
     const unsigned rk = rev4_bits((unsigned)k, nl);
     out[obase + (unsigned long long)k * d.out_elem_coef] =
-        X[f * ldst + (int)rk];
+        X[f * ldst + dif_swz((int)rk)];
   }
 }
 
@@ -556,9 +566,11 @@ __global__ void k_r2c_post(const float2* __restrict__ z,
                                     0.5f * (zk.y + zmc.y));
     const float2 dif = make_float2(zk.x - zmc.x, zk.y - zmc.y);
     const float2 odd = make_float2(0.5f * dif.y, -0.5f * dif.x);
-    double sw, cw;
-    sincos(-M_PI * (double)k / (double)m, &sw, &cw);
-    const float2 w = make_float2((float)cw, (float)sw);
+    // argument reduced in fp64 (k/m exact), fast f32 sincos (~1.5e-6 rad)
+    const float phiw = (float)(-M_PI * (double)k / (double)m);
+    float sw, cw;
+    __sincosf(phiw, &sw, &cw);
+    const float2 w = make_float2(cw, sw);
     const float2 wo = cmulf(w, odd);
     const float2 xk = make_float2(even.x + wo.x, even.y + wo.y);
     x[k] = xk;

@@ -249,15 +249,18 @@ __global__ void k_sk_flags(const float2* __restrict__ wf,
   if (first_zero) atomicAdd(zero_count, 1u);
 }
 
+// 2-D grid: y = row (whole workgroup early-outs on unflagged rows — the
+// 1-D version spent 86% VALU on a per-element 64-bit division)
 __global__ void k_sk_zap_rows(float2* __restrict__ wf,
                               const uint8_t* __restrict__ flags, size_t rows,
                               size_t len) {
-  const size_t total = rows * len;
+  const size_t row = blockIdx.y;
+  if (!flags[row]) return;
   const size_t stride = (size_t)gridDim.x * blockDim.x;
-  for (size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
-       i += stride) {
-    if (flags[i / len]) wf[i] = make_float2(0.0f, 0.0f);
-  }
+  float2* rp = wf + row * len;
+  for (size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x; i < len;
+       i += stride)
+    rp[i] = make_float2(0.0f, 0.0f);
 }
 
 // ---------------- time series + detection ----------------
@@ -572,8 +575,10 @@ hipError_t sk_flags(const float2* wf, const float2* s2s4, size_t rows,
 
 hipError_t sk_zap_rows(float2* wf, const uint8_t* flags, size_t rows,
                        size_t len, hipStream_t stream) {
-  hipLaunchKernelGGL(k_sk_zap_rows, grid_for(rows * len), dim3(kBlock), 0,
-                     stream, wf, flags, rows, len);
+  if (rows > 65535) return hipErrorInvalidValue;
+  dim3 grid(grid_for(len).x, (uint32_t)rows);
+  hipLaunchKernelGGL(k_sk_zap_rows, grid, dim3(kBlock), 0, stream, wf, flags,
+                     rows, len);
   SRTB_CHECK_LAUNCH();
   return hipSuccess;
 }
