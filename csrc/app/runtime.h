@@ -9,6 +9,7 @@
 #include <array>
 #include <chrono>
 #include <csignal>
+#include <cstdlib>
 #include <cstdio>
 #include <cstring>
 #include <execinfo.h>
@@ -28,7 +29,11 @@ enum class LogLevel : int { kNone = 0, kError = 1, kWarning = 2, kInfo = 3,
                             kDebug = 4 };
 
 inline std::atomic<int>& log_level() {
-  static std::atomic<int> level{3};
+  // default from SRTB_LOG_LEVEL env like the reference (log/log.hpp)
+  static std::atomic<int> level{[] {
+    const char* e = std::getenv("SRTB_LOG_LEVEL");
+    return e ? std::atoi(e) : 3;
+  }()};
   return level;
 }
 

@@ -222,7 +222,11 @@ def main(argv=None) -> int:
             if max_blocks is not None and n_blocks >= max_blocks:
                 break
     else:
-        # UDP ingest: shard endpoints across ranks
+        # UDP ingest: endpoints sharded across ranks; each rank runs one
+        # receiver thread per assigned endpoint (the reference spawns N udp
+        # receiver pipes, main.cpp:230-272), feeding a shared work queue
+        import queue as _queue
+        import threading
         from .io.udp import BlockAssembler, UdpPacketProvider, run_receiver
         backend = bk.get_backend(cfg.baseband_format_type)
         n_endpoints = len(cfg.udp_receiver_address)
@@ -230,25 +234,40 @@ def main(argv=None) -> int:
         if not my:
             print(f"rank {rank}: no UDP endpoints assigned")
             return 0
-        ep = my[0]  # one endpoint per rank in this runner
-        pipe = make(cfg, reserved)
         bits = abs(cfg.baseband_input_bits)
         block_bytes = cfg.baseband_input_count * bits // 8 * \
             bk.get_data_stream_count(cfg.baseband_format_type)
-        assembler = BlockAssembler(backend, block_bytes)
-        provider = UdpPacketProvider(cfg.udp_receiver_address[ep],
-                                     cfg.udp_receiver_port[ep])
-        state = {"n": 0}
-
-        def on_block(blk, ts):
-            products = pipe.process_block(blk, ts)
-            writer.push(products)
-            state["n"] += 1
+        pipes = {ep: make(cfg, reserved) for ep in my}
+        q: "_queue.Queue" = _queue.Queue(maxsize=2 * len(my))
+        state = {"n": 0, "stop": False}
 
         def stop():
-            return max_blocks is not None and state["n"] >= max_blocks
+            return state["stop"] or (max_blocks is not None and
+                                     state["n"] >= max_blocks)
 
-        run_receiver(provider, assembler, on_block, stop)
+        def receiver(ep):
+            assembler = BlockAssembler(backend, block_bytes)
+            provider = UdpPacketProvider(cfg.udp_receiver_address[ep],
+                                         cfg.udp_receiver_port[ep])
+            try:
+                run_receiver(provider, assembler,
+                             lambda blk, ts: q.put((ep, blk, ts)), stop)
+            finally:
+                provider.close()
+
+        threads = [threading.Thread(target=receiver, args=(ep,), daemon=True)
+                   for ep in my]
+        for t in threads:
+            t.start()
+        while not stop():
+            try:
+                ep, blk, ts = q.get(timeout=0.2)
+            except _queue.Empty:
+                continue
+            products = pipes[ep].process_block(blk, ts)
+            writer.push(products)
+            state["n"] += 1
+        state["stop"] = True
         n_blocks = state["n"]
 
     if write_all_f is not None:

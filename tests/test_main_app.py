@@ -148,3 +148,40 @@ baseband_output_file_prefix = {tmp_path}/wa_
     data = np.fromfile(tmp_path / "wa_all_r0.bin", dtype=np.uint8)
     orig = np.fromfile(rec, dtype=np.uint8)
     np.testing.assert_array_equal(data, orig)
+
+
+def test_main_udp_two_endpoints(tmp_path):
+    """One rank serving two UDP endpoints (reference N input pipes)."""
+    import socket
+    import threading
+    import time as _time
+    from srtb_amd.main import main as srtb_main
+
+    ports = [29931, 29932]
+    payload = 4096
+    block_samples = payload * 2
+
+    def sender():
+        s = socket.socket(socket.AF_INET, socket.SOCK_DGRAM)
+        _time.sleep(0.8)
+        for c in range(5):  # enough for 2 blocks per endpoint
+            for p in ports:
+                pkt = struct.pack("<Q", c) + bytes([c]) * payload
+                s.sendto(pkt, ("127.0.0.1", p))
+            _time.sleep(0.01)
+        s.close()
+
+    t = threading.Thread(target=sender)
+    t.start()
+    rc = srtb_main([
+        "--baseband_format_type", "fastmb_roach2",
+        "--baseband_input_count", str(block_samples),
+        "--baseband_input_bits", "8",
+        "--spectrum_channel_count", "16",
+        "--baseband_reserve_sample", "0",
+        "--udp_receiver_address", "127.0.0.1, 127.0.0.1",
+        "--udp_receiver_port", f"{ports[0]}, {ports[1]}",
+        "--baseband_output_file_prefix", str(tmp_path) + "/u_",
+        "--device", "cpu", "--max-blocks", "4"])
+    t.join()
+    assert rc == 0
