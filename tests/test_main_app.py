@@ -150,9 +150,11 @@ baseband_output_file_prefix = {tmp_path}/wa_
     np.testing.assert_array_equal(data, orig)
 
 
+@pytest.mark.timeout(60)
 def test_main_udp_two_endpoints(tmp_path):
     """One rank serving two UDP endpoints (reference N input pipes)."""
     import socket
+    import struct
     import threading
     import time as _time
     from srtb_amd.main import main as srtb_main
