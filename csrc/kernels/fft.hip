@@ -345,11 +345,15 @@ __global__ void __launch_bounds__(256)
     if (pre.mean_power) thr_mean = pre.threshold * (float)(*pre.mean_power);
   }
   float2 v[N];
+  // a column spans (N-1)*stride < L <= 2^29 elements, so 32-bit offsets
+  // from the column base suffice — halves the in-flight address registers
+  const float2* __restrict__ colbase = in + base;
+  const uint32_t stride32 = (uint32_t)d.in_stride;
 #pragma unroll
   for (int i = 0; i < N; ++i) {
-    const unsigned long long flat =
-        base + (unsigned long long)i * d.in_stride;
-    float2 x = in[flat];
+    const uint32_t off = (uint32_t)i * stride32;
+    float2 x = colbase[off];
+    const unsigned long long flat = base + off;
     if constexpr (PREOP) {
       // fused rfi_dedisperse (spectrum.hip k_rfi_dedisp_fused semantics);
       // `flat` IS the spectrum bin index for in-place column passes
@@ -370,6 +374,7 @@ __global__ void __launch_bounds__(256)
     v[col_sigma<N>(i)] = x;
   }
   col_fft<N, SIGN>(v, tw_n);
+  float2* __restrict__ ocolbase = out + base;
 #pragma unroll
   for (int k = 0; k < N; ++k) {
     float2 r = v[k];
@@ -378,7 +383,7 @@ __global__ void __launch_bounds__(256)
           (q0 * d.tw_f0 * (unsigned long long)k) & d.tw_mask;
       r = cmulf(r, tw_eval(m_, d.tw_angle));
     }
-    out[base + (unsigned long long)k * d.in_stride] = r;
+    ocolbase[(uint32_t)k * stride32] = r;
   }
 }
 
