@@ -46,6 +46,8 @@ def main():
     ap.add_argument("--no-rfi", action="store_true")
     ap.add_argument("--fft", choices=["native", "hipfft"], default="native",
                     help="FFT backend (hand-written Stockham vs hipFFT)")
+    ap.add_argument("--graph", action="store_true",
+                    help="capture the per-block chain into hipGraphs")
     ap.add_argument("--slots", type=int, default=4,
                     help="double-buffered engine slots (streams)")
     args = ap.parse_args()
@@ -91,7 +93,8 @@ def main():
         snr_threshold=snr, max_boxcar=max_boxcar, nsamps_reserved=0,
         zap_ranges=zap_ranges, use_phase_table=args.phase_table,
         enable_rfi_s1=not args.no_rfi, enable_sk=True, n_slots=args.slots,
-        fft_backend=0 if args.fft == "native" else 1)
+        fft_backend=0 if args.fft == "native" else 1,
+        use_hip_graph=args.graph)
 
     # synthetic 2-bit baseband noise, pinned, one buffer per slot
     rng = np.random.default_rng(1234 + rank)

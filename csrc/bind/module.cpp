@@ -481,7 +481,7 @@ class PyEngine {
            int64_t max_boxcar, int64_t nsamps_reserved,
            std::vector<std::vector<int64_t>> zap_ranges, bool use_phase_table,
            bool enable_rfi_s1, bool enable_sk, int64_t n_slots,
-           int64_t fft_backend, int64_t window_kind) {
+           int64_t fft_backend, int64_t window_kind, bool use_hip_graph) {
     EngineConfig c;
     c.baseband_input_count = n;
     c.baseband_input_bits = (int)nbits;
@@ -506,6 +506,7 @@ class PyEngine {
     c.enable_sk = enable_sk;
     c.fft_backend = (int)fft_backend;
     c.window_kind = (int)window_kind;
+    c.use_hip_graph = use_hip_graph;
     eng_ = std::make_unique<PipelineEngine>(c, (int)n_slots);
   }
 
@@ -620,7 +621,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def(py::init<int64_t, int64_t, int64_t, double, double, double, double,
                     double, double, double, int64_t, int64_t,
                     std::vector<std::vector<int64_t>>, bool, bool, bool,
-                    int64_t, int64_t, int64_t>(),
+                    int64_t, int64_t, int64_t, bool>(),
            py::arg("n"), py::arg("nbits"), py::arg("channels"),
            py::arg("freq_low"), py::arg("bandwidth"), py::arg("sample_rate"),
            py::arg("dm"), py::arg("rfi_threshold") = 10.0,
@@ -630,7 +631,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
            py::arg("use_phase_table") = false,
            py::arg("enable_rfi_s1") = true, py::arg("enable_sk") = true,
            py::arg("n_slots") = 2, py::arg("fft_backend") = 0,
-           py::arg("window_kind") = 0)
+           py::arg("window_kind") = 0, py::arg("use_hip_graph") = false)
       .def("submit", &PyEngine::submit, py::arg("raw"),
            py::arg("dm_override") = std::nan(""))
       .def("submit_samples", &PyEngine::submit_samples, py::arg("samples"),

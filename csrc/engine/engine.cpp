@@ -128,6 +128,7 @@ PipelineEngine::~PipelineEngine() {
     hipFree(s.thresholds);
     hipHostFree(s.h_counters);
     hipHostFree(s.h_thresholds);
+    if (s.graph_exec) hipGraphExecDestroy(s.graph_exec);
     if (s.done) hipEventDestroy(s.done);
     if (s.stream) hipStreamDestroy(s.stream);
   }
@@ -136,7 +137,8 @@ PipelineEngine::~PipelineEngine() {
 }
 
 void PipelineEngine::enqueue_chain(Slot& s, const uint8_t* dev_raw,
-                                   const float* dev_samples, double dm) {
+                                   const float* dev_samples, double dm,
+                                   bool record_event) {
   hipStream_t st = s.stream;
   if (std::isnan(dm)) dm = cfg_.dm;
   const float2* table = phase_table_;
@@ -259,8 +261,10 @@ void PipelineEngine::enqueue_chain(Slot& s, const uint8_t* dev_raw,
                            (1 + n_boxcars_) * sizeof(float),
                            hipMemcpyDeviceToHost, st),
             "thr d2h");
-  check_hip(hipEventRecord(s.done, st), "event record");
-  s.busy = true;
+  if (record_event) {
+    check_hip(hipEventRecord(s.done, st), "event record");
+    s.busy = true;
+  }
 }
 
 int PipelineEngine::submit(const void* host_bytes, size_t nbytes,
@@ -272,6 +276,37 @@ int PipelineEngine::submit(const void* host_bytes, size_t nbytes,
   if (s.busy) {
     check_hip(hipEventSynchronize(s.done), "slot wait");
     s.busy = false;
+  }
+  const bool graph_ok = cfg_.use_hip_graph && std::isnan(dm_override);
+  if (graph_ok && s.graph_exec && s.graph_host_src == host_bytes) {
+    // steady state: replay the captured chain (H2D + kernels + D2H)
+    check_hip(hipGraphLaunch(s.graph_exec, s.stream), "graph launch");
+    check_hip(hipEventRecord(s.done, s.stream), "event record");
+    s.busy = true;
+    return id;
+  }
+  if (graph_ok && s.graph_pending >= 1 && !s.graph_exec) {
+    // second submission from the same host buffer: capture it
+    check_hip(hipStreamBeginCapture(s.stream, hipStreamCaptureModeThreadLocal),
+              "begin capture");
+    check_hip(hipMemcpyAsync(s.raw, host_bytes, nbytes,
+                             hipMemcpyHostToDevice, s.stream),
+              "raw h2d");
+    enqueue_chain(s, s.raw, nullptr, dm_override, /*record_event=*/false);
+    hipGraph_t g = nullptr;
+    check_hip(hipStreamEndCapture(s.stream, &g), "end capture");
+    check_hip(hipGraphInstantiate(&s.graph_exec, g, nullptr, nullptr, 0),
+              "graph instantiate");
+    check_hip(hipGraphDestroy(g), "graph destroy");
+    s.graph_host_src = host_bytes;
+    check_hip(hipGraphLaunch(s.graph_exec, s.stream), "graph launch");
+    check_hip(hipEventRecord(s.done, s.stream), "event record");
+    s.busy = true;
+    return id;
+  }
+  if (graph_ok) {
+    s.graph_pending = 1;
+    s.graph_host_src = host_bytes;
   }
   check_hip(hipMemcpyAsync(s.raw, host_bytes, nbytes, hipMemcpyHostToDevice,
                            s.stream),

@@ -54,6 +54,10 @@ struct EngineConfig {
   // FFT window fused into unpack: 0 = rectangle (reference default),
   // 1 = hann, 2 = hamming
   int window_kind = 0;
+  // capture the steady-state per-block chain (~55 launches) into a hipGraph
+  // per slot and replay it; falls back to direct enqueue whenever the
+  // submission differs from the captured one (dm override, other host ptr)
+  bool use_hip_graph = false;
 };
 
 struct BlockResult {
@@ -119,7 +123,8 @@ class PipelineEngine {
  private:
   struct Slot;
   void enqueue_chain(Slot& s, const uint8_t* dev_raw,
-                     const float* dev_samples, double dm);
+                     const float* dev_samples, double dm,
+                     bool record_event = true);
 
   EngineConfig cfg_;
   size_t n_, nc_, s_, l_, ts_count_, raw_bytes_;
@@ -159,6 +164,10 @@ class PipelineEngine {
     float* h_thresholds = nullptr;  // pinned
     FftPlanSet plans;               // hipFFT fallback
     NativeFft nfwd, nbwd;           // hand-written path
+    // hipGraph replay state
+    hipGraphExec_t graph_exec = nullptr;
+    const void* graph_host_src = nullptr;  // host ptr the capture used
+    int graph_pending = 0;  // submissions seen for this slot
   };
   std::vector<std::unique_ptr<Slot>> slots_;
   int next_slot_ = 0;
