@@ -218,6 +218,29 @@ std::vector<torch::Tensor> t_sk_mitigate(torch::Tensor wf, double sk_threshold) 
   return {flags, zero_count};
 }
 
+torch::Tensor t_sk_v1_mitigate(torch::Tensor wf, double sk_threshold,
+                               bool normalize) {
+  CHECK_CUDA(wf);
+  CHECK_CONTIG(wf);
+  TORCH_CHECK(wf.dim() == 2);  // [M][bins], time-major
+  const size_t M = wf.size(0), bins = wf.size(1);
+  auto s2s4 = torch::empty({(int64_t)bins, 2},
+                           wf.options().dtype(torch::kFloat32));
+  check(sk_v1_stats(cptr(wf), M, bins,
+                    reinterpret_cast<float2*>(s2s4.data_ptr<float>()),
+                    cur_stream()),
+        "sk_v1_stats");
+  double hi = sk_threshold, lo = 2.0 - hi;
+  if (lo > hi) std::swap(lo, hi);
+  const double corr = ((double)M - 1.0) / ((double)M + 1.0);
+  check(sk_v1_zap(cptr(wf), M, bins,
+                  reinterpret_cast<const float2*>(s2s4.data_ptr<float>()),
+                  (float)(lo * corr + 1.0), (float)(hi * corr + 1.0),
+                  normalize, cur_stream()),
+        "sk_v1_zap");
+  return s2s4;
+}
+
 torch::Tensor t_time_series(torch::Tensor wf,
                             c10::optional<torch::Tensor> flags,
                             int64_t ts_count) {
@@ -599,6 +622,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("dm"), py::arg("table") = c10::nullopt);
   m.def("sk_row_stats", &t_sk_row_stats);
   m.def("sk_mitigate", &t_sk_mitigate);
+  m.def("sk_v1_mitigate", &t_sk_v1_mitigate, py::arg("wf"),
+        py::arg("sk_threshold"), py::arg("normalize") = false);
   m.def("time_series", &t_time_series, py::arg("wf"), py::arg("flags"),
         py::arg("ts_count"));
   m.def("subtract_mean", &t_subtract_mean);

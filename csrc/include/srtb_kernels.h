@@ -111,6 +111,15 @@ hipError_t sk_flags(const float2* wf, const float2* s2s4, size_t rows,
 hipError_t sk_zap_rows(float2* wf, const uint8_t* flags, size_t rows,
                        size_t len, hipStream_t stream);
 
+// SK method 1 (reference rfi_mitigation.hpp:183-274): time-major [M][bins]
+// waterfall; thresholds already corrected like sk_flags.  normalize applies
+// the optional per-bin 1/sqrt(mean|x|^2) of surviving bins.
+hipError_t sk_v1_stats(const float2* wf, size_t M, size_t bins, float2* s2s4,
+                       hipStream_t stream);
+hipError_t sk_v1_zap(float2* wf, size_t M, size_t bins, const float2* s2s4,
+                     float lo_corrected, float hi_corrected, bool normalize,
+                     hipStream_t stream);
+
 // ---------------- signal detection ----------------
 // ts[j] = sum_i |wf[i][j]|^2 over non-flagged rows, j < ts_count
 // (reference signal_detect_pipe.hpp:305-316; flags may be null).

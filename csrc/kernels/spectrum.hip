@@ -263,6 +263,43 @@ __global__ void k_sk_zap_rows(float2* __restrict__ wf,
     rp[i] = make_float2(0.0f, 0.0f);
 }
 
+// SK method 1 (reference rfi_mitigation.hpp:183-274): TIME-MAJOR layout
+// [M][bins]; one thread per frequency bin walks its column (stride = bins,
+// coalesced across threads), then a zero/normalize pass.
+__global__ void k_sk_v1_stats(const float2* __restrict__ wf, size_t M,
+                              size_t bins, float2* __restrict__ s2s4) {
+  const size_t j = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (j >= bins) return;
+  float s2 = 0.0f, s4 = 0.0f;
+  for (size_t m = 0; m < M; ++m) {
+    const float p = norm2(wf[m * bins + j]);
+    s2 += p;
+    s4 += p * p;
+  }
+  s2s4[j] = make_float2(s2, s4);
+}
+
+template <bool kNormalize>
+__global__ void k_sk_v1_zap(float2* __restrict__ wf, size_t M, size_t bins,
+                            const float2* __restrict__ s2s4, float lo_,
+                            float hi_) {
+  const size_t total = M * bins;
+  const size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += stride) {
+    const size_t j = i % bins;  // frequency bin (bins need not be pow2 here)
+    const float2 p = s2s4[j];
+    const float sk = (float)M * (p.y / (p.x * p.x));
+    const bool zap = (sk > hi_) || (sk < lo_) || !(p.x > 0.0f);
+    if (zap) {
+      wf[i] = make_float2(0.0f, 0.0f);
+    } else if constexpr (kNormalize) {
+      const float scale = rsqrtf(p.x / (float)M);
+      wf[i] = make_float2(wf[i].x * scale, wf[i].y * scale);
+    }
+  }
+}
+
 // ---------------- time series + detection ----------------
 
 // ts[j] = sum over non-flagged rows of |wf[row][j]|^2.
@@ -579,6 +616,27 @@ hipError_t sk_zap_rows(float2* wf, const uint8_t* flags, size_t rows,
   dim3 grid(grid_for(len).x, (uint32_t)rows);
   hipLaunchKernelGGL(k_sk_zap_rows, grid, dim3(kBlock), 0, stream, wf, flags,
                      rows, len);
+  SRTB_CHECK_LAUNCH();
+  return hipSuccess;
+}
+
+hipError_t sk_v1_stats(const float2* wf, size_t M, size_t bins, float2* s2s4,
+                       hipStream_t stream) {
+  hipLaunchKernelGGL(k_sk_v1_stats, grid_for(bins), dim3(kBlock), 0, stream,
+                     wf, M, bins, s2s4);
+  SRTB_CHECK_LAUNCH();
+  return hipSuccess;
+}
+
+hipError_t sk_v1_zap(float2* wf, size_t M, size_t bins, const float2* s2s4,
+                     float lo_, float hi_, bool normalize,
+                     hipStream_t stream) {
+  if (normalize)
+    hipLaunchKernelGGL((k_sk_v1_zap<true>), grid_for(M * bins), dim3(kBlock),
+                       0, stream, wf, M, bins, s2s4, lo_, hi_);
+  else
+    hipLaunchKernelGGL((k_sk_v1_zap<false>), grid_for(M * bins), dim3(kBlock),
+                       0, stream, wf, M, bins, s2s4, lo_, hi_);
   SRTB_CHECK_LAUNCH();
   return hipSuccess;
 }

@@ -305,26 +305,32 @@ def rfi_mitigate_sk(wf: np.ndarray, sk_threshold: float) -> np.ndarray:
     return wf
 
 
-def rfi_mitigate_sk_v1(wf: np.ndarray, sk_threshold: float,
+def rfi_mitigate_sk_v1(wf_tf: np.ndarray, sk_threshold: float,
                        normalize: bool = False) -> np.ndarray:
-    """Method 1 (reference rfi_mitigation.hpp:183-274): same SK statistic, but
-    thresholds are used raw (the -1 constant folded into the threshold) and an
-    optional per-channel normalization by sqrt(mean power) is applied."""
-    wf = np.asarray(wf).copy()
-    M = wf.shape[1]
+    """Method 1 (reference rfi_mitigation.hpp:183-274): TIME-MAJOR layout
+    [M][fft_bins] (one thread per frequency bin walks the M time samples of
+    its column), same corrected SK thresholds as method 2, zeroing whole
+    frequency columns; optional per-column normalization by sqrt(mean |x|^2)
+    of the surviving data."""
+    wf = np.asarray(wf_tf).copy()
+    M, bins = wf.shape
     hi = float(sk_threshold)
     lo = 2.0 - hi
     if lo > hi:
         lo, hi = hi, lo
     corr = (M - 1.0) / (M + 1.0)
     lo_, hi_ = lo * corr + 1.0, hi * corr + 1.0
-    sk = spectral_kurtosis_sk(wf)
-    zap = (sk > hi_) | (sk < lo_)
-    wf[zap, :] = 0
+    p = wf.real.astype(np.float64) ** 2 + wf.imag.astype(np.float64) ** 2
+    s2 = p.sum(axis=0)
+    s4 = (p * p).sum(axis=0)
+    with np.errstate(divide="ignore", invalid="ignore"):
+        sk = M * s4 / (s2 * s2)
+    zap = (sk > hi_) | (sk < lo_) | (s2 == 0)
+    wf[:, zap] = 0
     if normalize:
-        p = (np.abs(wf.astype(np.complex128)) ** 2).mean(axis=1, keepdims=True)
-        scale = np.where(p > 0, 1.0 / np.sqrt(p), 0.0)
-        wf = (wf * scale).astype(wf.dtype)
+        mp = np.where(zap, 1.0, s2 / M)
+        scale = np.where(zap, 0.0, 1.0 / np.sqrt(mp))
+        wf = (wf * scale[None, :]).astype(wf.dtype)
     return wf
 
 

@@ -285,3 +285,18 @@ def test_correlate_spectra():
     f2 = np.array([1 - 1j, 1 + 1j], dtype=np.complex64)
     out = ref.correlate_spectra(f1, f2, 0.5)
     np.testing.assert_allclose(out, 0.5 * f1 * np.conj(f2), rtol=1e-6)
+
+
+def test_sk_v1_time_major_layout():
+    rng = np.random.default_rng(8)
+    M, bins = 512, 64
+    wf = (rng.normal(size=(M, bins)) + 1j * rng.normal(size=(M, bins))
+          ).astype(np.complex64)
+    wf[:, 9] = 0
+    wf[::50, 9] = 30.0  # bursty bin -> SK >> 1
+    out = ref.rfi_mitigate_sk_v1(wf, 1.05)
+    assert (out[:, 9] == 0).all()
+    assert (np.abs(out[:, 8]) > 0).any()
+    outn = ref.rfi_mitigate_sk_v1(wf, 1.05, normalize=True)
+    surv = np.abs(outn[:, 8].astype(np.complex128)) ** 2
+    np.testing.assert_allclose(surv.mean(), 1.0, rtol=0.05)

@@ -283,3 +283,17 @@ def test_correlate(C):
     np.testing.assert_allclose(corr.cpu().numpy(), expect, rtol=1e-4, atol=1e-5)
     np.testing.assert_allclose(mag.cpu().numpy(), np.abs(expect), rtol=1e-4,
                                atol=1e-5)
+
+
+@pytest.mark.parametrize("normalize", [False, True])
+def test_sk_v1_mitigate(C, normalize):
+    rng = np.random.default_rng(30)
+    M, bins = 256, 96  # bins deliberately non-pow2
+    wf = (rng.normal(size=(M, bins)) + 1j * rng.normal(size=(M, bins))
+          ).astype(np.complex64)
+    wf[:, 5] = 0
+    wf[::40, 5] = 25.0
+    expect = ref.rfi_mitigate_sk_v1(wf, 1.05, normalize=normalize)
+    g = to_gpu(wf)
+    C.sk_v1_mitigate(g, 1.05, normalize)
+    np.testing.assert_allclose(g.cpu().numpy(), expect, rtol=1e-4, atol=1e-4)
