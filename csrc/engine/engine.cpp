@@ -1,6 +1,9 @@
 #include "engine.h"
 
+#include <roctracer/roctx.h>
+
 #include <cmath>
+#include <cstdlib>
 #include <cstring>
 #include <stdexcept>
 
@@ -8,6 +11,26 @@ namespace srtb_hip {
 
 namespace {
 constexpr double kD = 4.148808e3;  // dispersion constant (MHz^2 pc^-1 cm^3 s)
+
+// rocTX ranges per pipeline stage (SRTB_ROCTX=1): the rocprofv3 marker
+// domain shows unpack/fft/rfi/sk/detect spans per block — the reference's
+// per-pipe thread names, in profiler form (SURVEY §5 tracing).
+struct RoctxRange {
+  bool on;
+  explicit RoctxRange(const char* name)
+      : on([] {
+          static const bool enabled = [] {
+            const char* e = std::getenv("SRTB_ROCTX");
+            return e && std::atoi(e) != 0;
+          }();
+          return enabled;
+        }()) {
+    if (on) roctxRangePush(name);
+  }
+  ~RoctxRange() {
+    if (on) roctxRangePop();
+  }
+};
 }
 
 PipelineEngine::PipelineEngine(const EngineConfig& cfg, int n_slots)
@@ -145,6 +168,7 @@ void PipelineEngine::enqueue_chain(Slot& s, const uint8_t* dev_raw,
   if (table && dm != cfg_.dm)
     throw std::runtime_error("dm override requires use_phase_table=false");
 
+  RoctxRange r_chain("srtb_block_chain");
   const float* fft_in = s.samples;
   if (dev_raw) {
     // 1. unpack (+ window fused; default rectangle → none)

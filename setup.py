@@ -89,7 +89,7 @@ def build_apps(objects):
             continue
         cmd = [HIPCC, *HIPCC_FLAGS, "-x", "hip", src_path, "-x", "none", *objects,
                f"-I{ROOT}/csrc/include", f"-L{ROCM}/lib", "-lhipfft",
-               "-o", out]
+               "-lroctx64", "-o", out]
         print("+", " ".join(cmd), flush=True)
         subprocess.check_call(cmd)
 
@@ -105,7 +105,7 @@ def main():
         sources=["csrc/bind/module.cpp"],
         extra_objects=objects,
         include_dirs=[os.path.join(ROOT, "csrc/include")],
-        libraries=["hipfft"],
+        libraries=["hipfft", "roctx64"],
         extra_compile_args={"cxx": ["-O2", "-std=c++17"], "nvcc": ["-O2"]},
     )
 

@@ -50,11 +50,13 @@ def make_engine(C, cfg, **kw):
     return C.PipelineEngine(**args)
 
 
-@pytest.mark.parametrize("n,s", [(1 << 18, 1 << 6), (1 << 20, 1 << 6)])
-def test_engine_matches_cpu_oracle(C, n, s):
-    """Second shape has waterfall len 2^13 > 4096 → exercises the multi-pass
+@pytest.mark.parametrize("n,s,fft", [(1 << 18, 1 << 6, 0),
+                                      (1 << 20, 1 << 6, 0),
+                                      (1 << 18, 1 << 6, 1)])
+def test_engine_matches_cpu_oracle(C, n, s, fft):
+    """Shape 2 has waterfall len 2^13 > 4096 → exercises the multi-pass
     native FFT with the RFI+dedispersion preop fused into its first column
-    pass."""
+    pass; variant 3 runs the hipFFT fallback backend."""
     cfg = small_cfg()
     cfg.baseband_input_count = n
     cfg.spectrum_channel_count = s
@@ -68,7 +70,8 @@ def test_engine_matches_cpu_oracle(C, n, s):
     res_cpu = CpuPipeline(cfg).process_block(raw)
     eng = make_engine(C, cfg,
                       rfi_threshold=cfg.mitigate_rfi_average_method_threshold,
-                      sk_threshold=cfg.mitigate_rfi_spectral_kurtosis_threshold)
+                      sk_threshold=cfg.mitigate_rfi_spectral_kurtosis_threshold,
+                      fft_backend=fft)
     slot = eng.submit(torch.from_numpy(raw.copy()))
     res = eng.wait(slot)
 
