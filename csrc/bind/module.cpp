@@ -284,7 +284,9 @@ std::vector<torch::Tensor> t_count_signal(torch::Tensor ts, double snr) {
 torch::Tensor t_inclusive_scan(torch::Tensor ts) {
   CHECK_CUDA(ts);
   auto out = torch::empty_like(ts);
-  auto scratch = torch::empty({4096}, ts.options());
+  auto scratch = torch::empty({std::max<int64_t>(4096,
+                                  scan_scratch_size(ts.numel()))},
+                              ts.options());
   check(inclusive_scan(ts.data_ptr<float>(), out.data_ptr<float>(),
                        ts.numel(), scratch.data_ptr<float>(), cur_stream()),
         "inclusive_scan");

@@ -87,7 +87,10 @@ PipelineEngine::PipelineEngine(const EngineConfig& cfg, int n_slots)
               "ts partial alloc");
     check_hip(hipMalloc(&s.cumsum, ts_count_ * sizeof(float)), "cumsum alloc");
     check_hip(hipMalloc(&s.box, ts_count_ * sizeof(float)), "box alloc");
-    check_hip(hipMalloc(&s.scan_scratch, 4096 * sizeof(float)), "scan alloc");
+    check_hip(hipMalloc(&s.scan_scratch,
+                        std::max(4096, scan_scratch_size(ts_count_)) *
+                            sizeof(float)),
+              "scan alloc");
     check_hip(hipMalloc(&s.partials, np * sizeof(double)), "partials alloc");
     check_hip(hipMalloc(&s.mean_power, sizeof(double)), "mean alloc");
     check_hip(hipMalloc(&s.sums, 2 * sizeof(double)), "sums alloc");

@@ -147,7 +147,9 @@ hipError_t count_above(const float* ts, size_t n, const double* sumsq,
                        float snr, unsigned* out_count, float* out_threshold,
                        hipStream_t stream);
 
-// Inclusive prefix sum (float in → float out); scratch: >= 2048 floats.
+// Inclusive prefix sum (float in → float out); scratch must hold
+// scan_scratch_size(n) floats (one per 2048-element block).
+int scan_scratch_size(size_t n);
 hipError_t inclusive_scan(const float* in, float* out, size_t n,
                           float* scratch, hipStream_t stream);
 

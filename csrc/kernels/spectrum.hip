@@ -698,11 +698,17 @@ hipError_t count_above(const float* ts, size_t n, const double* sumsq,
   return hipSuccess;
 }
 
+int scan_scratch_size(size_t n) {
+  const size_t per_block = (size_t)kBlock * kScanItems;
+  return (int)((n + per_block - 1) / per_block);
+}
+
 hipError_t inclusive_scan(const float* in, float* out, size_t n,
                           float* scratch, hipStream_t stream) {
   const size_t per_block = (size_t)kBlock * kScanItems;
   const size_t nb = (n + per_block - 1) / per_block;
-  if (nb > 2048) return hipErrorInvalidValue;  // 4M elements max; ts ~ 2^18
+  // the block-sums pass is one workgroup scanning nb partials serially per
+  // thread chunk — correct for any nb (scratch must hold nb floats)
   hipLaunchKernelGGL(k_scan_blocks, dim3((uint32_t)nb), dim3(kBlock), 0,
                      stream, in, out, scratch, n);
   hipLaunchKernelGGL(k_scan_block_sums, dim3(1), dim3(kBlock), 0, stream,

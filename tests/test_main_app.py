@@ -187,3 +187,24 @@ def test_main_udp_two_endpoints(tmp_path):
         "--device", "cpu", "--max-blocks", "4"])
     t.join()
     assert rc == 0
+
+
+def test_main_with_shipped_j1644_config(tmp_path):
+    """The shipped configs/srtb_config_1644-4559.cfg loads through the real
+    path (BASELINE config 1 plumbing) with size overridden for CPU speed."""
+    import os
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    rng = np.random.default_rng(3)
+    n = 1 << 18
+    raw = rng.integers(0, 256, n * 2 // 8, dtype=np.uint8)  # 2-bit packed
+    rec = tmp_path / "j.bin"
+    raw.tofile(rec)
+    rc = main(["--config_file_name",
+               os.path.join(root, "configs", "srtb_config_1644-4559.cfg"),
+               "--baseband_input_count", str(n),
+               "--spectrum_channel_count", "2 ** 6",
+               "--input_file_path", str(rec),
+               "--baseband_output_file_prefix", str(tmp_path) + "/j_",
+               "--gui_enable", "0",
+               "--device", "cpu"])
+    assert rc == 0
