@@ -41,14 +41,24 @@ def main():
     cfg.dm = -478.80
     print("synthesizing dispersed pulse (DM -478.80, 2-bit)...", flush=True)
     rng = np.random.default_rng(1644)
-    t_pulse = 0.55 * args.n / cfg.baseband_sample_rate
-    pulse_block = synthesize_dispersed_pulse(cfg, t_pulse, pulse_amp=6.0,
+    # the pulse must land in the VALID window of its replay block: with
+    # overlap on, only the first (N - nsamps_reserved) samples of a block
+    # survive truncation (~65% here), and block 2 starts at N - reserved
+    from srtb_amd import ref as _ref
+    reserved = _ref.nsamps_reserved(args.n, args.channels, 1437.0, -64.0,
+                                    128e6, -478.80)
+    t_pulse = 0.25 * (args.n - reserved) / cfg.baseband_sample_rate
+    pulse_block = synthesize_dispersed_pulse(cfg, t_pulse, pulse_amp=12.0,
                                              noise_sigma=2.0, rng=rng)
     lv = rng.integers(0, 4, args.n, dtype=np.uint8).reshape(-1, 4)
     noise_block = ((lv[:, 0] << 6) | (lv[:, 1] << 4) | (lv[:, 2] << 2)
                    | lv[:, 3]).astype(np.uint8)
     rec = os.path.join(args.out, "j1644_synth.bin")
-    np.concatenate([noise_block, pulse_block]).tofile(rec)
+    # layout: noise block, then the pulse block aligned to replay block 2's
+    # start (= N - reserved samples in)
+    step_bytes = (args.n - reserved) * 2 // 8
+    buf = np.concatenate([noise_block[:step_bytes], pulse_block])
+    buf.tofile(rec)
 
     print("replaying through bin/srtb-backend ...", flush=True)
     out = subprocess.run(
