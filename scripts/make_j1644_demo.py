@@ -48,7 +48,7 @@ def main():
     reserved = _ref.nsamps_reserved(args.n, args.channels, 1437.0, -64.0,
                                     128e6, -478.80)
     t_pulse = 0.25 * (args.n - reserved) / cfg.baseband_sample_rate
-    pulse_block = synthesize_dispersed_pulse(cfg, t_pulse, pulse_amp=12.0,
+    pulse_block = synthesize_dispersed_pulse(cfg, t_pulse, pulse_amp=1.0,
                                              noise_sigma=2.0, rng=rng)
     lv = rng.integers(0, 4, args.n, dtype=np.uint8).reshape(-1, 4)
     noise_block = ((lv[:, 0] << 6) | (lv[:, 1] << 4) | (lv[:, 2] << 2)
