@@ -109,6 +109,11 @@ PipelineEngine::PipelineEngine(const EngineConfig& cfg, int n_slots)
     if (native_fft_) {
       s.nfwd.plan(nc_, 1, -1, s.stream);
       s.nbwd.plan(l_, s_, +1, s.stream);
+      const int wpr = s.nbwd.dif_sk_wgs_per_row();
+      if (cfg.enable_sk && wpr > 0)
+        check_hip(hipMalloc(&s.sk_dif_partials,
+                            s_ * (size_t)wpr * sizeof(float2)),
+                  "sk dif partials");
       check_hip(hipStreamSynchronize(s.stream), "fft table sync");
     } else {
       s.plans.create(n_, l_, s_, s.stream);
@@ -141,6 +146,7 @@ PipelineEngine::~PipelineEngine() {
     hipFree(s.samples);
     hipFree(s.spec);
     hipFree(s.s2s4);
+    if (s.sk_dif_partials) hipFree(s.sk_dif_partials);
     hipFree(s.flags);
     hipFree(s.ts);
     hipFree(s.ts_partial);
@@ -222,7 +228,8 @@ void PipelineEngine::enqueue_chain(Slot& s, const uint8_t* dev_raw,
     pre.f_c = f_c_;
     pre.df = df_;
     pre.dm = dm;
-    s.nbwd.exec(s.spec, reinterpret_cast<float2*>(s.samples), st, &pre);
+    s.nbwd.exec(s.spec, reinterpret_cast<float2*>(s.samples), st, &pre,
+                s.sk_dif_partials);
     wf = reinterpret_cast<float2*>(s.samples);
   } else {
     check_hip(rfi_dedisperse_fused(
@@ -246,8 +253,14 @@ void PipelineEngine::enqueue_chain(Slot& s, const uint8_t* dev_raw,
 
   const uint8_t* ts_flags = nullptr;
   if (cfg_.enable_sk) {
-    // 6. spectral kurtosis: row stats → flags (+zero count) → zap rows
-    check_hip(sk_row_stats(wf, s_, l_, s.s2s4, st), "sk stats");
+    // 6. spectral kurtosis: row stats (fused into the backward DIF store
+    //    when available — saves the 4 GB re-read) → flags → zap rows
+    if (fuse_into_bwd && s.sk_dif_partials)
+      check_hip(sk_combine_partials(s.sk_dif_partials, s_,
+                                    s.nbwd.dif_sk_wgs_per_row(), s.s2s4, st),
+                "sk combine");
+    else
+      check_hip(sk_row_stats(wf, s_, l_, s.s2s4, st), "sk stats");
     check_hip(sk_flags(wf, s.s2s4, s_, l_, sk_lo_, sk_hi_, s.flags,
                        s.counters + 0, st),
               "sk flags");

@@ -149,6 +149,7 @@ class PipelineEngine {
     float* samples = nullptr;       // [N] unpacked
     float2* spec = nullptr;         // [Nc+1] spectrum / waterfall (in-place)
     float2* s2s4 = nullptr;         // [S]
+    float2* sk_dif_partials = nullptr;  // [S * wgs_per_row] (fused SK stats)
     uint8_t* flags = nullptr;       // [S]
     float* ts = nullptr;            // [ts_count]
     float* ts_partial = nullptr;    // [chunks][ts_count] two-stage scratch

@@ -168,11 +168,24 @@ class NativeFft {
     return !passes_.empty() && passes_[0].kind == PassKind::kCol;
   }
 
+  // SK-stat accumulation can fuse into the final DIF pass when each
+  // workgroup stays within one batch row; returns workgroups per row
+  // (partials buffer = batch * wgs_per_row float2), or 0 if unsupported.
+  int dif_sk_wgs_per_row() const {
+    if (passes_.empty() || passes_.back().kind != PassKind::kDif) return 0;
+    const Pass& p = passes_.back();
+    const size_t per_row = (len_ / p.dif.n);  // instances per batch row
+    const int F = dif_f(p);
+    if (per_row % F != 0) return 0;
+    return (int)(per_row / F);
+  }
+
   // Execute the planned transform.  out may equal in only for 1-pass plans.
   // preop (optional) is applied to the FIRST pass's loads (requires
   // first_pass_fusable()).
   void exec(const float2* in, float2* out, hipStream_t stream,
-            const FftPreop* preop = nullptr) {
+            const FftPreop* preop = nullptr,
+            float2* dif_sk_partials = nullptr) {
     if (passes_.empty()) throw std::runtime_error("NativeFft: not planned");
     if (passes_.size() > 1 && in == out)
       throw std::runtime_error("NativeFft: multi-pass needs out != in");
@@ -200,14 +213,9 @@ class NativeFft {
           dst = cur;  // in place
           break;
         case PassKind::kDif: {
-          int F = 32;
-          while (F > 1 &&
-                 ((size_t)p.dif.n + (size_t)F * (p.dif.n + 2)) *
-                         sizeof(float2) > 160 * 1024)
-            F >>= 1;
-          while (F > 1 && p.n_ffts % F != 0) F >>= 1;
+          const int F = dif_f(p);
           check_hip(fft_dif_final(cur, dst, p.dif, p.n_ffts, F, sign_,
-                                  p.tw_n, stream),
+                                  p.tw_n, dif_sk_partials, stream),
                     "fft_dif_final");
           break;
         }
@@ -246,6 +254,16 @@ class NativeFft {
     int t = 0;
     while ((1ull << t) < v) ++t;
     return t;
+  }
+
+  int dif_f(const Pass& p) const {
+    int F = 32;
+    while (F > 1 &&
+           ((size_t)p.dif.n + (size_t)F * (p.dif.n + 2)) * sizeof(float2) >
+               160 * 1024)
+      F >>= 1;
+    while (F > 1 && p.n_ffts % F != 0) F >>= 1;
+    return F;
   }
 
   int pick_f(uint32_t n, const Pass& p, size_t n_ffts) const {

@@ -219,9 +219,20 @@ struct DifFinalDesc {
   int pf_bits[4] = {0, 0, 0, 0};
   unsigned long long pf_coef[4] = {0, 0, 0, 0};
 };
+// sk_partials (optional): DifFinalDesc rows own whole batch rows when
+// (n_ffts/batch) % F == 0; each workgroup then accumulates <sum|x|^2,
+// sum|x|^4> of its stored elements into
+// sk_partials[row * wgs_per_row + wg_in_row] (deterministic, no atomics) —
+// this replaces the separate 4 GB sk_row_stats read.  wgs_per_row =
+// (L/n)/F.
 hipError_t fft_dif_final(const float2* in, float2* out, const DifFinalDesc& d,
                          size_t n_ffts, int F, int sign, const float2* tw_n,
-                         hipStream_t stream);
+                         float2* sk_partials, hipStream_t stream);
+
+// combine the DIF-written partials into per-row <S2,S4>
+hipError_t sk_combine_partials(const float2* partials, size_t rows,
+                               int wgs_per_row, float2* s2s4,
+                               hipStream_t stream);
 
 // packed-real R2C finish: Z = C2C(x_even + i*x_odd) of length m -> true
 // spectrum X[0..m) (Nyquist dropped).  In-place safe (x may alias z).

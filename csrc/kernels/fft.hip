@@ -437,10 +437,11 @@ __device__ inline unsigned rev4_bits(unsigned k, int t2 /* log2(n) */) {
   return ((r & 0x55555555u) << 1) | ((r & 0xAAAAAAAAu) >> 1);
 }
 
-template <int SIGN>
+template <int SIGN, bool SK>
 __global__ void __launch_bounds__(256)
     k_fft_dif_final(const float2* __restrict__ in, float2* __restrict__ out,
-                    DifFinalDescDev d, const float2* __restrict__ tw_n) {
+                    DifFinalDescDev d, const float2* __restrict__ tw_n,
+                    float2* __restrict__ sk_partials, int wgs_per_row) {
   extern __shared__ float2 lds[];
   const int n = d.n;
   const int nl = d.n_log2;
@@ -524,15 +525,47 @@ __global__ void __launch_bounds__(256)
   }
 
   // ---- store: digit-reversed LDS read, contiguous-lane output runs ----
+  float sk2 = 0.0f, sk4 = 0.0f;
   for (int e = threadIdx.x; e < total; e += blockDim.x) {
     const int f = e & (F - 1);
     const int k = e >> d.f_log2;
     unsigned long long in_blk, obase;
     dif_addr(fft0 + f, d, in_blk, obase);
     const unsigned rk = rev4_bits((unsigned)k, nl);
-    out[obase + (unsigned long long)k * d.out_elem_coef] =
-        X[f * ldst + dif_swz((int)rk)];
+    const float2 v = X[f * ldst + dif_swz((int)rk)];
+    out[obase + (unsigned long long)k * d.out_elem_coef] = v;
+    if constexpr (SK) {
+      const float p = v.x * v.x + v.y * v.y;
+      sk2 += p;
+      sk4 += p * p;
+    }
   }
+  if constexpr (SK) {
+    // all F instances of this WG belong to one batch row (the planner
+    // guarantees (L/n) % F == 0); deterministic per-WG partial, no atomics
+    const float b2 = block_reduce_sum(sk2);
+    const float b4 = block_reduce_sum(sk4);
+    if (threadIdx.x == 0) {
+      const unsigned long long row = fft0 >> d.j_bits;
+      const unsigned long long wg_in_row =
+          (fft0 & ((1ull << d.j_bits) - 1)) >> d.f_log2;
+      sk_partials[row * wgs_per_row + wg_in_row] = make_float2(b2, b4);
+    }
+  }
+}
+
+__global__ void k_sk_combine_partials(const float2* __restrict__ partials,
+                                      size_t rows, int wgs_per_row,
+                                      float2* __restrict__ s2s4) {
+  const size_t r = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (r >= rows) return;
+  float s2 = 0.0f, s4 = 0.0f;
+  for (int w = 0; w < wgs_per_row; ++w) {
+    const float2 p = partials[r * wgs_per_row + w];
+    s2 += p.x;
+    s4 += p.y;
+  }
+  s2s4[r] = make_float2(s2, s4);
 }
 
 // twiddle-table builder (fp64 on device)
@@ -745,7 +778,8 @@ hipError_t fft_col_pass(const float2* in, float2* out, const FftPassDesc& hd,
 
 hipError_t fft_dif_final(const float2* in, float2* out,
                          const DifFinalDesc& hd, size_t n_ffts, int F,
-                         int sign, const float2* tw_n, hipStream_t stream) {
+                         int sign, const float2* tw_n, float2* sk_partials,
+                         hipStream_t stream) {
   if (hd.n & (hd.n - 1)) return hipErrorInvalidValue;
   DifFinalDescDev d;
   d.n = hd.n;
@@ -767,12 +801,38 @@ hipError_t fft_dif_final(const float2* in, float2* out,
   const size_t lds_bytes =
       ((size_t)hd.n + (size_t)F * (hd.n + 2)) * sizeof(float2);
   if (lds_bytes > 160 * 1024) return hipErrorInvalidValue;
-  if (sign < 0)
-    hipLaunchKernelGGL((k_fft_dif_final<-1>), dim3(grid), dim3(256),
-                       lds_bytes, stream, in, out, d, tw_n);
-  else
-    hipLaunchKernelGGL((k_fft_dif_final<1>), dim3(grid), dim3(256), lds_bytes,
-                       stream, in, out, d, tw_n);
+  int wgs_per_row = 0;
+  if (sk_partials) {
+    const unsigned long long per_row = 1ull << d.j_bits;  // instances/row
+    if (per_row % F != 0) return hipErrorInvalidValue;
+    wgs_per_row = (int)(per_row / F);
+  }
+  if (sign < 0) {
+    if (sk_partials)
+      hipLaunchKernelGGL((k_fft_dif_final<-1, true>), dim3(grid), dim3(256),
+                         lds_bytes, stream, in, out, d, tw_n, sk_partials,
+                         wgs_per_row);
+    else
+      hipLaunchKernelGGL((k_fft_dif_final<-1, false>), dim3(grid), dim3(256),
+                         lds_bytes, stream, in, out, d, tw_n, nullptr, 0);
+  } else {
+    if (sk_partials)
+      hipLaunchKernelGGL((k_fft_dif_final<1, true>), dim3(grid), dim3(256),
+                         lds_bytes, stream, in, out, d, tw_n, sk_partials,
+                         wgs_per_row);
+    else
+      hipLaunchKernelGGL((k_fft_dif_final<1, false>), dim3(grid), dim3(256),
+                         lds_bytes, stream, in, out, d, tw_n, nullptr, 0);
+  }
+  SRTB_CHECK_LAUNCH();
+  return hipSuccess;
+}
+
+hipError_t sk_combine_partials(const float2* partials, size_t rows,
+                               int wgs_per_row, float2* s2s4,
+                               hipStream_t stream) {
+  hipLaunchKernelGGL(k_sk_combine_partials, grid_for(rows), dim3(kBlock), 0,
+                     stream, partials, rows, wgs_per_row, s2s4);
   SRTB_CHECK_LAUNCH();
   return hipSuccess;
 }
