@@ -325,6 +325,7 @@ __global__ void k_time_series_partial(const float2* __restrict__ wf,
   if (j2 >= ts_count) return;
   const bool pair = (j2 + 1 < ts_count);
   float acc0 = 0.0f, acc1 = 0.0f;
+#pragma unroll 4
   for (size_t i = r0; i < r1; ++i) {
     if (kFlags && flags[i]) continue;
     const float2* rp = wf + i * len + j2;

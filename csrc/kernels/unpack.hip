@@ -31,6 +31,7 @@ __global__ void k_unpack_subbyte(const uint32_t* __restrict__ in,
   constexpr int per_word = 4 * per_byte;
   constexpr uint32_t mask = (1u << NBITS) - 1u;
   const size_t stride = (size_t)gridDim.x * blockDim.x;
+#pragma unroll 4
   for (size_t w = (size_t)blockIdx.x * blockDim.x + threadIdx.x; w < n_words;
        w += stride) {
     const uint32_t v = in[w];
