@@ -4,8 +4,9 @@
 // (1/2/4-bit MSB-first unsigned fields; 8/16/32-bit signed/unsigned casts;
 // 2-pol per-sample interleave; SNAP-1 "1 1 2 2"; GZNU A1 4-byte-word
 // deinterleave with offset-binary fix), redesigned for CDNA4:
-//  - each lane consumes one aligned uint32 (4 bytes) and emits 4..32 floats
-//    as float4 stores → 16 B/lane global writes, fully coalesced
+//  - sub-byte formats: one byte per lane (wave-contiguous float4/float2
+//    stores; see the k_unpack_subbyte note); byte casts: one uint32 per
+//    lane with a float4 store
 //  - wave64-sized blocks, grid-stride loops capped per G11.
 
 #include "common.h"

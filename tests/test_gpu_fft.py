@@ -81,3 +81,17 @@ def test_native_rfft_j1644_scale(C):
     err = (out - ref).abs().max().item()
     scale = ref.abs().max().item()
     assert err < 2e-4 * scale
+
+
+@pytest.mark.parametrize("maxcol", ["4", "8", "16", "64"])
+def test_fft_forced_column_widths(C, maxcol, monkeypatch):
+    """Every register-column width must agree with torch.fft — the balanced
+    default factorization otherwise leaves N=64 (used by the 2^30 bench
+    path) and small widths untested."""
+    monkeypatch.setenv("SRTB_FFT_MAXCOL", maxcol)
+    n = 1 << 20  # rest 12 bits over the 256 final
+    x = rand_c64((2, n), seed=int(maxcol))
+    xt = torch.from_numpy(x).cuda()
+    out = C.native_fft(xt, -1).cpu().numpy()
+    ref_t = torch.fft.fft(torch.from_numpy(x).cuda(), dim=1).cpu().numpy()
+    assert rel_err(out, ref_t) < 1e-4
