@@ -1,0 +1,113 @@
+#!/usr/bin/env python3
+"""Stability soak: run the PipelineEngine continuously for --seconds on the
+J1644-4559 config and assert host RSS and device VRAM stay flat (no leaks,
+no slow allocator growth, no stream/event exhaustion).
+
+All engine device/pinned buffers are allocated at construction, so both
+curves must be flat after warmup.  Run on a GPU box:
+
+    python scripts/soak.py --seconds 240
+"""
+
+import argparse
+import os
+import subprocess
+import sys
+import time
+
+import numpy as np
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+sys.path.insert(0, ROOT)
+
+
+def rss_mb():
+    with open("/proc/self/status") as f:
+        for line in f:
+            if line.startswith("VmRSS"):
+                return int(line.split()[1]) / 1024.0
+    return 0.0
+
+
+def vram_mb():
+    """Used VRAM of GPU 0 via rocm-smi (engine uses raw hipMalloc, so
+    torch.cuda.memory_allocated doesn't see it)."""
+    try:
+        out = subprocess.run(
+            ["rocm-smi", "--showmeminfo", "vram", "--csv"],
+            capture_output=True, text=True, timeout=30).stdout
+        for line in out.splitlines():
+            if line.startswith("card"):
+                return int(line.split(",")[2]) / (1024 * 1024)
+    except Exception:
+        pass
+    return 0.0
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--seconds", type=int, default=240)
+    ap.add_argument("--n", type=int, default=2**30)
+    ap.add_argument("--channels", type=int, default=2**11)
+    ap.add_argument("--slots", type=int, default=4)
+    args = ap.parse_args()
+
+    import torch
+    from srtb_amd.ops import native
+    C = native()
+    torch.cuda.set_device(0)
+
+    eng = C.PipelineEngine(
+        n=args.n, nbits=2, channels=args.channels, freq_low=1437.0,
+        bandwidth=-64.0, sample_rate=128e6, dm=-478.80,
+        rfi_threshold=1.5, sk_threshold=1.05, snr_threshold=8.0,
+        max_boxcar=256, nsamps_reserved=0, zap_ranges=[[480, 608]],
+        n_slots=args.slots)
+
+    rng = np.random.default_rng(0)
+    pinned = [torch.from_numpy(
+        rng.integers(0, 256, eng.raw_bytes, dtype=np.uint8)).pin_memory()
+        for _ in range(args.slots)]
+
+    # warmup
+    slots = [eng.submit(pinned[i]) for i in range(args.slots)]
+    for s in slots:
+        eng.wait(s)
+
+    rss0, vram0 = rss_mb(), vram_mb()
+    print(f"after warmup: RSS {rss0:.1f} MB, VRAM {vram0:.1f} MB", flush=True)
+
+    t0 = time.time()
+    t_report = t0
+    blocks = 0
+    inflight = []
+    while time.time() - t0 < args.seconds:
+        while len(inflight) < args.slots:
+            inflight.append(eng.submit(pinned[blocks % args.slots]))
+            blocks += 1
+        eng.wait(inflight.pop(0))
+        now = time.time()
+        if now - t_report >= 30.0:
+            r, v = rss_mb(), vram_mb()
+            rate = blocks * args.n / (now - t0) / 1e6
+            print(f"t={now - t0:6.0f}s blocks={blocks:6d} "
+                  f"rate={rate:8.0f} Msps RSS={r:8.1f} MB "
+                  f"VRAM={v:8.1f} MB", flush=True)
+            t_report = now
+    for s in inflight:
+        eng.wait(s)
+    eng.synchronize()
+
+    el = time.time() - t0
+    r1, v1 = rss_mb(), vram_mb()
+    print(f"done: {blocks} blocks in {el:.1f}s = "
+          f"{blocks * args.n / el / 1e6:.0f} Msamples/s sustained", flush=True)
+    print(f"RSS  {rss0:.1f} -> {r1:.1f} MB (delta {r1 - rss0:+.1f})")
+    print(f"VRAM {vram0:.1f} -> {v1:.1f} MB (delta {v1 - vram0:+.1f})")
+    ok = (r1 - rss0) < 64.0 and abs(v1 - vram0) < 256.0
+    print("SOAK " + ("PASS" if ok else "FAIL"))
+    sys.exit(0 if ok else 1)
+
+
+if __name__ == "__main__":
+    main()
