@@ -114,13 +114,16 @@ def main():
         return counts_dev
 
     def run_steps(k):
-        prev = None
+        # keep ALL slots in flight (submit-ahead depth = n_slots); waiting
+        # with only one block queued leaves the GPU idle during the host-side
+        # stat aggregation (~20% measured on the J1644 config)
+        inflight = []
         for i in range(k):
-            slot = eng.submit(pinned[i % len(pinned)])
-            if prev is not None:
-                agg(eng.wait(prev))
-            prev = slot
-        agg(eng.wait(prev))
+            inflight.append(eng.submit(pinned[i % len(pinned)]))
+            if len(inflight) >= args.slots:
+                agg(eng.wait(inflight.pop(0)))
+        while inflight:
+            agg(eng.wait(inflight.pop(0)))
         eng.synchronize()
 
     # warmup
