@@ -277,12 +277,13 @@ __device__ inline void bfly4(float2& a, float2& b, float2& c, float2& d,
   d = make_float2(t1.x - t3.x, t1.y - t3.y);
 }
 
-// one DIT stage over the register array; L = output block size, radix r
-template <int N, int L, int R, int SIGN>
+// one DIT stage over the register array; L = output block size, radix r.
+// TS overrides the twiddle index scale when the register array holds only a
+// fraction of the full column (the lane-pair kernel below).
+template <int N, int L, int R, int SIGN, int TS = N / L>
 __device__ inline void col_stage(float2 (&v)[N],
                                  const float2* __restrict__ tw_n) {
-  constexpr int M = L / R;
-  constexpr int TS = N / L;  // tw index scale: W_L^x = tw_n[x*TS]
+  constexpr int M = L / R;  // TS: W_L^x = tw_n[x*TS]
 #pragma unroll
   for (int g = 0; g < N; g += L) {
 #pragma unroll
@@ -384,6 +385,115 @@ __global__ void __launch_bounds__(256)
       r = cmulf(r, tw_eval(m_, d.tw_angle));
     }
     ocolbase[(uint32_t)k * stride32] = r;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// N = 64 column pass, two lanes per column.  The mono-column k_fft_col<64>
+// holds 64 complex elements in registers (~400 VGPRs → 1 wave/SIMD); here
+// lane p = tid&1 of an adjacent lane pair holds post-σ positions
+// 32p..32p+31, so stages 1-2 (L=4, L=16) are lane-local, and the final
+// L=64 stage exchanges halves with two shfl_xor(…, 1) per butterfly (a
+// quad-perm DPP swap on gfx950 — no LDS).  Both lanes evaluate the full
+// radix-4 butterfly from the exchanged quartet and keep their own two
+// outputs.  32 data registers per lane → ~3 waves/SIMD, and per-instruction
+// global accesses stay coalesced (each half-wave touches 32 consecutive
+// columns: 2 × 256 B runs).
+// ---------------------------------------------------------------------------
+
+constexpr int col_sigma_inv64(int pos) {
+  for (int i = 0; i < 64; ++i)
+    if (col_sigma<64>(i) == pos) return i;
+  return 0;
+}
+
+__device__ inline float2 shfl_xor1(float2 x) {
+  return make_float2(__shfl_xor(x.x, 1, 64), __shfl_xor(x.y, 1, 64));
+}
+
+template <bool TWIDDLE, int SIGN, bool PREOP>
+__global__ void __launch_bounds__(256)
+    k_fft_col_pair64(const float2* __restrict__ in, float2* __restrict__ out,
+                     FftPassDescDev d, unsigned long long n_ffts,
+                     const float2* __restrict__ tw_n,
+                     const float2* __restrict__ tw_hi,
+                     const float2* __restrict__ tw_lo, FftPreopDev pre) {
+  const unsigned long long tid =
+      (unsigned long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const unsigned long long id = tid >> 1;  // column index
+  const int p = (int)(tid & 1);            // which half of the column
+  if (id >= n_ffts) return;
+  unsigned long long q0, q1, q2;
+  digits(id, d, q0, q1, q2);
+  const unsigned long long base = q0 * d.in_c0 + q1 * d.in_c1 + q2 * d.in_c2;
+  float thr_mean = 0.f;
+  if constexpr (PREOP) {
+    if (pre.mean_power) thr_mean = pre.threshold * (float)(*pre.mean_power);
+  }
+  float2 v[32];
+  const float2* __restrict__ colbase = in + base;
+  const uint32_t stride32 = (uint32_t)d.in_stride;
+#pragma unroll
+  for (int r = 0; r < 32; ++r) {
+    // position 32p + r holds element σ⁻¹(32p + r); both candidates are
+    // compile-time constants, selected per-lane
+    const int i = p ? col_sigma_inv64(32 + r) : col_sigma_inv64(r);
+    const uint32_t off = (uint32_t)i * stride32;
+    float2 x = colbase[off];
+    if constexpr (PREOP) {
+      const unsigned long long flat = base + off;
+      bool zap = pre.mean_power && (norm2(x) > thr_mean);
+      for (int z = 0; z < pre.n_zap; ++z)
+        zap |= (flat >= pre.zap[z].lo) & (flat <= pre.zap[z].hi);
+      if (zap) {
+        x = make_float2(0.f, 0.f);
+      } else {
+        if (pre.mean_power) {
+          x.x *= pre.norm_coeff;
+          x.y *= pre.norm_coeff;
+        }
+        x = cmulf(x, srtb_dedisp_factor(flat, pre.f_min, pre.f_c, pre.df,
+                                        pre.dm));
+      }
+    }
+    v[r] = x;
+  }
+  // stages 1-2 are lane-local; the twiddle scale is that of the FULL
+  // length-64 transform (TS = 64/L), not of the 32-element register array
+  col_stage<32, 4, 4, SIGN, 16>(v, tw_n);
+  col_stage<32, 16, 4, SIGN, 4>(v, tw_n);
+  // stage 3: quartets {j, j+16, j+32, j+48}, TS = 1
+#pragma unroll
+  for (int j = 0; j < 16; ++j) {
+    const float2 ea = shfl_xor1(v[j]);
+    const float2 eb = shfl_xor1(v[16 + j]);
+    float2 a = p ? ea : v[j];
+    float2 b = p ? eb : v[16 + j];
+    float2 c = p ? v[j] : ea;
+    float2 e = p ? v[16 + j] : eb;
+    float2 w1, w2, w3;
+    if (j == 0) {
+      w1 = w2 = w3 = make_float2(1.f, 0.f);
+    } else {
+      w1 = tw_n[j];
+      w2 = tw_n[2 * j];
+      w3 = tw_n[3 * j];
+    }
+    bfly4<SIGN>(a, b, c, e, w1, w2, w3);
+    v[j] = p ? c : a;
+    v[16 + j] = p ? e : b;
+  }
+  float2* __restrict__ ocolbase = out + base;
+#pragma unroll
+  for (int r = 0; r < 32; ++r) {
+    const int k = 32 * p + r;
+    float2 rr = v[r];
+    if constexpr (TWIDDLE) {
+      const unsigned long long m_ =
+          (q0 * d.tw_f0 * (unsigned long long)k) & d.tw_mask;
+      rr = cmulf(rr, tw_eval(m_, d.tw_angle));
+    }
+    ocolbase[(uint32_t)k * stride32] = rr;
   }
 }
 
@@ -760,18 +870,36 @@ hipError_t fft_col_pass(const float2* in, float2* out, const FftPassDesc& hd,
                       else COL_LAUNCH(N, false, 1, false); }                 \
     }                                                                        \
     break;
+  // N = 64 uses the lane-pair kernel (two threads per column)
+  const uint32_t grid2 = (uint32_t)((2 * n_ffts + 255) / 256);
+#define COL_LAUNCH_P(TW, SG, PR)                                             \
+  hipLaunchKernelGGL((k_fft_col_pair64<TW, SG, PR>), dim3(grid2), dim3(256), \
+                     0, stream, in, out, d, n_ffts, tw_n, tw_hi, tw_lo, pre)
   switch (hd.n) {
     COL_DISPATCH(2)
     COL_DISPATCH(4)
     COL_DISPATCH(8)
     COL_DISPATCH(16)
     COL_DISPATCH(32)
-    COL_DISPATCH(64)
+    case 64:
+      if (twiddle) {
+        if (sign < 0) { if (preop) COL_LAUNCH_P(true, -1, true);
+                        else COL_LAUNCH_P(true, -1, false); }
+        else          { if (preop) COL_LAUNCH_P(true, 1, true);
+                        else COL_LAUNCH_P(true, 1, false); }
+      } else {
+        if (sign < 0) { if (preop) COL_LAUNCH_P(false, -1, true);
+                        else COL_LAUNCH_P(false, -1, false); }
+        else          { if (preop) COL_LAUNCH_P(false, 1, true);
+                        else COL_LAUNCH_P(false, 1, false); }
+      }
+      break;
     default:
       return hipErrorInvalidValue;
   }
 #undef COL_DISPATCH
 #undef COL_LAUNCH
+#undef COL_LAUNCH_P
   SRTB_CHECK_LAUNCH();
   return hipSuccess;
 }
