@@ -64,6 +64,70 @@ __global__ void k_unpack_subbyte(const uint8_t* __restrict__ in,
   }
 }
 
+// ---- sub-byte unpack, wave-block variant: dense dword loads + __shfl ----
+// Byte-per-lane above fixed the STORE side but still loads one byte per
+// lane (64 B touched per 64-lane load instruction).  Here each wave owns a
+// 256-byte block per iteration: every lane loads one uint32 (dense 256 B
+// run), then __shfl redistributes so each lane decodes the byte whose
+// OUTPUT vector it stores — both loads and stores are wave-contiguous.
+template <int NBITS, bool kWindow>
+__global__ void k_unpack_subbyte_w(const uint32_t* __restrict__ in,
+                                   float* __restrict__ out, size_t n_blocks,
+                                   const float* __restrict__ window) {
+  constexpr int per_byte = 8 / NBITS;
+  constexpr uint32_t mask = (1u << NBITS) - 1u;
+  const int lane = threadIdx.x & 63;
+  const size_t wave = ((size_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const size_t wstride = ((size_t)gridDim.x * blockDim.x) >> 6;
+  for (size_t blk = wave; blk < n_blocks; blk += wstride) {
+    const uint32_t myw = in[blk * 64 + lane];
+    const size_t sample_base = blk * (256 * per_byte);
+    if constexpr (NBITS == 1) {
+      // float4-centric: float4 fi covers half a byte; 512 float4s per block
+#pragma unroll
+      for (int r = 0; r < 8; ++r) {
+        const int fi = 64 * r + lane;
+        const uint32_t dw = __shfl(myw, 8 * r + (lane >> 3), 64);
+        const uint32_t bv = (dw >> (8 * ((lane >> 1) & 3))) & 0xffu;
+        const int half = lane & 1;  // low or high 4 samples of the byte
+        const size_t sbase = sample_base + (size_t)fi * 4;
+        float4 o;
+        float* op = &o.x;
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          const int s = half * 4 + i;  // MSB-first within the byte
+          op[i] = wmul<kWindow>(window, sbase + i,
+                                (float)((bv >> (7 - s)) & 1u));
+        }
+        reinterpret_cast<float4*>(out + sbase)[0] = o;
+      }
+    } else {
+      // byte-centric: 256 bytes per block, 4 rounds of 64
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int bi = 64 * r + lane;
+        const uint32_t dw = __shfl(myw, 16 * r + (lane >> 2), 64);
+        const uint32_t bv = (dw >> (8 * (lane & 3))) & 0xffu;
+        const size_t sbase = sample_base + (size_t)bi * per_byte;
+        if constexpr (NBITS == 2) {
+          float4 o;
+          float* op = &o.x;
+#pragma unroll
+          for (int i = 0; i < 4; ++i)
+            op[i] = wmul<kWindow>(window, sbase + i,
+                                  (float)((bv >> ((3 - i) * 2)) & mask));
+          reinterpret_cast<float4*>(out + sbase)[0] = o;
+        } else {  // NBITS == 4
+          float2 o;
+          o.x = wmul<kWindow>(window, sbase + 0, (float)((bv >> 4) & mask));
+          o.y = wmul<kWindow>(window, sbase + 1, (float)(bv & mask));
+          reinterpret_cast<float2*>(out + sbase)[0] = o;
+        }
+      }
+    }
+  }
+}
+
 // ---- byte casts: one uint32 = 4 samples per lane ----
 template <bool kSigned, bool kWindow>
 __global__ void k_unpack_cast8(const uint32_t* __restrict__ in,
@@ -230,15 +294,40 @@ hipError_t unpack(const uint8_t* in, float* out, size_t out_count, int nbits,
     case 2:
     case 4: {
       const size_t n_bytes = out_count * nbits / 8;
-      const dim3 g = grid_for(n_bytes);
+      // main body: whole 256-byte wave blocks (dense loads); byte-per-lane
+      // tail kernel covers the remainder (only for non-multiple-of-256
+      // inputs — never on power-of-two block sizes)
+      const size_t n_blocks = n_bytes >> 8;
+      const size_t tail_off = n_blocks << 8;
+      const size_t tail = n_bytes - tail_off;
 #define CASE(B)                                                              \
   if (nbits == B) {                                                          \
-    if (w)                                                                   \
-      hipLaunchKernelGGL((k_unpack_subbyte<B, true>), g, dim3(kBlock), 0,    \
-                         stream, in, out, n_bytes, window);                  \
-    else                                                                     \
-      hipLaunchKernelGGL((k_unpack_subbyte<B, false>), g, dim3(kBlock), 0,   \
-                         stream, in, out, n_bytes, window);                  \
+    constexpr int PB = 8 / B;                                                \
+    if (n_blocks) {                                                          \
+      const dim3 g = grid_for(n_blocks * 64);                                \
+      if (w)                                                                 \
+        hipLaunchKernelGGL((k_unpack_subbyte_w<B, true>), g, dim3(kBlock),   \
+                           0, stream,                                        \
+                           reinterpret_cast<const uint32_t*>(in), out,       \
+                           n_blocks, window);                                \
+      else                                                                   \
+        hipLaunchKernelGGL((k_unpack_subbyte_w<B, false>), g, dim3(kBlock),  \
+                           0, stream,                                        \
+                           reinterpret_cast<const uint32_t*>(in), out,       \
+                           n_blocks, window);                                \
+    }                                                                        \
+    if (tail) {                                                              \
+      const dim3 gt = grid_for(tail);                                        \
+      const float* wt = w ? window + tail_off * PB : nullptr;                \
+      if (w)                                                                 \
+        hipLaunchKernelGGL((k_unpack_subbyte<B, true>), gt, dim3(kBlock), 0, \
+                           stream, in + tail_off, out + tail_off * PB, tail, \
+                           wt);                                              \
+      else                                                                   \
+        hipLaunchKernelGGL((k_unpack_subbyte<B, false>), gt, dim3(kBlock),   \
+                           0, stream, in + tail_off, out + tail_off * PB,    \
+                           tail, wt);                                        \
+    }                                                                        \
   }
       CASE(1) CASE(2) CASE(4)
 #undef CASE
