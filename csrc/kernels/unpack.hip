@@ -104,28 +104,37 @@ __global__ void k_unpack_subbyte_w(const uint32_t* __restrict__ in,
         }
         reinterpret_cast<float4*>(out + sbase)[0] = o;
       }
-    } else {
-      // byte-centric: 256 bytes per block, 4 rounds of 64
+    } else if constexpr (NBITS == 2) {
+      // byte-centric: one float4 per byte, 256 bytes per block, 4 rounds
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int bi = 64 * r + lane;
         const uint32_t dw = __shfl(myw, 16 * r + (lane >> 2), 64);
         const uint32_t bv = (dw >> (8 * (lane & 3))) & 0xffu;
         const size_t sbase = sample_base + (size_t)bi * per_byte;
-        if constexpr (NBITS == 2) {
-          float4 o;
-          float* op = &o.x;
+        float4 o;
+        float* op = &o.x;
 #pragma unroll
-          for (int i = 0; i < 4; ++i)
-            op[i] = wmul<kWindow>(window, sbase + i,
-                                  (float)((bv >> ((3 - i) * 2)) & mask));
-          reinterpret_cast<float4*>(out + sbase)[0] = o;
-        } else {  // NBITS == 4
-          float2 o;
-          o.x = wmul<kWindow>(window, sbase + 0, (float)((bv >> 4) & mask));
-          o.y = wmul<kWindow>(window, sbase + 1, (float)(bv & mask));
-          reinterpret_cast<float2*>(out + sbase)[0] = o;
-        }
+        for (int i = 0; i < 4; ++i)
+          op[i] = wmul<kWindow>(window, sbase + i,
+                                (float)((bv >> ((3 - i) * 2)) & mask));
+        reinterpret_cast<float4*>(out + sbase)[0] = o;
+      }
+    } else {
+      // NBITS == 4, float4-centric: one float4 covers two bytes, 2 rounds
+#pragma unroll
+      for (int r = 0; r < 2; ++r) {
+        const int fi = 64 * r + lane;
+        const uint32_t dw = __shfl(myw, 32 * r + (lane >> 1), 64);
+        const uint32_t b0 = (dw >> (8 * ((lane & 1) * 2))) & 0xffu;
+        const uint32_t b1 = (dw >> (8 * ((lane & 1) * 2 + 1))) & 0xffu;
+        const size_t sbase = sample_base + (size_t)fi * 4;
+        float4 o;
+        o.x = wmul<kWindow>(window, sbase + 0, (float)((b0 >> 4) & mask));
+        o.y = wmul<kWindow>(window, sbase + 1, (float)(b0 & mask));
+        o.z = wmul<kWindow>(window, sbase + 2, (float)((b1 >> 4) & mask));
+        o.w = wmul<kWindow>(window, sbase + 3, (float)(b1 & mask));
+        reinterpret_cast<float4*>(out + sbase)[0] = o;
       }
     }
   }
@@ -300,11 +309,7 @@ hipError_t unpack(const uint8_t* in, float* out, size_t out_count, int nbits,
       // main body: whole 256-byte wave blocks (dense loads); byte-per-lane
       // tail kernel covers the remainder (only for non-multiple-of-256
       // inputs — never on power-of-two block sizes)
-      // 4-bit stays byte-per-lane: its float2 stores are already dense and
-      // the byte loads only touch 0.5 GB — measured at roofline (0.66 ms
-      // for 2^30 samples) vs 1.10 ms for the wave-block variant, which
-      // halves output volume per load and becomes load-latency-bound
-      const size_t n_blocks = (nbits == 4) ? 0 : (n_bytes >> 8);
+      const size_t n_blocks = n_bytes >> 8;
       const size_t tail_off = n_blocks << 8;
       const size_t tail = n_bytes - tail_off;
 #define CASE(B)                                                              \
