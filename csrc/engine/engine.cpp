@@ -264,7 +264,11 @@ void PipelineEngine::enqueue_chain(Slot& s, const uint8_t* dev_raw,
     check_hip(sk_flags(wf, s.s2s4, s_, l_, sk_lo_, sk_hi_, s.flags,
                        s.counters + 0, st),
               "sk flags");
-    check_hip(sk_zap_rows(wf, s.flags, s_, l_, st), "sk zap");
+    // NOTE: the waterfall itself is NOT zapped here — the time-series pass
+    // below applies the flags directly, and zeroing the flagged rows of the
+    // 4 GB waterfall (~1 ms/block) only matters when the waterfall is
+    // actually read (product dump on detection, display).  waterfall_ptr()
+    // performs the zap on demand.
     ts_flags = s.flags;
   }
   // 7. time series over non-zapped rows
@@ -388,6 +392,7 @@ BlockResult PipelineEngine::wait(int slot) {
   Slot& s = *slots_.at(slot);
   check_hip(hipEventSynchronize(s.done), "wait");
   s.busy = false;
+  s.wf_zap_pending = cfg_.enable_sk;  // see waterfall_ptr()
   BlockResult r;
   r.zero_count = s.h_counters[0];
   r.counts.emplace_back(1u, s.h_counters[1]);
@@ -406,7 +411,15 @@ void PipelineEngine::synchronize() {
 
 float2* PipelineEngine::waterfall_ptr(int slot) {
   Slot& s = *slots_.at(slot);
-  return s.wf ? s.wf : s.spec;
+  float2* wf = s.wf ? s.wf : s.spec;
+  if (s.wf_zap_pending) {
+    // deferred SK row zap (flags are still resident for this slot); only
+    // blocks whose waterfall is actually consumed pay the 4 GB pass
+    check_hip(sk_zap_rows(wf, s.flags, s_, l_, s.stream), "sk zap lazy");
+    check_hip(hipStreamSynchronize(s.stream), "sk zap sync");
+    s.wf_zap_pending = false;
+  }
+  return wf;
 }
 float* PipelineEngine::time_series_ptr(int slot) { return slots_.at(slot)->ts; }
 float* PipelineEngine::cumsum_ptr(int slot) { return slots_.at(slot)->cumsum; }

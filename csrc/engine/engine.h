@@ -144,6 +144,7 @@ class PipelineEngine {
     hipStream_t stream = nullptr;
     hipEvent_t done = nullptr;
     bool busy = false;
+    bool wf_zap_pending = false;    // SK zap deferred until waterfall read
     float2* wf = nullptr;           // where the waterfall landed last block
     uint8_t* raw = nullptr;         // device raw bytes
     float* samples = nullptr;       // [N] unpacked

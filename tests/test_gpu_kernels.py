@@ -31,6 +31,24 @@ def test_unpack_matches_oracle(C, nbits):
     np.testing.assert_array_equal(out, expect)
 
 
+@pytest.mark.parametrize("nbits", [1, 2, 4])
+@pytest.mark.parametrize("n_bytes", [300, 255, 256 + 17])
+def test_unpack_subbyte_tail(C, nbits, n_bytes):
+    """Non-multiple-of-256-byte inputs take the wave-block main kernel PLUS
+    the byte-per-lane tail kernel; pow2-sized tests never exercise the
+    seam."""
+    rng = np.random.default_rng(10 * nbits + n_bytes)
+    raw = rng.integers(0, 256, n_bytes, dtype=np.uint8)
+    expect = ref.unpack(raw, nbits)
+    out = C.unpack(to_gpu(raw), nbits, expect.size).cpu().numpy()
+    np.testing.assert_array_equal(out, expect)
+    # and with a window, which the tail kernel must index absolutely
+    w = ref.window_coefficients("hann", expect.size)
+    expect_w = ref.unpack(raw, nbits, window=w)
+    out_w = C.unpack(to_gpu(raw), nbits, expect.size, to_gpu(w)).cpu().numpy()
+    np.testing.assert_allclose(out_w, expect_w, rtol=1e-6)
+
+
 def test_unpack_16bit(C):
     rng = np.random.default_rng(2)
     raw = rng.integers(0, 256, 1 << 12, dtype=np.uint8)
