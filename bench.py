@@ -44,7 +44,8 @@ def main():
                     help="cache dedispersion phase factors (fixed DM); "
                     "default recomputes fp64 phase per block like the reference")
     ap.add_argument("--no-rfi", action="store_true")
-    ap.add_argument("--fft", choices=["native", "hipfft"], default="native",
+    ap.add_argument("--fft", choices=["native", "hipfft", "auto"],
+                    default="native",
                     help="FFT backend (hand-written Stockham vs hipFFT)")
     ap.add_argument("--graph", action="store_true",
                     help="capture the per-block chain into hipGraphs")
@@ -93,7 +94,7 @@ def main():
         snr_threshold=snr, max_boxcar=max_boxcar, nsamps_reserved=0,
         zap_ranges=zap_ranges, use_phase_table=args.phase_table,
         enable_rfi_s1=not args.no_rfi, enable_sk=True, n_slots=args.slots,
-        fft_backend=0 if args.fft == "native" else 1,
+        fft_backend={"native": 0, "hipfft": 1, "auto": 2}[args.fft],
         use_hip_graph=args.graph)
 
     # synthetic 2-bit baseband noise, pinned, one buffer per slot

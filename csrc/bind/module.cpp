@@ -657,7 +657,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
            py::arg("zap_ranges") = std::vector<std::vector<int64_t>>{},
            py::arg("use_phase_table") = false,
            py::arg("enable_rfi_s1") = true, py::arg("enable_sk") = true,
-           py::arg("n_slots") = 2, py::arg("fft_backend") = 0,
+           py::arg("n_slots") = 2, py::arg("fft_backend") = 2,
            py::arg("window_kind") = 0, py::arg("use_hip_graph") = false)
       .def("submit", &PyEngine::submit, py::arg("raw"),
            py::arg("dm_override") = std::nan(""))

@@ -102,18 +102,25 @@ PipelineEngine::PipelineEngine(const EngineConfig& cfg, int n_slots)
               "pinned cnt");
     check_hip(hipHostMalloc(&s.h_thresholds, (1 + n_boxcars_) * sizeof(float)),
               "pinned thr");
-    // hand-written FFT when both shapes are pow2 (fwd: packed C2C of n/2;
-    // bwd: batched C2C of l_); hipFFT otherwise
-    native_fft_ = (cfg.fft_backend == 0) && NativeFft::supported(nc_) &&
-                  NativeFft::supported(l_);
+    // backend selection (see EngineConfig::fft_backend): the forward
+    // 2^29-class packed C2C always favors the native plan; the batched
+    // backward favors rocFFT below the measured l_ = 2^17 crossover
+    const int be = cfg.fft_backend;
+    native_fft_ = (be == 0 || be == 2) && NativeFft::supported(nc_);
+    native_bwd_ = native_fft_ && NativeFft::supported(l_) &&
+                  (be == 0 || l_ >= (1ull << 17));
     if (native_fft_) {
       s.nfwd.plan(nc_, 1, -1, s.stream);
-      s.nbwd.plan(l_, s_, +1, s.stream);
-      const int wpr = s.nbwd.dif_sk_wgs_per_row();
-      if (cfg.enable_sk && wpr > 0)
-        check_hip(hipMalloc(&s.sk_dif_partials,
-                            s_ * (size_t)wpr * sizeof(float2)),
-                  "sk dif partials");
+      if (native_bwd_) {
+        s.nbwd.plan(l_, s_, +1, s.stream);
+        const int wpr = s.nbwd.dif_sk_wgs_per_row();
+        if (cfg.enable_sk && wpr > 0)
+          check_hip(hipMalloc(&s.sk_dif_partials,
+                              s_ * (size_t)wpr * sizeof(float2)),
+                    "sk dif partials");
+      } else {
+        s.plans.create_c2c_only(l_, s_, s.stream);
+      }
       check_hip(hipStreamSynchronize(s.stream), "fft table sync");
     } else {
       s.plans.create(n_, l_, s_, s.stream);
@@ -216,7 +223,7 @@ void PipelineEngine::enqueue_chain(Slot& s, const uint8_t* dev_raw,
   // read+write of the 4 GB spectrum); otherwise the standalone fused kernel.
   float2* wf;
   const bool fuse_into_bwd =
-      native_fft_ && !table && s.nbwd.first_pass_fusable();
+      native_bwd_ && !table && s.nbwd.first_pass_fusable();
   if (fuse_into_bwd) {
     FftPreop pre;
     pre.mean_power = cfg_.enable_rfi_s1 ? s.mean_power : nullptr;
@@ -237,7 +244,7 @@ void PipelineEngine::enqueue_chain(Slot& s, const uint8_t* dev_raw,
                   cfg_.rfi_threshold, norm_coeff_, cfg_.zap_ranges,
                   cfg_.n_zap_ranges, f_min_, f_c_, df_, dm, table, st),
               "rfi+dedisp");
-    if (native_fft_) {
+    if (native_bwd_) {
       s.nbwd.exec(s.spec, reinterpret_cast<float2*>(s.samples), st);
       wf = reinterpret_cast<float2*>(s.samples);
     } else {

@@ -50,7 +50,11 @@ struct EngineConfig {
   bool enable_sk = true;
   // 0 = hand-written Stockham FFT when shapes are pow2 (measured 2.2x
   // rocFFT on the J1644 waterfall, 1.1x on the 2^30 forward), 1 = hipFFT
-  int fft_backend = 0;
+  // 0 = native always, 1 = hipFFT always, 2 = auto (native forward; the
+  // batched backward picks rocFFT below the measured 2^17 crossover —
+  // rocFFT's specialized sbcc/sbrc kernels win 1.4-2.9x for lengths
+  // ≤ 2^16, the native column+DIF plan wins 2.2x at 2^18)
+  int fft_backend = 2;
   // FFT window fused into unpack: 0 = rectangle (reference default),
   // 1 = hann, 2 = hamming
   int window_kind = 0;
@@ -137,7 +141,8 @@ class PipelineEngine {
 
   float2* phase_table_ = nullptr;  // shared across slots (read-only)
   float* window_ = nullptr;        // fused FFT window table (null = rect)
-  bool native_fft_ = false;        // hand-written FFT active
+  bool native_fft_ = false;        // hand-written FORWARD FFT active
+  bool native_bwd_ = false;        // hand-written BACKWARD (waterfall) FFT
 
 
   struct Slot {

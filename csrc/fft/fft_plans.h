@@ -66,6 +66,28 @@ class FftPlanSet {
     check_fft(hipfftSetStream(c2c_, stream), "SetStream c2c");
   }
 
+  // C2C plan only (the mixed "auto" backend: native forward FFT + hipFFT
+  // batched backward) — skips the R2C plan, whose work area for a 2^30-point
+  // transform would waste gigabytes on a plan that never runs.
+  void create_c2c_only(size_t c2c_len, size_t c2c_batch, hipStream_t stream) {
+    destroy();
+    size_t ws_c2c = 0;
+    long long n2[1] = {(long long)c2c_len};
+    check_fft(hipfftCreate(&c2c_), "hipfftCreate c2c");
+    check_fft(hipfftSetAutoAllocation(c2c_, 0), "SetAutoAllocation c2c");
+    check_fft(hipfftMakePlanMany64(c2c_, 1, n2, nullptr, 1,
+                                   (long long)c2c_len, nullptr, 1,
+                                   (long long)c2c_len, HIPFFT_C2C,
+                                   (long long)c2c_batch, &ws_c2c),
+              "MakePlanMany64 c2c");
+    work_size_ = ws_c2c;
+    if (work_size_) {
+      check_hip(hipMalloc(&work_area_, work_size_), "hipMalloc fft work");
+      check_fft(hipfftSetWorkArea(c2c_, work_area_), "SetWorkArea c2c");
+    }
+    check_fft(hipfftSetStream(c2c_, stream), "SetStream c2c");
+  }
+
   void exec_r2c(float* in, float2* out) {
     check_fft(hipfftExecR2C(r2c_, in, reinterpret_cast<hipfftComplex*>(out)),
               "ExecR2C");

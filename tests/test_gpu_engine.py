@@ -52,11 +52,14 @@ def make_engine(C, cfg, **kw):
 
 @pytest.mark.parametrize("n,s,fft", [(1 << 18, 1 << 6, 0),
                                       (1 << 20, 1 << 6, 0),
-                                      (1 << 18, 1 << 6, 1)])
+                                      (1 << 18, 1 << 6, 1),
+                                      (1 << 20, 1 << 6, 2)])
 def test_engine_matches_cpu_oracle(C, n, s, fft):
     """Shape 2 has waterfall len 2^13 > 4096 → exercises the multi-pass
     native FFT with the RFI+dedispersion preop fused into its first column
-    pass; variant 3 runs the hipFFT fallback backend."""
+    pass; variant 3 runs the hipFFT fallback backend; variant 4 runs the
+    auto backend, whose small waterfall length (8192) picks the mixed
+    native-forward + rocFFT-backward path."""
     cfg = small_cfg()
     cfg.baseband_input_count = n
     cfg.spectrum_channel_count = s
