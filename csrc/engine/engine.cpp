@@ -105,6 +105,7 @@ PipelineEngine::PipelineEngine(const EngineConfig& cfg, int n_slots)
     // backend selection (see EngineConfig::fft_backend): the forward
     // 2^29-class packed C2C always favors the native plan; the batched
     // backward favors rocFFT below the measured l_ = 2^17 crossover
+    fused_unpack_off_ = std::getenv("SRTB_NO_FUSED_UNPACK") != nullptr;
     const int be = cfg.fft_backend;
     native_fft_ = (be == 0 || be == 2) && NativeFft::supported(nc_);
     native_bwd_ = native_fft_ && NativeFft::supported(l_) &&
@@ -186,24 +187,31 @@ void PipelineEngine::enqueue_chain(Slot& s, const uint8_t* dev_raw,
 
   RoctxRange r_chain("srtb_block_chain");
   const float* fft_in = s.samples;
-  if (dev_raw) {
+  // fused unpack: for the 2-bit rectangular-window native-forward case the
+  // FFT's first column pass decodes the raw bytes itself (0.25 GB of byte
+  // reads replace the unpack kernel's 4 GB write + the pass's 4 GB read)
+  const bool fuse_unpack = dev_raw && native_fft_ && !fused_unpack_off_ &&
+                           cfg_.baseband_input_bits == 2 && !window_ &&
+                           slots_[0]->nfwd.first_pass_fusable();
+  if (dev_raw && !fuse_unpack) {
     // 1. unpack (+ window fused; default rectangle → none)
     check_hip(unpack(dev_raw, s.samples, n_, cfg_.baseband_input_bits,
                      window_, st),
               "unpack");
-  } else if (native_fft_) {
+  } else if (!dev_raw && native_fft_) {
     // native fwd runs column passes in place on its input: copy the caller's
     // samples into the slot buffer first (D2D, overlapped on stream)
     check_hip(hipMemcpyAsync(s.samples, dev_samples, n_ * sizeof(float),
                              hipMemcpyDeviceToDevice, st),
               "samples d2d");
-  } else {
+  } else if (!dev_raw) {
     fft_in = dev_samples;
   }
   if (native_fft_) {
     // 2. forward C2C of the packed-real view + r2c post-process with FUSED
     //    mean-|X|^2 (saves the separate 4 GB mean_power pass)
-    s.nfwd.exec(reinterpret_cast<float2*>(s.samples), s.spec, st);
+    s.nfwd.exec(reinterpret_cast<float2*>(s.samples), s.spec, st,
+                nullptr, nullptr, fuse_unpack ? dev_raw : nullptr);
     check_hip(r2c_post_process(
                   s.spec, s.spec, nc_,
                   cfg_.enable_rfi_s1 ? s.partials : nullptr,

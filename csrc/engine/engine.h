@@ -48,8 +48,6 @@ struct EngineConfig {
   bool use_phase_table = false;  // cache dedispersion factors for fixed DM
   bool enable_rfi_s1 = true;
   bool enable_sk = true;
-  // 0 = hand-written Stockham FFT when shapes are pow2 (measured 2.2x
-  // rocFFT on the J1644 waterfall, 1.1x on the 2^30 forward), 1 = hipFFT
   // 0 = native always, 1 = hipFFT always, 2 = auto (native forward; the
   // batched backward picks rocFFT below the measured 2^17 crossover —
   // rocFFT's specialized sbcc/sbrc kernels win 1.4-2.9x for lengths
@@ -142,6 +140,7 @@ class PipelineEngine {
   float2* phase_table_ = nullptr;  // shared across slots (read-only)
   float* window_ = nullptr;        // fused FFT window table (null = rect)
   bool native_fft_ = false;        // hand-written FORWARD FFT active
+  bool fused_unpack_off_ = false;  // SRTB_NO_FUSED_UNPACK kill switch
   bool native_bwd_ = false;        // hand-written BACKWARD (waterfall) FFT
 
 

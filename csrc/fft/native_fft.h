@@ -185,12 +185,15 @@ class NativeFft {
   // first_pass_fusable()).
   void exec(const float2* in, float2* out, hipStream_t stream,
             const FftPreop* preop = nullptr,
-            float2* dif_sk_partials = nullptr) {
+            float2* dif_sk_partials = nullptr,
+            const uint8_t* decode2_raw = nullptr) {
     if (passes_.empty()) throw std::runtime_error("NativeFft: not planned");
     if (passes_.size() > 1 && in == out)
       throw std::runtime_error("NativeFft: multi-pass needs out != in");
     if (preop && !first_pass_fusable())
       throw std::runtime_error("NativeFft: preop needs a column first pass");
+    if (decode2_raw && (preop || !first_pass_fusable()))
+      throw std::runtime_error("NativeFft: decode2 needs a column first pass");
     float2* cur = const_cast<float2*>(in);
     for (size_t i = 0; i < passes_.size(); ++i) {
       Pass& p = passes_[i];
@@ -208,7 +211,8 @@ class NativeFft {
         }
         case PassKind::kCol:
           check_hip(fft_col_pass(cur, cur, p.d, p.n_ffts, sign_, p.tw_n,
-                                 p.tw_hi, p.tw_lo, stream, pre),
+                                 p.tw_hi, p.tw_lo, stream, pre,
+                                 i == 0 ? decode2_raw : nullptr),
                     "fft_col_pass");
           dst = cur;  // in place
           break;

@@ -76,6 +76,18 @@ __device__ inline void digits(unsigned long long id, const FftPassDescDev& d,
   q2 = r >> d.d1_log2;
 }
 
+// fused 2-bit unpack for the FORWARD first pass: complex element `flat` of
+// the packed-real view is real samples (2*flat, 2*flat+1), both inside raw
+// byte flat>>1 (MSB-first fields).  Reading 0.25 GB of bytes replaces a
+// 4 GB float re-read AND the standalone unpack kernel's 4 GB write.
+__device__ inline float2 dec2_load(const uint8_t* __restrict__ raw,
+                                   unsigned long long flat) {
+  const uint32_t bv = raw[flat >> 1];
+  const int j0 = (int)(flat & 1) * 2;
+  return make_float2((float)((bv >> ((3 - j0) * 2)) & 3u),
+                     (float)((bv >> ((2 - j0) * 2)) & 3u));
+}
+
 // LDS layout: [tw: n][X: F*(n+2)][Y: F*(n+2)] float2s.
 template <bool LOAD_FFAST, bool STORE_FFAST, bool TWIDDLE, int SIGN>
 __global__ void __launch_bounds__(256)
@@ -328,13 +340,14 @@ struct FftPreopDev {
   double f_min, f_c, df, dm;
 };
 
-template <int N, bool TWIDDLE, int SIGN, bool PREOP>
+template <int N, bool TWIDDLE, int SIGN, bool PREOP, bool DEC2 = false>
 __global__ void __launch_bounds__(256)
     k_fft_col(const float2* __restrict__ in, float2* __restrict__ out,
               FftPassDescDev d, unsigned long long n_ffts,
               const float2* __restrict__ tw_n,
               const float2* __restrict__ tw_hi,
-              const float2* __restrict__ tw_lo, FftPreopDev pre) {
+              const float2* __restrict__ tw_lo, FftPreopDev pre,
+              const uint8_t* __restrict__ raw2) {
   const unsigned long long id =
       (unsigned long long)blockIdx.x * blockDim.x + threadIdx.x;
   if (id >= n_ffts) return;
@@ -353,8 +366,10 @@ __global__ void __launch_bounds__(256)
 #pragma unroll
   for (int i = 0; i < N; ++i) {
     const uint32_t off = (uint32_t)i * stride32;
-    float2 x = colbase[off];
     const unsigned long long flat = base + off;
+    float2 x;
+    if constexpr (DEC2) x = dec2_load(raw2, flat);
+    else x = colbase[off];
     if constexpr (PREOP) {
       // fused rfi_dedisperse (spectrum.hip k_rfi_dedisp_fused semantics);
       // `flat` IS the spectrum bin index for in-place column passes
@@ -411,13 +426,14 @@ __device__ inline float2 shfl_xor1(float2 x) {
   return make_float2(__shfl_xor(x.x, 1, 64), __shfl_xor(x.y, 1, 64));
 }
 
-template <bool TWIDDLE, int SIGN, bool PREOP>
+template <bool TWIDDLE, int SIGN, bool PREOP, bool DEC2 = false>
 __global__ void __launch_bounds__(256)
     k_fft_col_pair64(const float2* __restrict__ in, float2* __restrict__ out,
                      FftPassDescDev d, unsigned long long n_ffts,
                      const float2* __restrict__ tw_n,
                      const float2* __restrict__ tw_hi,
-                     const float2* __restrict__ tw_lo, FftPreopDev pre) {
+                     const float2* __restrict__ tw_lo, FftPreopDev pre,
+                     const uint8_t* __restrict__ raw2) {
   const unsigned long long tid =
       (unsigned long long)blockIdx.x * blockDim.x + threadIdx.x;
   const unsigned long long id = tid >> 1;  // column index
@@ -439,7 +455,9 @@ __global__ void __launch_bounds__(256)
     // compile-time constants, selected per-lane
     const int i = p ? col_sigma_inv64(32 + r) : col_sigma_inv64(r);
     const uint32_t off = (uint32_t)i * stride32;
-    float2 x = colbase[off];
+    float2 x;
+    if constexpr (DEC2) x = dec2_load(raw2, base + off);
+    else x = colbase[off];
     if constexpr (PREOP) {
       const unsigned long long flat = base + off;
       bool zap = pre.mean_power && (norm2(x) > thr_mean);
@@ -823,7 +841,9 @@ hipError_t fft_stockham_pass(const float2* in, float2* out,
 hipError_t fft_col_pass(const float2* in, float2* out, const FftPassDesc& hd,
                         size_t n_ffts, int sign, const float2* tw_n,
                         const float2* tw_hi, const float2* tw_lo,
-                        hipStream_t stream, const FftPreop* preop) {
+                        hipStream_t stream, const FftPreop* preop,
+                        const uint8_t* raw2) {
+  if (raw2 && (preop || hd.tw_mod == 0)) return hipErrorInvalidValue;
   FftPreopDev pre{};
   if (preop) {
     pre.mean_power = preop->mean_power;
@@ -855,10 +875,18 @@ hipError_t fft_col_pass(const float2* in, float2* out, const FftPassDesc& hd,
 
 #define COL_LAUNCH(N, TW, SG, PR)                                           \
   hipLaunchKernelGGL((k_fft_col<N, TW, SG, PR>), dim3(grid), dim3(256), 0,   \
-                     stream, in, out, d, n_ffts, tw_n, tw_hi, tw_lo, pre)
+                     stream, in, out, d, n_ffts, tw_n, tw_hi, tw_lo, pre,    \
+                     raw2)
+#define COL_LAUNCH_D(N, SG)                                                  \
+  hipLaunchKernelGGL((k_fft_col<N, true, SG, false, true>), dim3(grid),      \
+                     dim3(256), 0, stream, in, out, d, n_ffts, tw_n, tw_hi,  \
+                     tw_lo, pre, raw2)
 #define COL_DISPATCH(N)                                                      \
   case N:                                                                    \
-    if (twiddle) {                                                           \
+    if (raw2) {                                                              \
+      if (sign < 0) COL_LAUNCH_D(N, -1);                                     \
+      else COL_LAUNCH_D(N, 1);                                               \
+    } else if (twiddle) {                                                           \
       if (sign < 0) { if (preop) COL_LAUNCH(N, true, -1, true);              \
                       else COL_LAUNCH(N, true, -1, false); }                 \
       else          { if (preop) COL_LAUNCH(N, true, 1, true);               \
@@ -874,7 +902,12 @@ hipError_t fft_col_pass(const float2* in, float2* out, const FftPassDesc& hd,
   const uint32_t grid2 = (uint32_t)((2 * n_ffts + 255) / 256);
 #define COL_LAUNCH_P(TW, SG, PR)                                             \
   hipLaunchKernelGGL((k_fft_col_pair64<TW, SG, PR>), dim3(grid2), dim3(256), \
-                     0, stream, in, out, d, n_ffts, tw_n, tw_hi, tw_lo, pre)
+                     0, stream, in, out, d, n_ffts, tw_n, tw_hi, tw_lo, pre, \
+                     raw2)
+#define COL_LAUNCH_PD(SG)                                                    \
+  hipLaunchKernelGGL((k_fft_col_pair64<true, SG, false, true>), dim3(grid2), \
+                     dim3(256), 0, stream, in, out, d, n_ffts, tw_n, tw_hi,  \
+                     tw_lo, pre, raw2)
   switch (hd.n) {
     COL_DISPATCH(2)
     COL_DISPATCH(4)
@@ -882,6 +915,11 @@ hipError_t fft_col_pass(const float2* in, float2* out, const FftPassDesc& hd,
     COL_DISPATCH(16)
     COL_DISPATCH(32)
     case 64:
+      if (raw2) {
+        if (sign < 0) COL_LAUNCH_PD(-1);
+        else COL_LAUNCH_PD(1);
+        break;
+      }
       if (twiddle) {
         if (sign < 0) { if (preop) COL_LAUNCH_P(true, -1, true);
                         else COL_LAUNCH_P(true, -1, false); }
@@ -899,7 +937,9 @@ hipError_t fft_col_pass(const float2* in, float2* out, const FftPassDesc& hd,
   }
 #undef COL_DISPATCH
 #undef COL_LAUNCH
+#undef COL_LAUNCH_D
 #undef COL_LAUNCH_P
+#undef COL_LAUNCH_PD
   SRTB_CHECK_LAUNCH();
   return hipSuccess;
 }
