@@ -190,9 +190,10 @@ void PipelineEngine::enqueue_chain(Slot& s, const uint8_t* dev_raw,
   // fused unpack: for the 2-bit rectangular-window native-forward case the
   // FFT's first column pass decodes the raw bytes itself (0.25 GB of byte
   // reads replace the unpack kernel's 4 GB write + the pass's 4 GB read)
+  const int in_bits = cfg_.baseband_input_bits;
   const bool fuse_unpack = dev_raw && native_fft_ && !fused_unpack_off_ &&
-                           cfg_.baseband_input_bits == 2 && !window_ &&
-                           slots_[0]->nfwd.first_pass_fusable();
+                           (in_bits == 1 || in_bits == 2 || in_bits == 4) &&
+                           !window_ && slots_[0]->nfwd.first_pass_fusable();
   if (dev_raw && !fuse_unpack) {
     // 1. unpack (+ window fused; default rectangle → none)
     check_hip(unpack(dev_raw, s.samples, n_, cfg_.baseband_input_bits,
@@ -211,7 +212,7 @@ void PipelineEngine::enqueue_chain(Slot& s, const uint8_t* dev_raw,
     // 2. forward C2C of the packed-real view + r2c post-process with FUSED
     //    mean-|X|^2 (saves the separate 4 GB mean_power pass)
     s.nfwd.exec(reinterpret_cast<float2*>(s.samples), s.spec, st,
-                nullptr, nullptr, fuse_unpack ? dev_raw : nullptr);
+                nullptr, nullptr, fuse_unpack ? dev_raw : nullptr, in_bits);
     check_hip(r2c_post_process(
                   s.spec, s.spec, nc_,
                   cfg_.enable_rfi_s1 ? s.partials : nullptr,

@@ -186,7 +186,7 @@ class NativeFft {
   void exec(const float2* in, float2* out, hipStream_t stream,
             const FftPreop* preop = nullptr,
             float2* dif_sk_partials = nullptr,
-            const uint8_t* decode2_raw = nullptr) {
+            const uint8_t* decode2_raw = nullptr, int decode_bits = 2) {
     if (passes_.empty()) throw std::runtime_error("NativeFft: not planned");
     if (passes_.size() > 1 && in == out)
       throw std::runtime_error("NativeFft: multi-pass needs out != in");
@@ -212,7 +212,7 @@ class NativeFft {
         case PassKind::kCol:
           check_hip(fft_col_pass(cur, cur, p.d, p.n_ffts, sign_, p.tw_n,
                                  p.tw_hi, p.tw_lo, stream, pre,
-                                 i == 0 ? decode2_raw : nullptr),
+                                 i == 0 ? decode2_raw : nullptr, decode_bits),
                     "fft_col_pass");
           dst = cur;  // in place
           break;

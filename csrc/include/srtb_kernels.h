@@ -203,14 +203,14 @@ struct FftPreop {
 // Register-resident column FFT pass (N in {2,4,8,16,32,64}; one FFT per
 // thread fully in VGPRs; in-place: out must alias layout of in addressing;
 // uses the same FftPassDesc fields; out_* ignored, stores to input layout).
-// raw2 (optional): fused 2-bit unpack — the pass loads and decodes raw
-// bytes instead of reading `in` (forward first pass only; requires an
-// inter-pass twiddle and no preop).
+// raw2 (optional): fused sub-byte unpack (raw_bits in {1,2,4}) — the pass
+// loads and decodes raw bytes instead of reading `in` (forward first pass
+// only; requires an inter-pass twiddle and no preop).
 hipError_t fft_col_pass(const float2* in, float2* out, const FftPassDesc& d,
                         size_t n_ffts, int sign, const float2* tw_n,
                         const float2* tw_hi, const float2* tw_lo,
                         hipStream_t stream, const FftPreop* preop = nullptr,
-                        const uint8_t* raw2 = nullptr);
+                        const uint8_t* raw2 = nullptr, int raw_bits = 2);
 
 // Final composite pass: in-place radix-4 DIF in LDS with base-4
 // digit-reversal folded into the store, plus multi-digit output scatter.

@@ -76,16 +76,28 @@ __device__ inline void digits(unsigned long long id, const FftPassDescDev& d,
   q2 = r >> d.d1_log2;
 }
 
-// fused 2-bit unpack for the FORWARD first pass: complex element `flat` of
-// the packed-real view is real samples (2*flat, 2*flat+1), both inside raw
-// byte flat>>1 (MSB-first fields).  Reading 0.25 GB of bytes replaces a
-// 4 GB float re-read AND the standalone unpack kernel's 4 GB write.
-__device__ inline float2 dec2_load(const uint8_t* __restrict__ raw,
+// fused sub-byte unpack for the FORWARD first pass: complex element `flat`
+// of the packed-real view is real samples (2*flat, 2*flat+1), which live in
+// the same raw byte (MSB-first fields).  Reading 0.12-0.5 GB of bytes
+// replaces a 4 GB float re-read AND the standalone unpack kernel's 4 GB
+// write (measured +7.6%% end-to-end on the 2-bit J1644 config).
+template <int NBITS>
+__device__ inline float2 decN_load(const uint8_t* __restrict__ raw,
                                    unsigned long long flat) {
-  const uint32_t bv = raw[flat >> 1];
-  const int j0 = (int)(flat & 1) * 2;
-  return make_float2((float)((bv >> ((3 - j0) * 2)) & 3u),
-                     (float)((bv >> ((2 - j0) * 2)) & 3u));
+  if constexpr (NBITS == 4) {
+    const uint32_t bv = raw[flat];
+    return make_float2((float)((bv >> 4) & 15u), (float)(bv & 15u));
+  } else if constexpr (NBITS == 2) {
+    const uint32_t bv = raw[flat >> 1];
+    const int j0 = (int)(flat & 1) * 2;
+    return make_float2((float)((bv >> ((3 - j0) * 2)) & 3u),
+                       (float)((bv >> ((2 - j0) * 2)) & 3u));
+  } else {  // NBITS == 1
+    const uint32_t bv = raw[flat >> 2];
+    const int j0 = (int)(flat & 3) * 2;
+    return make_float2((float)((bv >> (7 - j0)) & 1u),
+                       (float)((bv >> (6 - j0)) & 1u));
+  }
 }
 
 // LDS layout: [tw: n][X: F*(n+2)][Y: F*(n+2)] float2s.
@@ -340,7 +352,7 @@ struct FftPreopDev {
   double f_min, f_c, df, dm;
 };
 
-template <int N, bool TWIDDLE, int SIGN, bool PREOP, bool DEC2 = false>
+template <int N, bool TWIDDLE, int SIGN, bool PREOP, int DEC = 0>
 __global__ void __launch_bounds__(256)
     k_fft_col(const float2* __restrict__ in, float2* __restrict__ out,
               FftPassDescDev d, unsigned long long n_ffts,
@@ -368,7 +380,7 @@ __global__ void __launch_bounds__(256)
     const uint32_t off = (uint32_t)i * stride32;
     const unsigned long long flat = base + off;
     float2 x;
-    if constexpr (DEC2) x = dec2_load(raw2, flat);
+    if constexpr (DEC != 0) x = decN_load<DEC>(raw2, flat);
     else x = colbase[off];
     if constexpr (PREOP) {
       // fused rfi_dedisperse (spectrum.hip k_rfi_dedisp_fused semantics);
@@ -426,7 +438,7 @@ __device__ inline float2 shfl_xor1(float2 x) {
   return make_float2(__shfl_xor(x.x, 1, 64), __shfl_xor(x.y, 1, 64));
 }
 
-template <bool TWIDDLE, int SIGN, bool PREOP, bool DEC2 = false>
+template <bool TWIDDLE, int SIGN, bool PREOP, int DEC = 0>
 __global__ void __launch_bounds__(256)
     k_fft_col_pair64(const float2* __restrict__ in, float2* __restrict__ out,
                      FftPassDescDev d, unsigned long long n_ffts,
@@ -456,7 +468,7 @@ __global__ void __launch_bounds__(256)
     const int i = p ? col_sigma_inv64(32 + r) : col_sigma_inv64(r);
     const uint32_t off = (uint32_t)i * stride32;
     float2 x;
-    if constexpr (DEC2) x = dec2_load(raw2, base + off);
+    if constexpr (DEC != 0) x = decN_load<DEC>(raw2, base + off);
     else x = colbase[off];
     if constexpr (PREOP) {
       const unsigned long long flat = base + off;
@@ -842,8 +854,10 @@ hipError_t fft_col_pass(const float2* in, float2* out, const FftPassDesc& hd,
                         size_t n_ffts, int sign, const float2* tw_n,
                         const float2* tw_hi, const float2* tw_lo,
                         hipStream_t stream, const FftPreop* preop,
-                        const uint8_t* raw2) {
+                        const uint8_t* raw2, int raw_bits) {
   if (raw2 && (preop || hd.tw_mod == 0)) return hipErrorInvalidValue;
+  if (raw2 && raw_bits != 1 && raw_bits != 2 && raw_bits != 4)
+    return hipErrorInvalidValue;
   FftPreopDev pre{};
   if (preop) {
     pre.mean_power = preop->mean_power;
@@ -877,15 +891,21 @@ hipError_t fft_col_pass(const float2* in, float2* out, const FftPassDesc& hd,
   hipLaunchKernelGGL((k_fft_col<N, TW, SG, PR>), dim3(grid), dim3(256), 0,   \
                      stream, in, out, d, n_ffts, tw_n, tw_hi, tw_lo, pre,    \
                      raw2)
-#define COL_LAUNCH_D(N, SG)                                                  \
-  hipLaunchKernelGGL((k_fft_col<N, true, SG, false, true>), dim3(grid),      \
+#define COL_LAUNCH_D(N, SG, B)                                               \
+  hipLaunchKernelGGL((k_fft_col<N, true, SG, false, B>), dim3(grid),         \
                      dim3(256), 0, stream, in, out, d, n_ffts, tw_n, tw_hi,  \
                      tw_lo, pre, raw2)
+#define COL_DISPATCH_D(N)                                                    \
+  if (raw_bits == 1) { if (sign < 0) COL_LAUNCH_D(N, -1, 1);                 \
+                       else COL_LAUNCH_D(N, 1, 1); }                         \
+  else if (raw_bits == 2) { if (sign < 0) COL_LAUNCH_D(N, -1, 2);            \
+                            else COL_LAUNCH_D(N, 1, 2); }                    \
+  else { if (sign < 0) COL_LAUNCH_D(N, -1, 4);                               \
+         else COL_LAUNCH_D(N, 1, 4); }
 #define COL_DISPATCH(N)                                                      \
   case N:                                                                    \
     if (raw2) {                                                              \
-      if (sign < 0) COL_LAUNCH_D(N, -1);                                     \
-      else COL_LAUNCH_D(N, 1);                                               \
+      COL_DISPATCH_D(N)                                                      \
     } else if (twiddle) {                                                           \
       if (sign < 0) { if (preop) COL_LAUNCH(N, true, -1, true);              \
                       else COL_LAUNCH(N, true, -1, false); }                 \
@@ -904,8 +924,8 @@ hipError_t fft_col_pass(const float2* in, float2* out, const FftPassDesc& hd,
   hipLaunchKernelGGL((k_fft_col_pair64<TW, SG, PR>), dim3(grid2), dim3(256), \
                      0, stream, in, out, d, n_ffts, tw_n, tw_hi, tw_lo, pre, \
                      raw2)
-#define COL_LAUNCH_PD(SG)                                                    \
-  hipLaunchKernelGGL((k_fft_col_pair64<true, SG, false, true>), dim3(grid2), \
+#define COL_LAUNCH_PD(SG, B)                                                 \
+  hipLaunchKernelGGL((k_fft_col_pair64<true, SG, false, B>), dim3(grid2),    \
                      dim3(256), 0, stream, in, out, d, n_ffts, tw_n, tw_hi,  \
                      tw_lo, pre, raw2)
   switch (hd.n) {
@@ -916,8 +936,12 @@ hipError_t fft_col_pass(const float2* in, float2* out, const FftPassDesc& hd,
     COL_DISPATCH(32)
     case 64:
       if (raw2) {
-        if (sign < 0) COL_LAUNCH_PD(-1);
-        else COL_LAUNCH_PD(1);
+        if (raw_bits == 1) { if (sign < 0) COL_LAUNCH_PD(-1, 1);
+                             else COL_LAUNCH_PD(1, 1); }
+        else if (raw_bits == 2) { if (sign < 0) COL_LAUNCH_PD(-1, 2);
+                                  else COL_LAUNCH_PD(1, 2); }
+        else { if (sign < 0) COL_LAUNCH_PD(-1, 4);
+               else COL_LAUNCH_PD(1, 4); }
         break;
       }
       if (twiddle) {
@@ -936,6 +960,7 @@ hipError_t fft_col_pass(const float2* in, float2* out, const FftPassDesc& hd,
       return hipErrorInvalidValue;
   }
 #undef COL_DISPATCH
+#undef COL_DISPATCH_D
 #undef COL_LAUNCH
 #undef COL_LAUNCH_D
 #undef COL_LAUNCH_P

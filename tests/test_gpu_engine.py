@@ -121,8 +121,12 @@ def test_engine_quiet_on_noise(C):
     assert all(c == 0 for _, c in res["counts"]), res
 
 
-def test_engine_2bit_path(C):
-    cfg = small_cfg(bits=2)
+@pytest.mark.parametrize("bits", [1, 2, 4])
+def test_engine_subbyte_paths(C, bits):
+    """Sub-byte formats run with the unpack fused into the forward FFT's
+    first column pass (decode-at-load); the dispersed pulse must still land
+    in the right time-series bin."""
+    cfg = small_cfg(bits=bits)
     t_pulse = 0.5 * cfg.baseband_input_count / cfg.baseband_sample_rate
     raw = synthesize_dispersed_pulse(cfg, t_pulse, pulse_amp=12.0,
                                      noise_sigma=2.0)
@@ -133,6 +137,25 @@ def test_engine_2bit_path(C):
     expect_bin = int(t_pulse * cfg.baseband_sample_rate) // (
         2 * cfg.spectrum_channel_count)
     assert abs(int(np.argmax(ts)) - expect_bin) <= 2
+
+
+def test_engine_subbyte_fused_matches_unfused(C, monkeypatch):
+    """The fused decode-at-load path must produce the same time series as
+    the standalone unpack kernel path (SRTB_NO_FUSED_UNPACK)."""
+    cfg = small_cfg(bits=2)
+    rng = np.random.default_rng(7)
+    raw = rng.integers(0, 256, cfg.baseband_input_count // 4,
+                       dtype=np.uint8)
+    eng = make_engine(C, cfg)
+    slot = eng.submit(torch.from_numpy(raw.copy()))
+    eng.wait(slot)
+    ts_fused = eng.time_series(slot).cpu().numpy().copy()
+    monkeypatch.setenv("SRTB_NO_FUSED_UNPACK", "1")
+    eng2 = make_engine(C, cfg)
+    slot2 = eng2.submit(torch.from_numpy(raw.copy()))
+    eng2.wait(slot2)
+    ts_plain = eng2.time_series(slot2).cpu().numpy()
+    np.testing.assert_allclose(ts_fused, ts_plain, rtol=1e-5, atol=1e-3)
 
 
 def test_engine_double_buffering_order(C):
