@@ -128,4 +128,17 @@ def synthesize_dispersed_pulse(cfg: Config, pulse_t: float, pulse_amp: float,
         lv = lv.reshape(-1, 4)
         packed = (lv[:, 0] << 6) | (lv[:, 1] << 4) | (lv[:, 2] << 2) | lv[:, 3]
         return packed.astype(np.uint8)
+    if bits == 4:
+        # 4-bit levels 0..15 centered on 8, step = noise_sigma/2
+        lv = np.clip(np.round(sig / (noise_sigma / 2) + 8), 0, 15
+                     ).astype(np.uint8)
+        lv = lv.reshape(-1, 2)
+        return ((lv[:, 0] << 4) | lv[:, 1]).astype(np.uint8)
+    if bits == 1:
+        # 1-bit sign quantization, MSB-first packing
+        lv = (sig > 0).astype(np.uint8).reshape(-1, 8)
+        packed = np.zeros(lv.shape[0], dtype=np.uint8)
+        for j in range(8):
+            packed |= lv[:, j] << (7 - j)
+        return packed
     raise ValueError(f"unsupported bits {bits} for synthesis")
