@@ -92,6 +92,11 @@ __device__ inline float2 decN_load(const uint8_t* __restrict__ raw,
     const int j0 = (int)(flat & 1) * 2;
     return make_float2((float)((bv >> ((3 - j0) * 2)) & 3u),
                        (float)((bv >> ((2 - j0) * 2)) & 3u));
+  } else if constexpr (NBITS == 8) {
+    return make_float2((float)raw[2 * flat], (float)raw[2 * flat + 1]);
+  } else if constexpr (NBITS == -8) {
+    const int8_t* r8 = reinterpret_cast<const int8_t*>(raw);
+    return make_float2((float)r8[2 * flat], (float)r8[2 * flat + 1]);
   } else {  // NBITS == 1
     const uint32_t bv = raw[flat >> 2];
     const int j0 = (int)(flat & 3) * 2;
@@ -856,7 +861,8 @@ hipError_t fft_col_pass(const float2* in, float2* out, const FftPassDesc& hd,
                         hipStream_t stream, const FftPreop* preop,
                         const uint8_t* raw2, int raw_bits) {
   if (raw2 && (preop || hd.tw_mod == 0)) return hipErrorInvalidValue;
-  if (raw2 && raw_bits != 1 && raw_bits != 2 && raw_bits != 4)
+  if (raw2 && raw_bits != 1 && raw_bits != 2 && raw_bits != 4 &&
+      raw_bits != 8 && raw_bits != -8)
     return hipErrorInvalidValue;
   FftPreopDev pre{};
   if (preop) {
@@ -900,8 +906,12 @@ hipError_t fft_col_pass(const float2* in, float2* out, const FftPassDesc& hd,
                        else COL_LAUNCH_D(N, 1, 1); }                         \
   else if (raw_bits == 2) { if (sign < 0) COL_LAUNCH_D(N, -1, 2);            \
                             else COL_LAUNCH_D(N, 1, 2); }                    \
-  else { if (sign < 0) COL_LAUNCH_D(N, -1, 4);                               \
-         else COL_LAUNCH_D(N, 1, 4); }
+  else if (raw_bits == 4) { if (sign < 0) COL_LAUNCH_D(N, -1, 4);            \
+                            else COL_LAUNCH_D(N, 1, 4); }                    \
+  else if (raw_bits == 8) { if (sign < 0) COL_LAUNCH_D(N, -1, 8);            \
+                            else COL_LAUNCH_D(N, 1, 8); }                    \
+  else { if (sign < 0) COL_LAUNCH_D(N, -1, -8);                              \
+         else COL_LAUNCH_D(N, 1, -8); }
 #define COL_DISPATCH(N)                                                      \
   case N:                                                                    \
     if (raw2) {                                                              \
@@ -940,8 +950,12 @@ hipError_t fft_col_pass(const float2* in, float2* out, const FftPassDesc& hd,
                              else COL_LAUNCH_PD(1, 1); }
         else if (raw_bits == 2) { if (sign < 0) COL_LAUNCH_PD(-1, 2);
                                   else COL_LAUNCH_PD(1, 2); }
-        else { if (sign < 0) COL_LAUNCH_PD(-1, 4);
-               else COL_LAUNCH_PD(1, 4); }
+        else if (raw_bits == 4) { if (sign < 0) COL_LAUNCH_PD(-1, 4);
+                                  else COL_LAUNCH_PD(1, 4); }
+        else if (raw_bits == 8) { if (sign < 0) COL_LAUNCH_PD(-1, 8);
+                                  else COL_LAUNCH_PD(1, 8); }
+        else { if (sign < 0) COL_LAUNCH_PD(-1, -8);
+               else COL_LAUNCH_PD(1, -8); }
         break;
       }
       if (twiddle) {
