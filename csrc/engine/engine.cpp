@@ -232,8 +232,7 @@ void PipelineEngine::enqueue_chain(Slot& s, const uint8_t* dev_raw,
   // first column pass when the native planner allows it (saves a full
   // read+write of the 4 GB spectrum); otherwise the standalone fused kernel.
   float2* wf;
-  const bool fuse_into_bwd =
-      native_bwd_ && !table && s.nbwd.first_pass_fusable();
+  const bool fuse_into_bwd = native_bwd_ && s.nbwd.first_pass_fusable();
   if (fuse_into_bwd) {
     FftPreop pre;
     pre.mean_power = cfg_.enable_rfi_s1 ? s.mean_power : nullptr;
@@ -245,6 +244,7 @@ void PipelineEngine::enqueue_chain(Slot& s, const uint8_t* dev_raw,
     pre.f_c = f_c_;
     pre.df = df_;
     pre.dm = dm;
+    pre.table = table;
     s.nbwd.exec(s.spec, reinterpret_cast<float2*>(s.samples), st, &pre,
                 s.sk_dif_partials);
     wf = reinterpret_cast<float2*>(s.samples);

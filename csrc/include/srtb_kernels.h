@@ -198,6 +198,10 @@ struct FftPreop {
   int n_zap = 0;
   ZapRange zap[16] = {};
   double f_min = 0, f_c = 0, df = 0, dm = 0;
+  // optional cached dedispersion factors [nc]; when set the pass reads the
+  // table (one extra coalesced float2 load) instead of the fp64 phase
+  // computation, which would otherwise double the pass's VALU cost
+  const float2* table = nullptr;
 };
 
 // Register-resident column FFT pass (N in {2,4,8,16,32,64}; one FFT per

@@ -355,6 +355,7 @@ struct FftPreopDev {
   int n_zap;
   ZapRange zap[16];
   double f_min, f_c, df, dm;
+  const float2* table;  // cached dedispersion factors (null = compute fp64)
 };
 
 template <int N, bool TWIDDLE, int SIGN, bool PREOP, int DEC = 0>
@@ -400,8 +401,9 @@ __global__ void __launch_bounds__(256)
           x.x *= pre.norm_coeff;
           x.y *= pre.norm_coeff;
         }
-        x = cmulf(x, srtb_dedisp_factor(flat, pre.f_min, pre.f_c, pre.df,
-                                        pre.dm));
+        x = cmulf(x, pre.table ? pre.table[flat]
+                               : srtb_dedisp_factor(flat, pre.f_min, pre.f_c,
+                                                    pre.df, pre.dm));
       }
     }
     v[col_sigma<N>(i)] = x;
@@ -875,6 +877,7 @@ hipError_t fft_col_pass(const float2* in, float2* out, const FftPassDesc& hd,
     pre.f_c = preop->f_c;
     pre.df = preop->df;
     pre.dm = preop->dm;
+    pre.table = preop->table;
   }
   FftPassDescDev d;
   d.n = hd.n;
