@@ -18,6 +18,7 @@
 #include <cstdlib>
 #include <cmath>
 #include <stdexcept>
+#include <string>
 #include <vector>
 
 #include "../include/srtb_kernels.h"
@@ -83,7 +84,29 @@ class NativeFft {
       return;
     }
     std::vector<uint32_t> f;
-    {
+    // SRTB_FFT_FACTORS="64,64,64,8,256": explicit factorization override
+    // (all factors incl. the final DIF length; product must equal len) —
+    // for plan-shape experiments without planner changes
+    if (const char* fe = std::getenv("SRTB_FFT_FACTORS")) {
+      size_t prod = 1;
+      std::string str(fe);
+      for (size_t pos = 0; pos < str.size();) {
+        size_t comma = str.find(',', pos);
+        if (comma == std::string::npos) comma = str.size();
+        const int v = std::atoi(str.substr(pos, comma - pos).c_str());
+        if (v > 1) { f.push_back((uint32_t)v); prod *= (size_t)v; }
+        pos = comma + 1;
+      }
+      if (prod != len || f.size() < 2 || f.size() > 5)
+        f.clear();  // invalid override for this length: fall through
+      else {
+        for (size_t i = 0; i + 1 < f.size(); ++i)
+          if (f[i] > 64) f.clear();
+      }
+    }
+    if (!f.empty()) {
+      // accepted override
+    } else {
       const char* mc = std::getenv("SRTB_FFT_MAXCOL");
       const int maxcol_log2 = mc ? ilog2z(std::atoi(mc)) : 5;  // default 32
       const char* fe = std::getenv("SRTB_FFT_FINAL");
