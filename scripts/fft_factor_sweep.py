@@ -27,8 +27,9 @@ print("err=%%.2e fwd2^29=%%.3f ms" %% (err, t))
 
 
 def main():
-    variants = ["", "64,64,64,8,256", "64,64,32,16,256", "64,64,64,2048",
-                "64,32,32,32,256", "32,64,64,8,256", "8,64,64,64,256"]
+    variants = sys.argv[1:] or [
+        "", "64,64,64,8,256", "64,64,32,16,256",
+        "64,32,32,32,256", "32,64,64,8,256", "8,64,64,64,256"]
     child = CHILD.replace("ROOT_PLACEHOLDER", ROOT)
     for fac in variants:
         env = dict(os.environ)
