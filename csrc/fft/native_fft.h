@@ -116,16 +116,29 @@ class NativeFft {
       if (final_log2 > 12) final_log2 = 12;
       if (final_log2 >= t) final_log2 = (t % 2) ? t - 1 : t - 2;
       const int rest = t - final_log2;
-      // at most 4 column passes fit the scatter descriptor; distribute the
-      // bits evenly so the wide (register-hungry) factors are as small as
-      // possible: e.g. 21 bits -> [64,32,32,32], 10 -> [32,32]
-      int ncols = (rest + maxcol_log2 - 1) / maxcol_log2;
-      if (ncols > 4) ncols = 4;  // widen factors instead (up to 64 each)
-      const int base_b = rest / ncols, extra = rest % ncols;
-      for (int i = 0; i < ncols; ++i) {
-        const int b = base_b + (i < extra ? 1 : 0);
-        if (b > 6) throw std::runtime_error("NativeFft: factor too large");
-        f.push_back(1u << b);
+      if (rest >= 18 && rest <= 24) {
+        // large residual: greedy 64s + one remainder factor.  The N=64
+        // lane-pair kernel is the fastest pass per 8 GB sweep (4 waves/SIMD,
+        // measured 2.2 vs 2.75 ms under contention), and 18..20 bits save a
+        // whole pass vs balanced 32s: 2^29 fwd [64,64,64,8] measured 9.41 ms
+        // vs [64,32,32,32] 9.88 (fft_factor_sweep.py).
+        int left = rest;
+        while (left >= 6 && (int)f.size() < 4) {
+          f.push_back(64);
+          left -= 6;
+        }
+        if (left > 0) f.push_back(1u << left);
+      } else {
+        // small residual: distribute evenly over 32-max columns (measured
+        // best at the 2^18 waterfall: [32,32] beats [64,16])
+        int ncols = (rest + maxcol_log2 - 1) / maxcol_log2;
+        if (ncols > 4) ncols = 4;  // widen factors instead (up to 64 each)
+        const int base_b = rest / ncols, extra = rest % ncols;
+        for (int i = 0; i < ncols; ++i) {
+          const int b = base_b + (i < extra ? 1 : 0);
+          if (b > 6) throw std::runtime_error("NativeFft: factor too large");
+          f.push_back(1u << b);
+        }
       }
       f.push_back(1u << final_log2);
     }
