@@ -489,8 +489,9 @@ __global__ void __launch_bounds__(256)
           x.x *= pre.norm_coeff;
           x.y *= pre.norm_coeff;
         }
-        x = cmulf(x, srtb_dedisp_factor(flat, pre.f_min, pre.f_c, pre.df,
-                                        pre.dm));
+        x = cmulf(x, pre.table ? pre.table[flat]
+                               : srtb_dedisp_factor(flat, pre.f_min, pre.f_c,
+                                                    pre.df, pre.dm));
       }
     }
     v[r] = x;
@@ -524,6 +525,99 @@ __global__ void __launch_bounds__(256)
 #pragma unroll
   for (int r = 0; r < 32; ++r) {
     const int k = 32 * p + r;
+    float2 rr = v[r];
+    if constexpr (TWIDDLE) {
+      const unsigned long long m_ =
+          (q0 * d.tw_f0 * (unsigned long long)k) & d.tw_mask;
+      rr = cmulf(rr, tw_eval(m_, d.tw_angle));
+    }
+    ocolbase[(uint32_t)k * stride32] = rr;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// N = 32 column pass, two lanes per column (same construction as pair64):
+// lane p holds post-σ positions 16p..16p+15, stages L=4 and L=16 are
+// lane-local, the final radix-2 stage (L=32) exchanges halves with one
+// shfl_xor(…,1) per butterfly.  16 data registers per lane → higher
+// occupancy than the mono col<32> kernel (~143 VGPRs, 3 waves/SIMD),
+// which matters most for the VALU-heavy fused-preop (fp64 dedispersion)
+// backward pass.
+// ---------------------------------------------------------------------------
+
+constexpr int col_sigma_inv32(int pos) {
+  for (int i = 0; i < 32; ++i)
+    if (col_sigma<32>(i) == pos) return i;
+  return 0;
+}
+
+template <bool TWIDDLE, int SIGN, bool PREOP, int DEC = 0>
+__global__ void __launch_bounds__(256)
+    k_fft_col_pair32(const float2* __restrict__ in, float2* __restrict__ out,
+                     FftPassDescDev d, unsigned long long n_ffts,
+                     const float2* __restrict__ tw_n,
+                     const float2* __restrict__ tw_hi,
+                     const float2* __restrict__ tw_lo, FftPreopDev pre,
+                     const uint8_t* __restrict__ raw2) {
+  const unsigned long long tid =
+      (unsigned long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const unsigned long long id = tid >> 1;  // column index
+  const int p = (int)(tid & 1);            // which half of the column
+  if (id >= n_ffts) return;
+  unsigned long long q0, q1, q2;
+  digits(id, d, q0, q1, q2);
+  const unsigned long long base = q0 * d.in_c0 + q1 * d.in_c1 + q2 * d.in_c2;
+  float thr_mean = 0.f;
+  if constexpr (PREOP) {
+    if (pre.mean_power) thr_mean = pre.threshold * (float)(*pre.mean_power);
+  }
+  float2 v[16];
+  const float2* __restrict__ colbase = in + base;
+  const uint32_t stride32 = (uint32_t)d.in_stride;
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int i = p ? col_sigma_inv32(16 + r) : col_sigma_inv32(r);
+    const uint32_t off = (uint32_t)i * stride32;
+    float2 x;
+    if constexpr (DEC != 0) x = decN_load<DEC>(raw2, base + off);
+    else x = colbase[off];
+    if constexpr (PREOP) {
+      const unsigned long long flat = base + off;
+      bool zap = pre.mean_power && (norm2(x) > thr_mean);
+      for (int z = 0; z < pre.n_zap; ++z)
+        zap |= (flat >= pre.zap[z].lo) & (flat <= pre.zap[z].hi);
+      if (zap) {
+        x = make_float2(0.f, 0.f);
+      } else {
+        if (pre.mean_power) {
+          x.x *= pre.norm_coeff;
+          x.y *= pre.norm_coeff;
+        }
+        x = cmulf(x, pre.table ? pre.table[flat]
+                               : srtb_dedisp_factor(flat, pre.f_min, pre.f_c,
+                                                    pre.df, pre.dm));
+      }
+    }
+    v[r] = x;
+  }
+  // stages 1-2 lane-local; twiddle scale of the FULL length-32 transform
+  col_stage<16, 4, 4, SIGN, 8>(v, tw_n);
+  col_stage<16, 16, 4, SIGN, 2>(v, tw_n);
+  // final radix-2 stage (L=32, M=16, TS=1): pairs {j, j+16}
+#pragma unroll
+  for (int j = 0; j < 16; ++j) {
+    const float2 other = shfl_xor1(v[j]);
+    const float2 a = p ? other : v[j];
+    const float2 b = p ? v[j] : other;
+    const float2 w = (j == 0) ? make_float2(1.f, 0.f) : tw_n[j];
+    const float2 y = cmulf(b, w);
+    v[j] = p ? make_float2(a.x - y.x, a.y - y.y)
+             : make_float2(a.x + y.x, a.y + y.y);
+  }
+  float2* __restrict__ ocolbase = out + base;
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int k = 16 * p + r;
     float2 rr = v[r];
     if constexpr (TWIDDLE) {
       const unsigned long long m_ =
@@ -941,12 +1035,45 @@ hipError_t fft_col_pass(const float2* in, float2* out, const FftPassDesc& hd,
   hipLaunchKernelGGL((k_fft_col_pair64<true, SG, false, B>), dim3(grid2),    \
                      dim3(256), 0, stream, in, out, d, n_ffts, tw_n, tw_hi,  \
                      tw_lo, pre, raw2)
+#define COL_LAUNCH_P32(TW, SG, PR)                                           \
+  hipLaunchKernelGGL((k_fft_col_pair32<TW, SG, PR>), dim3(grid2), dim3(256), \
+                     0, stream, in, out, d, n_ffts, tw_n, tw_hi, tw_lo, pre, \
+                     raw2)
+#define COL_LAUNCH_P32D(SG, B)                                               \
+  hipLaunchKernelGGL((k_fft_col_pair32<true, SG, false, B>), dim3(grid2),    \
+                     dim3(256), 0, stream, in, out, d, n_ffts, tw_n, tw_hi,  \
+                     tw_lo, pre, raw2)
   switch (hd.n) {
     COL_DISPATCH(2)
     COL_DISPATCH(4)
     COL_DISPATCH(8)
     COL_DISPATCH(16)
-    COL_DISPATCH(32)
+    case 32:
+      if (raw2) {
+        if (raw_bits == 1) { if (sign < 0) COL_LAUNCH_P32D(-1, 1);
+                             else COL_LAUNCH_P32D(1, 1); }
+        else if (raw_bits == 2) { if (sign < 0) COL_LAUNCH_P32D(-1, 2);
+                                  else COL_LAUNCH_P32D(1, 2); }
+        else if (raw_bits == 4) { if (sign < 0) COL_LAUNCH_P32D(-1, 4);
+                                  else COL_LAUNCH_P32D(1, 4); }
+        else if (raw_bits == 8) { if (sign < 0) COL_LAUNCH_P32D(-1, 8);
+                                  else COL_LAUNCH_P32D(1, 8); }
+        else { if (sign < 0) COL_LAUNCH_P32D(-1, -8);
+               else COL_LAUNCH_P32D(1, -8); }
+        break;
+      }
+      if (twiddle) {
+        if (sign < 0) { if (preop) COL_LAUNCH_P32(true, -1, true);
+                        else COL_LAUNCH_P32(true, -1, false); }
+        else          { if (preop) COL_LAUNCH_P32(true, 1, true);
+                        else COL_LAUNCH_P32(true, 1, false); }
+      } else {
+        if (sign < 0) { if (preop) COL_LAUNCH_P32(false, -1, true);
+                        else COL_LAUNCH_P32(false, -1, false); }
+        else          { if (preop) COL_LAUNCH_P32(false, 1, true);
+                        else COL_LAUNCH_P32(false, 1, false); }
+      }
+      break;
     case 64:
       if (raw2) {
         if (raw_bits == 1) { if (sign < 0) COL_LAUNCH_PD(-1, 1);
@@ -982,6 +1109,8 @@ hipError_t fft_col_pass(const float2* in, float2* out, const FftPassDesc& hd,
 #undef COL_LAUNCH_D
 #undef COL_LAUNCH_P
 #undef COL_LAUNCH_PD
+#undef COL_LAUNCH_P32
+#undef COL_LAUNCH_P32D
   SRTB_CHECK_LAUNCH();
   return hipSuccess;
 }
