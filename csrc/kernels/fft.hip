@@ -27,6 +27,7 @@
 //    strided passes instances are q0-consecutive so a row of F elements is
 //    contiguous in HBM.
 
+#include <cstdlib>
 #include "common.h"
 #include "../include/srtb_kernels.h"
 
@@ -951,6 +952,11 @@ hipError_t fft_stockham_pass(const float2* in, float2* out,
   return hipSuccess;
 }
 
+static bool use_pair32() {
+  static const bool v = std::getenv("SRTB_FFT_PAIR32") != nullptr;
+  return v;
+}
+
 hipError_t fft_col_pass(const float2* in, float2* out, const FftPassDesc& hd,
                         size_t n_ffts, int sign, const float2* tw_n,
                         const float2* tw_hi, const float2* tw_lo,
@@ -1048,30 +1054,51 @@ hipError_t fft_col_pass(const float2* in, float2* out, const FftPassDesc& hd,
     COL_DISPATCH(4)
     COL_DISPATCH(8)
     COL_DISPATCH(16)
+    // N = 32: the lane-pair kernel (6 waves/SIMD) measured end-to-end
+    // neutral vs the mono kernel (3 waves) and is opt-in via
+    // SRTB_FFT_PAIR32=1 pending more soak coverage; mono is the default.
     case 32:
+      if (use_pair32()) {
+        if (raw2) {
+          if (raw_bits == 1) { if (sign < 0) COL_LAUNCH_P32D(-1, 1);
+                               else COL_LAUNCH_P32D(1, 1); }
+          else if (raw_bits == 2) { if (sign < 0) COL_LAUNCH_P32D(-1, 2);
+                                    else COL_LAUNCH_P32D(1, 2); }
+          else if (raw_bits == 4) { if (sign < 0) COL_LAUNCH_P32D(-1, 4);
+                                    else COL_LAUNCH_P32D(1, 4); }
+          else if (raw_bits == 8) { if (sign < 0) COL_LAUNCH_P32D(-1, 8);
+                                    else COL_LAUNCH_P32D(1, 8); }
+          else { if (sign < 0) COL_LAUNCH_P32D(-1, -8);
+                 else COL_LAUNCH_P32D(1, -8); }
+          break;
+        }
+        if (twiddle) {
+          if (sign < 0) { if (preop) COL_LAUNCH_P32(true, -1, true);
+                          else COL_LAUNCH_P32(true, -1, false); }
+          else          { if (preop) COL_LAUNCH_P32(true, 1, true);
+                          else COL_LAUNCH_P32(true, 1, false); }
+        } else {
+          if (sign < 0) { if (preop) COL_LAUNCH_P32(false, -1, true);
+                          else COL_LAUNCH_P32(false, -1, false); }
+          else          { if (preop) COL_LAUNCH_P32(false, 1, true);
+                          else COL_LAUNCH_P32(false, 1, false); }
+        }
+        break;
+      }
       if (raw2) {
-        if (raw_bits == 1) { if (sign < 0) COL_LAUNCH_P32D(-1, 1);
-                             else COL_LAUNCH_P32D(1, 1); }
-        else if (raw_bits == 2) { if (sign < 0) COL_LAUNCH_P32D(-1, 2);
-                                  else COL_LAUNCH_P32D(1, 2); }
-        else if (raw_bits == 4) { if (sign < 0) COL_LAUNCH_P32D(-1, 4);
-                                  else COL_LAUNCH_P32D(1, 4); }
-        else if (raw_bits == 8) { if (sign < 0) COL_LAUNCH_P32D(-1, 8);
-                                  else COL_LAUNCH_P32D(1, 8); }
-        else { if (sign < 0) COL_LAUNCH_P32D(-1, -8);
-               else COL_LAUNCH_P32D(1, -8); }
+        COL_DISPATCH_D(32)
         break;
       }
       if (twiddle) {
-        if (sign < 0) { if (preop) COL_LAUNCH_P32(true, -1, true);
-                        else COL_LAUNCH_P32(true, -1, false); }
-        else          { if (preop) COL_LAUNCH_P32(true, 1, true);
-                        else COL_LAUNCH_P32(true, 1, false); }
+        if (sign < 0) { if (preop) COL_LAUNCH(32, true, -1, true);
+                        else COL_LAUNCH(32, true, -1, false); }
+        else          { if (preop) COL_LAUNCH(32, true, 1, true);
+                        else COL_LAUNCH(32, true, 1, false); }
       } else {
-        if (sign < 0) { if (preop) COL_LAUNCH_P32(false, -1, true);
-                        else COL_LAUNCH_P32(false, -1, false); }
-        else          { if (preop) COL_LAUNCH_P32(false, 1, true);
-                        else COL_LAUNCH_P32(false, 1, false); }
+        if (sign < 0) { if (preop) COL_LAUNCH(32, false, -1, true);
+                        else COL_LAUNCH(32, false, -1, false); }
+        else          { if (preop) COL_LAUNCH(32, false, 1, true);
+                        else COL_LAUNCH(32, false, 1, false); }
       }
       break;
     case 64:
