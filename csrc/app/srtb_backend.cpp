@@ -299,7 +299,7 @@ int main(int argc, char** argv) {
   stop.store(true);
   if (input_thread.joinable()) input_thread.join();
   engine.synchronize();
-  for (auto b : bufs) hipHostFree(b);
+  for (auto b : bufs) (void)hipHostFree(b);
 
   SRTB_APP_LOGI("done: " << blocks << " blocks, " << detections
                          << " with detections");

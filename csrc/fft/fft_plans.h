@@ -103,9 +103,9 @@ class FftPlanSet {
   size_t work_size() const { return work_size_; }
 
   void destroy() {
-    if (r2c_) hipfftDestroy(r2c_), r2c_ = 0;
-    if (c2c_) hipfftDestroy(c2c_), c2c_ = 0;
-    if (work_area_) hipFree(work_area_), work_area_ = nullptr;
+    if (r2c_) (void)hipfftDestroy(r2c_), r2c_ = 0;
+    if (c2c_) (void)hipfftDestroy(c2c_), c2c_ = 0;
+    if (work_area_) (void)hipFree(work_area_), work_area_ = nullptr;
     work_size_ = 0;
   }
 

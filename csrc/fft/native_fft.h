@@ -265,7 +265,7 @@ class NativeFft {
   }
 
   void destroy() {
-    for (auto& t : tables_) hipFree(t.ptr);
+    for (auto& t : tables_) (void)hipFree(t.ptr);
     tables_.clear();
     passes_.clear();
   }
