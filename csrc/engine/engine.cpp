@@ -145,35 +145,35 @@ PipelineEngine::PipelineEngine(const EngineConfig& cfg, int n_slots)
 
 PipelineEngine::~PipelineEngine() {
   for (auto& sp : slots_) {
-    if (sp->stream) hipStreamSynchronize(sp->stream);
+    if (sp->stream) (void)hipStreamSynchronize(sp->stream);
   }
   for (auto& sp : slots_) {
     Slot& s = *sp;
     s.plans.destroy();
-    hipFree(s.raw);
-    hipFree(s.samples);
-    hipFree(s.spec);
-    hipFree(s.s2s4);
-    if (s.sk_dif_partials) hipFree(s.sk_dif_partials);
-    hipFree(s.flags);
-    hipFree(s.ts);
-    hipFree(s.ts_partial);
-    hipFree(s.cumsum);
-    hipFree(s.box);
-    hipFree(s.scan_scratch);
-    hipFree(s.partials);
-    hipFree(s.mean_power);
-    hipFree(s.sums);
-    hipFree(s.counters);
-    hipFree(s.thresholds);
-    hipHostFree(s.h_counters);
-    hipHostFree(s.h_thresholds);
-    if (s.graph_exec) hipGraphExecDestroy(s.graph_exec);
-    if (s.done) hipEventDestroy(s.done);
-    if (s.stream) hipStreamDestroy(s.stream);
+    (void)hipFree(s.raw);
+    (void)hipFree(s.samples);
+    (void)hipFree(s.spec);
+    (void)hipFree(s.s2s4);
+    if (s.sk_dif_partials) (void)hipFree(s.sk_dif_partials);
+    (void)hipFree(s.flags);
+    (void)hipFree(s.ts);
+    (void)hipFree(s.ts_partial);
+    (void)hipFree(s.cumsum);
+    (void)hipFree(s.box);
+    (void)hipFree(s.scan_scratch);
+    (void)hipFree(s.partials);
+    (void)hipFree(s.mean_power);
+    (void)hipFree(s.sums);
+    (void)hipFree(s.counters);
+    (void)hipFree(s.thresholds);
+    (void)hipHostFree(s.h_counters);
+    (void)hipHostFree(s.h_thresholds);
+    if (s.graph_exec) (void)hipGraphExecDestroy(s.graph_exec);
+    if (s.done) (void)hipEventDestroy(s.done);
+    if (s.stream) (void)hipStreamDestroy(s.stream);
   }
-  if (phase_table_) hipFree(phase_table_);
-  if (window_) hipFree(window_);
+  if (phase_table_) (void)hipFree(phase_table_);
+  if (window_) (void)hipFree(window_);
 }
 
 void PipelineEngine::enqueue_chain(Slot& s, const uint8_t* dev_raw,
