@@ -193,7 +193,8 @@ void PipelineEngine::enqueue_chain(Slot& s, const uint8_t* dev_raw,
   const int in_bits = cfg_.baseband_input_bits;
   const bool fuse_unpack = dev_raw && native_fft_ && !fused_unpack_off_ &&
                            (in_bits == 1 || in_bits == 2 || in_bits == 4 ||
-                            in_bits == 8 || in_bits == -8) &&
+                            in_bits == 8 || in_bits == -8 ||
+                            in_bits == 16 || in_bits == -16) &&
                            !window_ && slots_[0]->nfwd.first_pass_fusable();
   if (dev_raw && !fuse_unpack) {
     // 1. unpack (+ window fused; default rectangle → none)

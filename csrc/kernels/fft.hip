@@ -98,6 +98,12 @@ __device__ inline float2 decN_load(const uint8_t* __restrict__ raw,
   } else if constexpr (NBITS == -8) {
     const int8_t* r8 = reinterpret_cast<const int8_t*>(raw);
     return make_float2((float)r8[2 * flat], (float)r8[2 * flat + 1]);
+  } else if constexpr (NBITS == 16) {
+    const uint16_t* r16 = reinterpret_cast<const uint16_t*>(raw);
+    return make_float2((float)r16[2 * flat], (float)r16[2 * flat + 1]);
+  } else if constexpr (NBITS == -16) {
+    const int16_t* r16 = reinterpret_cast<const int16_t*>(raw);
+    return make_float2((float)r16[2 * flat], (float)r16[2 * flat + 1]);
   } else {  // NBITS == 1
     const uint32_t bv = raw[flat >> 2];
     const int j0 = (int)(flat & 3) * 2;
@@ -964,7 +970,7 @@ hipError_t fft_col_pass(const float2* in, float2* out, const FftPassDesc& hd,
                         const uint8_t* raw2, int raw_bits) {
   if (raw2 && (preop || hd.tw_mod == 0)) return hipErrorInvalidValue;
   if (raw2 && raw_bits != 1 && raw_bits != 2 && raw_bits != 4 &&
-      raw_bits != 8 && raw_bits != -8)
+      raw_bits != 8 && raw_bits != -8 && raw_bits != 16 && raw_bits != -16)
     return hipErrorInvalidValue;
   FftPreopDev pre{};
   if (preop) {
@@ -1013,8 +1019,12 @@ hipError_t fft_col_pass(const float2* in, float2* out, const FftPassDesc& hd,
                             else COL_LAUNCH_D(N, 1, 4); }                    \
   else if (raw_bits == 8) { if (sign < 0) COL_LAUNCH_D(N, -1, 8);            \
                             else COL_LAUNCH_D(N, 1, 8); }                    \
-  else { if (sign < 0) COL_LAUNCH_D(N, -1, -8);                              \
-         else COL_LAUNCH_D(N, 1, -8); }
+  else if (raw_bits == -8) { if (sign < 0) COL_LAUNCH_D(N, -1, -8);          \
+                             else COL_LAUNCH_D(N, 1, -8); }                  \
+  else if (raw_bits == 16) { if (sign < 0) COL_LAUNCH_D(N, -1, 16);          \
+                             else COL_LAUNCH_D(N, 1, 16); }                  \
+  else { if (sign < 0) COL_LAUNCH_D(N, -1, -16);                             \
+         else COL_LAUNCH_D(N, 1, -16); }
 #define COL_DISPATCH(N)                                                      \
   case N:                                                                    \
     if (raw2) {                                                              \
@@ -1068,8 +1078,12 @@ hipError_t fft_col_pass(const float2* in, float2* out, const FftPassDesc& hd,
                                     else COL_LAUNCH_P32D(1, 4); }
           else if (raw_bits == 8) { if (sign < 0) COL_LAUNCH_P32D(-1, 8);
                                     else COL_LAUNCH_P32D(1, 8); }
-          else { if (sign < 0) COL_LAUNCH_P32D(-1, -8);
-                 else COL_LAUNCH_P32D(1, -8); }
+          else if (raw_bits == -8) { if (sign < 0) COL_LAUNCH_P32D(-1, -8);
+                                     else COL_LAUNCH_P32D(1, -8); }
+          else if (raw_bits == 16) { if (sign < 0) COL_LAUNCH_P32D(-1, 16);
+                                     else COL_LAUNCH_P32D(1, 16); }
+          else { if (sign < 0) COL_LAUNCH_P32D(-1, -16);
+                 else COL_LAUNCH_P32D(1, -16); }
           break;
         }
         if (twiddle) {
@@ -1111,8 +1125,12 @@ hipError_t fft_col_pass(const float2* in, float2* out, const FftPassDesc& hd,
                                   else COL_LAUNCH_PD(1, 4); }
         else if (raw_bits == 8) { if (sign < 0) COL_LAUNCH_PD(-1, 8);
                                   else COL_LAUNCH_PD(1, 8); }
-        else { if (sign < 0) COL_LAUNCH_PD(-1, -8);
-               else COL_LAUNCH_PD(1, -8); }
+        else if (raw_bits == -8) { if (sign < 0) COL_LAUNCH_PD(-1, -8);
+                                   else COL_LAUNCH_PD(1, -8); }
+        else if (raw_bits == 16) { if (sign < 0) COL_LAUNCH_PD(-1, 16);
+                                   else COL_LAUNCH_PD(1, 16); }
+        else { if (sign < 0) COL_LAUNCH_PD(-1, -16);
+               else COL_LAUNCH_PD(1, -16); }
         break;
       }
       if (twiddle) {

@@ -139,13 +139,14 @@ def test_engine_subbyte_paths(C, bits):
     assert abs(int(np.argmax(ts)) - expect_bin) <= 2
 
 
-def test_engine_subbyte_fused_matches_unfused(C, monkeypatch):
+@pytest.mark.parametrize("bits", [2, -8, -16])
+def test_engine_subbyte_fused_matches_unfused(C, monkeypatch, bits):
     """The fused decode-at-load path must produce the same time series as
     the standalone unpack kernel path (SRTB_NO_FUSED_UNPACK)."""
-    cfg = small_cfg(bits=2)
+    cfg = small_cfg(bits=bits)
     rng = np.random.default_rng(7)
-    raw = rng.integers(0, 256, cfg.baseband_input_count // 4,
-                       dtype=np.uint8)
+    nbytes = cfg.baseband_input_count * abs(bits) // 8
+    raw = rng.integers(0, 256, nbytes, dtype=np.uint8)
     eng = make_engine(C, cfg)
     slot = eng.submit(torch.from_numpy(raw.copy()))
     eng.wait(slot)
