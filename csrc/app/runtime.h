@@ -18,6 +18,10 @@
 #include <sched.h>
 #include <sstream>
 #include <string>
+#include <condition_variable>
+#include <deque>
+#include <functional>
+#include <vector>
 #include <thread>
 #include <unistd.h>
 
@@ -162,5 +166,83 @@ inline bool set_thread_affinity(int cpu) {
 inline void set_thread_name(const char* name) {
   pthread_setname_np(pthread_self(), name);  // <=15 chars
 }
+
+// ---------------- async write pool ----------------
+// The reference posts baseband/spectrum/time-series product writes onto
+// boost::asio::thread_pools (write_signal_pipe.hpp:55-57,159,210,249) so a
+// detection dump never stalls the streaming pipeline; this is the plain-C++
+// equivalent.  Destructor drains the queue before joining (writes are never
+// lost at shutdown).
+class WritePool {
+ public:
+  explicit WritePool(int n_threads = 2) {
+    for (int i = 0; i < n_threads; ++i)
+      threads_.emplace_back([this] {
+        set_thread_name("srtb-writer");
+        run();
+      });
+  }
+  WritePool(const WritePool&) = delete;
+  WritePool& operator=(const WritePool&) = delete;
+
+  ~WritePool() {
+    {
+      std::unique_lock<std::mutex> lk(m_);
+      done_ = true;
+    }
+    cv_.notify_all();
+    for (auto& t : threads_) t.join();
+  }
+
+  void post(std::function<void()> f) {
+    {
+      std::unique_lock<std::mutex> lk(m_);
+      q_.push_back(std::move(f));
+    }
+    cv_.notify_one();
+  }
+
+  // Block until every posted task has finished (for tests / clean exits
+  // that must observe the files).
+  void drain() {
+    std::unique_lock<std::mutex> lk(m_);
+    idle_cv_.wait(lk, [this] { return q_.empty() && active_ == 0; });
+  }
+
+ private:
+  void run() {
+    for (;;) {
+      std::function<void()> f;
+      {
+        std::unique_lock<std::mutex> lk(m_);
+        cv_.wait(lk, [this] { return done_ || !q_.empty(); });
+        if (q_.empty()) {
+          if (done_) return;
+          continue;
+        }
+        f = std::move(q_.front());
+        q_.pop_front();
+        ++active_;
+      }
+      try {
+        f();
+      } catch (const std::exception& e) {
+        SRTB_APP_LOGE("write pool: " << e.what());
+      }
+      {
+        std::unique_lock<std::mutex> lk(m_);
+        --active_;
+        if (q_.empty() && active_ == 0) idle_cv_.notify_all();
+      }
+    }
+  }
+
+  std::mutex m_;
+  std::condition_variable cv_, idle_cv_;
+  std::deque<std::function<void()>> q_;
+  int active_ = 0;
+  bool done_ = false;
+  std::vector<std::thread> threads_;
+};
 
 }  // namespace srtb_app

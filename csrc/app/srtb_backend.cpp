@@ -231,9 +231,13 @@ int main(int argc, char** argv) {
   struct InFlight { int slot; int buf_index; uint64_t counter; };
   std::vector<InFlight> inflight;
   const size_t S = engine.n_channels(), Lw = engine.waterfall_len();
-  std::vector<std::complex<float>> h_wf;
-  std::vector<float> h_ts;
   uint64_t blocks = 0, detections = 0;
+  // product writes go to a small pool (reference posts them to asio
+  // thread_pools) so a detection dump never stalls block submission;
+  // device buffers are staged to host HERE (they recycle on next submit),
+  // file IO + fdatasync happen on the pool
+  srtb_app::WritePool writers(2);
+  const std::string out_prefix = cfg.baseband_output_file_prefix;
 
   auto drain_one = [&] {
     const InFlight w = inflight.front();
@@ -250,18 +254,26 @@ int main(int argc, char** argv) {
       SRTB_APP_LOGI("detection in block " << w.counter << " ("
                                           << positive << " samples over "
                                           << "threshold)");
-      // raw baseband still lives in its pinned buffer (recycled only after
-      // the q_free push below)
-      write_baseband_bin(cfg.baseband_output_file_prefix, w.counter,
-                         bufs[w.buf_index], raw_bytes);
-      h_wf.resize(S * Lw);
+      // raw baseband lives in a pinned buffer that recycles after the
+      // q_free push below — copy it out for the async write
+      {
+        std::vector<uint8_t> bb(bufs[w.buf_index],
+                                bufs[w.buf_index] + raw_bytes);
+        writers.post([out_prefix, counter = w.counter,
+                      bb = std::move(bb)] {
+          write_baseband_bin(out_prefix, counter, bb.data(), bb.size());
+        });
+      }
+      std::vector<std::complex<float>> h_wf(S * Lw);
       srtb_hip::check_hip(
           hipMemcpy(h_wf.data(), engine.waterfall_ptr(w.slot),
                     S * Lw * sizeof(float2), hipMemcpyDeviceToHost),
           "wf d2h");
-      write_spectrum_npy(cfg.baseband_output_file_prefix, w.counter,
-                         h_wf.data(), S, Lw);
-      h_ts.resize(engine.ts_count());
+      writers.post([out_prefix, counter = w.counter, S, Lw,
+                    wf = std::move(h_wf)] {
+        write_spectrum_npy(out_prefix, counter, wf.data(), S, Lw);
+      });
+      std::vector<float> h_ts(engine.ts_count());
       srtb_hip::check_hip(
           hipMemcpy(h_ts.data(), engine.time_series_ptr(w.slot),
                     engine.ts_count() * sizeof(float),
@@ -270,8 +282,10 @@ int main(int argc, char** argv) {
       for (auto& [len, cnt] : res.counts) {
         if (cnt == 0) continue;
         if (len == 1) {
-          write_time_series_tim(cfg.baseband_output_file_prefix, w.counter, 1,
-                                h_ts.data(), h_ts.size());
+          writers.post([out_prefix, counter = w.counter, ts = h_ts] {
+            write_time_series_tim(out_prefix, counter, 1, ts.data(),
+                                  ts.size());
+          });
         } else {
           float* box = engine.compute_boxcar(w.slot, len);
           std::vector<float> h_box(engine.ts_count() - len);
@@ -279,8 +293,11 @@ int main(int argc, char** argv) {
                                         h_box.size() * sizeof(float),
                                         hipMemcpyDeviceToHost),
                               "box d2h");
-          write_time_series_tim(cfg.baseband_output_file_prefix, w.counter,
-                                len, h_box.data(), h_box.size());
+          writers.post([out_prefix, counter = w.counter, len = len,
+                        bx = std::move(h_box)] {
+            write_time_series_tim(out_prefix, counter, len, bx.data(),
+                                  bx.size());
+          });
         }
       }
     }
@@ -299,6 +316,7 @@ int main(int argc, char** argv) {
   stop.store(true);
   if (input_thread.joinable()) input_thread.join();
   engine.synchronize();
+  writers.drain();  // all product files on disk before the summary line
   for (auto b : bufs) (void)hipHostFree(b);
 
   SRTB_APP_LOGI("done: " << blocks << " blocks, " << detections
