@@ -71,7 +71,8 @@ class SignalWriteScheduler:
     """
 
     def __init__(self, prefix: str, block_samples: int, sample_rate: float,
-                 real_time: bool = True, max_pending: int = 8):
+                 real_time: bool = True, max_pending: int = 8,
+                 async_writes: bool = False):
         self.prefix = prefix
         self.real_time = real_time
         self.overlap_window_ns = 0.45 * 1e9 * block_samples / sample_rate
@@ -79,12 +80,27 @@ class SignalWriteScheduler:
         self.pending_negative: deque[BlockProducts] = deque()
         self.max_pending = max_pending
         self.written: list[str] = []
+        # product writes off the pipeline thread (reference posts them to
+        # asio thread_pools, write_signal_pipe.hpp:55-57); call close() (or
+        # use as a context manager) to flush
+        self._pool = None
+        if async_writes:
+            from concurrent.futures import ThreadPoolExecutor
+            self._pool = ThreadPoolExecutor(max_workers=2,
+                                            thread_name_prefix="srtb-writer")
+            self._futures = []
 
     def _overlaps_positive(self, ts: int) -> bool:
         return any(abs(ts - t) < self.overlap_window_ns
                    for t in self.recent_positive)
 
     def _write(self, blk: BlockProducts) -> None:
+        if self._pool is not None:
+            self._futures.append(self._pool.submit(self._write_sync, blk))
+            return
+        self._write_sync(blk)
+
+    def _write_sync(self, blk: BlockProducts) -> None:
         if blk.raw is not None:
             self.written.append(
                 write_baseband_bin(self.prefix, blk.counter, blk.raw))
@@ -123,6 +139,21 @@ class SignalWriteScheduler:
                   cand.timestamp + self.overlap_window_ns <
                   self.recent_positive[-1] - 5 * self.overlap_window_ns):
                 self.pending_negative.popleft()  # can never match anymore
+
+    def close(self) -> None:
+        """Flush and join the async write pool (no-op in sync mode)."""
+        if self._pool is not None:
+            for f in self._futures:
+                f.result()
+            self._pool.shutdown(wait=True)
+            self._pool = None
+
+    def __enter__(self):
+        return self
+
+    def __exit__(self, *exc):
+        self.close()
+        return False
 
 
 # ---------------------------------------------------------------------------

@@ -224,3 +224,20 @@ def test_filterbank_header_roundtrip(tmp_path):
 def test_to_sigproc_dms():
     assert wr.to_sigproc_dms(12.5) == pytest.approx(123000.0)
     assert wr.to_sigproc_dms(-1.25) == pytest.approx(-11500.0)
+
+
+def test_signal_write_scheduler_async(tmp_path):
+    """async_writes=True posts product writes to a pool; close() flushes —
+    files must exist and match the sync-mode output."""
+    import numpy as np
+    from srtb_amd.io.writers import BlockProducts, SignalWriteScheduler
+    raw = np.arange(64, dtype=np.uint8)
+    ts = np.linspace(0, 1, 32, dtype=np.float32)
+    with SignalWriteScheduler(str(tmp_path) + "/a_", 1 << 20, 1e9,
+                              async_writes=True) as sch:
+        sch.push(BlockProducts(counter=7, timestamp=0, raw=raw,
+                               waterfall=None, time_series=[(1, ts)]))
+    files = sorted(p.name for p in tmp_path.iterdir())
+    assert files == ["a_7.1.tim", "a_7.bin"]
+    got = np.fromfile(tmp_path / "a_7.bin", dtype=np.uint8)
+    np.testing.assert_array_equal(got, raw)
