@@ -101,7 +101,7 @@ class NativeFft {
         f.clear();  // invalid override for this length: fall through
       else {
         for (size_t i = 0; i + 1 < f.size(); ++i)
-          if (f[i] > 64) f.clear();
+          if (f[i] > 64 && f[i] != 512) f.clear();  // 512 = LDS mid pass
       }
     }
     if (!f.empty()) {
@@ -156,7 +156,7 @@ class NativeFft {
       ensure_len_table(f[j], sign, stream);
       ensure_mod_table(M[j], sign, stream);
       Pass p;
-      p.kind = PassKind::kCol;
+      p.kind = (f[j] == 512) ? PassKind::kMid : PassKind::kCol;
       p.d.n = f[j];
       p.d.d0 = (uint32_t)M[j + 1];
       p.d.d1 = (uint32_t)Cp[j];
@@ -252,6 +252,12 @@ class NativeFft {
                     "fft_col_pass");
           dst = cur;  // in place
           break;
+        case PassKind::kMid:
+          check_hip(fft_mid512_pass(cur, cur, p.d, p.n_ffts, sign_, p.tw_n,
+                                    stream),
+                    "fft_mid512_pass");
+          dst = cur;  // in place
+          break;
         case PassKind::kDif: {
           const int F = dif_f(p);
           check_hip(fft_dif_final(cur, dst, p.dif, p.n_ffts, F, sign_,
@@ -271,7 +277,7 @@ class NativeFft {
   }
 
  private:
-  enum class PassKind { kStockham, kCol, kDif };
+  enum class PassKind { kStockham, kCol, kMid, kDif };
 
   struct Pass {
     PassKind kind = PassKind::kStockham;

@@ -233,6 +233,12 @@ struct DifFinalDesc {
 // sk_partials[row * wgs_per_row + wg_in_row] (deterministic, no atomics) —
 // this replaces the separate 4 GB sk_row_stats read.  wgs_per_row =
 // (L/n)/F.
+// EXPERIMENTAL strided 512-point LDS middle pass (see fft.hip); reachable
+// only via the SRTB_FFT_FACTORS plan override.
+hipError_t fft_mid512_pass(const float2* in, float2* out,
+                           const FftPassDesc& d, size_t n_ffts, int sign,
+                           const float2* tw_n, hipStream_t stream);
+
 hipError_t fft_dif_final(const float2* in, float2* out, const DifFinalDesc& d,
                          size_t n_ffts, int F, int sign, const float2* tw_n,
                          float2* sk_partials, hipStream_t stream);

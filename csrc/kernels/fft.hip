@@ -836,6 +836,131 @@ __global__ void k_build_twiddle(float2* __restrict__ t, size_t count,
 // (x may alias z).  k=0: X[0]=Re(Z0)+Im(Z0) (Nyquist dropped, count = Nc).
 // Optionally accumulates Σ|X|² partials (fused RFI-s1 mean-power — saves a
 // full 4 GB spectrum read).
+// ---------------------------------------------------------------------------
+// EXPERIMENTAL (round-2 lever, reachable only via SRTB_FFT_FACTORS):
+// strided middle pass of length 512 through LDS — replaces TWO register-
+// column passes, cutting a whole 8 GB sweep from the 2^29 plan
+// ([64,64,512]+256 = 4 passes vs [64,64,64,8]+256 = 5).
+//
+// 512 = 2·256: stage A is the radix-2 DIF split (odd half twiddled by
+// W512^i), then two independent in-place radix-4 DIF-256 per column (the
+// validated k_fft_dif_final butterflies + swizzle), and the store reads
+// LDS at 256·(k&1) + swz(rev4(k>>1)) so output lands in natural order.
+// F = 32 columns per workgroup → every global access instruction touches
+// 32 consecutive columns (256 B runs at any stride).  LDS: 32 rows of
+// 513 float2 (row stride 1026 dwords ≡ 2 mod 64 → cross-column accesses
+// are bank-conflict-free).  Math validated against numpy (oracle sim).
+// ---------------------------------------------------------------------------
+
+template <int SIGN>
+__global__ void __launch_bounds__(256)
+    k_fft_mid512(const float2* __restrict__ in, float2* __restrict__ out,
+                 FftPassDescDev d, unsigned long long n_ffts,
+                 const float2* __restrict__ tw_n) {
+  extern __shared__ float2 lds[];
+  constexpr int F = 32;       // columns per workgroup
+  constexpr int LDST = 513;   // row stride (float2)
+  float2* ltw = lds;          // 512-entry twiddle table
+  float2* X = lds + 512;
+  const unsigned long long col0 = (unsigned long long)blockIdx.x * F;
+
+  for (int j = threadIdx.x; j < 512; j += blockDim.x) ltw[j] = tw_n[j];
+
+  // ---- load: lane f fastest → 32 consecutive columns per instruction ----
+  unsigned long long base_f;  // this lane's column base (f = tid & 31)
+  {
+    const int f = threadIdx.x & 31;
+    unsigned long long q0, q1, q2;
+    digits(col0 + f, d, q0, q1, q2);
+    base_f = q0 * d.in_c0 + q1 * d.in_c1 + q2 * d.in_c2;
+  }
+  __syncthreads();
+  const uint32_t stride32 = (uint32_t)d.in_stride;
+  {
+    const int f = threadIdx.x & 31;
+    const bool active = (col0 + f) < n_ffts;
+    for (int r = 0; r < 64; ++r) {
+      const int i = r * 8 + (threadIdx.x >> 5);  // i in [0,512)
+      const int h = i >> 8, j = i & 255;
+      X[f * LDST + 256 * h + dif_swz(j)] =
+          active ? in[base_f + (uint32_t)i * stride32]
+                 : make_float2(0.f, 0.f);
+    }
+  }
+  __syncthreads();
+
+  // ---- stage A: radix-2 split, odd half × W512^i ----
+  for (int b = threadIdx.x; b < F * 256; b += blockDim.x) {
+    const int f = b & 31;
+    const int i = b >> 5;
+    float2* row = X + f * LDST;
+    const int e = dif_swz(i);
+    const float2 u = row[e];
+    const float2 v = row[256 + e];
+    row[e] = make_float2(u.x + v.x, u.y + v.y);
+    const float2 dmy = make_float2(u.x - v.x, u.y - v.y);
+    row[256 + e] = (i == 0) ? dmy : cmulf(dmy, ltw[i]);
+  }
+  __syncthreads();
+
+  // ---- stage B: two independent in-place radix-4 DIF-256 per column ----
+  // W_L^x within a 256-point half = tw512[x * 512/L] = ltw[x << (9-log2 L)]
+  for (int L = 256, tl = 1; L >= 4; L >>= 2, tl += 2) {
+    const int M = L >> 2;
+    const int m_log2 = __builtin_ctz((unsigned)M);
+    for (int b = threadIdx.x; b < F * 128; b += blockDim.x) {
+      const int f = b & 31;
+      const int rest = b >> 5;        // [0,128): quartet index + half
+      const int h = rest >> 6;
+      const int bb = rest & 63;
+      const int j = bb & (M - 1);
+      const int g = (bb >> m_log2) << (m_log2 + 2);
+      float2* row = X + f * LDST + 256 * h;
+      const int e0 = dif_swz(g + j);
+      const int e1 = dif_swz(g + j + M);
+      const int e2 = dif_swz(g + j + 2 * M);
+      const int e3 = dif_swz(g + j + 3 * M);
+      const float2 a = row[e0];
+      const float2 bv = row[e1];
+      const float2 c = row[e2];
+      const float2 dv = row[e3];
+      const float2 t0 = make_float2(a.x + c.x, a.y + c.y);
+      const float2 t1 = make_float2(a.x - c.x, a.y - c.y);
+      const float2 t2 = make_float2(bv.x + dv.x, bv.y + dv.y);
+      const float2 dmy = make_float2(bv.x - dv.x, bv.y - dv.y);
+      const float2 t3 = (SIGN > 0) ? make_float2(-dmy.y, dmy.x)
+                                   : make_float2(dmy.y, -dmy.x);
+      const float2 x0 = make_float2(t0.x + t2.x, t0.y + t2.y);
+      const float2 x1 = make_float2(t1.x + t3.x, t1.y + t3.y);
+      const float2 x2 = make_float2(t0.x - t2.x, t0.y - t2.y);
+      const float2 x3 = make_float2(t1.x - t3.x, t1.y - t3.y);
+      row[e0] = x0;
+      row[e1] = (j == 0) ? x1 : cmulf(x1, ltw[j << tl]);
+      row[e2] = (j == 0) ? x2 : cmulf(x2, ltw[(2 * j) << tl]);
+      row[e3] = (j == 0) ? x3 : cmulf(x3, ltw[(3 * j) << tl]);
+    }
+    __syncthreads();
+  }
+
+  // ---- store: natural k, inter-pass twiddle W^(q0*f0*k) ----
+  {
+    const int f = threadIdx.x & 31;
+    unsigned long long q0, q1, q2;
+    digits(col0 + f, d, q0, q1, q2);
+    const bool active = (col0 + f) < n_ffts;
+    for (int r = 0; r < 64; ++r) {
+      const int k = r * 8 + (threadIdx.x >> 5);
+      const unsigned rk = rev4_bits((unsigned)(k >> 1), 8);
+      const float2 v = X[f * LDST + 256 * (k & 1) + dif_swz((int)rk)];
+      float2 rr = v;
+      const unsigned long long m_ =
+          (q0 * d.tw_f0 * (unsigned long long)k) & d.tw_mask;
+      rr = cmulf(rr, tw_eval(m_, d.tw_angle));
+      if (active) out[base_f + (uint32_t)k * stride32] = rr;
+    }
+  }
+}
+
 template <bool MEANP>
 __global__ void k_r2c_post(const float2* __restrict__ z,
                            float2* __restrict__ x, size_t m,
@@ -1156,6 +1281,36 @@ hipError_t fft_col_pass(const float2* in, float2* out, const FftPassDesc& hd,
 #undef COL_LAUNCH_PD
 #undef COL_LAUNCH_P32
 #undef COL_LAUNCH_P32D
+  SRTB_CHECK_LAUNCH();
+  return hipSuccess;
+}
+
+hipError_t fft_mid512_pass(const float2* in, float2* out,
+                           const FftPassDesc& hd, size_t n_ffts, int sign,
+                           const float2* tw_n, hipStream_t stream) {
+  if (hd.n != 512 || hd.tw_mod == 0) return hipErrorInvalidValue;
+  FftPassDescDev d;
+  d.n = hd.n;
+  d.n_log2 = 9;
+  d.f_log2 = 5;
+  d.d0_log2 = hd.d0 ? ilog2(hd.d0) : 48;
+  d.d1_log2 = ilog2(hd.d1 ? hd.d1 : 1);
+  d.in_c0 = hd.in_c0; d.in_c1 = hd.in_c1; d.in_c2 = hd.in_c2;
+  d.in_stride = hd.in_stride;
+  d.out_c0 = hd.out_c0; d.out_c1 = hd.out_c1; d.out_c2 = hd.out_c2;
+  d.out_stride = hd.out_stride;
+  d.tw_f0 = hd.tw_f0; d.tw_f1 = hd.tw_f1;
+  d.tw_mask = hd.tw_mod ? hd.tw_mod - 1 : 0;
+  d.tw_lo_bits = hd.tw_lo_bits;
+  d.tw_angle = hd.tw_angle;
+  const uint32_t grid = (uint32_t)((n_ffts + 31) / 32);
+  const size_t lds_bytes = (512 + 32 * 513) * sizeof(float2);
+  if (sign < 0)
+    hipLaunchKernelGGL((k_fft_mid512<-1>), dim3(grid), dim3(256), lds_bytes,
+                       stream, in, out, d, n_ffts, tw_n);
+  else
+    hipLaunchKernelGGL((k_fft_mid512<1>), dim3(grid), dim3(256), lds_bytes,
+                       stream, in, out, d, n_ffts, tw_n);
   SRTB_CHECK_LAUNCH();
   return hipSuccess;
 }
