@@ -852,13 +852,13 @@ __global__ void k_build_twiddle(float2* __restrict__ t, size_t count,
 // are bank-conflict-free).  Math validated against numpy (oracle sim).
 // ---------------------------------------------------------------------------
 
-template <int SIGN>
+template <int SIGN, int F = 32>
 __global__ void __launch_bounds__(256)
     k_fft_mid512(const float2* __restrict__ in, float2* __restrict__ out,
                  FftPassDescDev d, unsigned long long n_ffts,
                  const float2* __restrict__ tw_n) {
   extern __shared__ float2 lds[];
-  constexpr int F = 32;       // columns per workgroup
+  constexpr int LOGF = (F == 16) ? 4 : 5;  // columns per workgroup
   constexpr int LDST = 513;   // row stride (float2)
   float2* ltw = lds;          // 512-entry twiddle table
   float2* X = lds + 512;
@@ -866,10 +866,10 @@ __global__ void __launch_bounds__(256)
 
   for (int j = threadIdx.x; j < 512; j += blockDim.x) ltw[j] = tw_n[j];
 
-  // ---- load: lane f fastest → 32 consecutive columns per instruction ----
-  unsigned long long base_f;  // this lane's column base (f = tid & 31)
+  // ---- load: lane f fastest → F consecutive columns per instruction ----
+  unsigned long long base_f;  // this lane's column base (f = tid & (F-1))
   {
-    const int f = threadIdx.x & 31;
+    const int f = threadIdx.x & (F - 1);
     unsigned long long q0, q1, q2;
     digits(col0 + f, d, q0, q1, q2);
     base_f = q0 * d.in_c0 + q1 * d.in_c1 + q2 * d.in_c2;
@@ -877,10 +877,11 @@ __global__ void __launch_bounds__(256)
   __syncthreads();
   const uint32_t stride32 = (uint32_t)d.in_stride;
   {
-    const int f = threadIdx.x & 31;
+    const int f = threadIdx.x & (F - 1);
     const bool active = (col0 + f) < n_ffts;
-    for (int r = 0; r < 64; ++r) {
-      const int i = r * 8 + (threadIdx.x >> 5);  // i in [0,512)
+    constexpr int TPC = 256 / F;  // i-slices per thread pass
+    for (int r = 0; r < 512 / TPC; ++r) {
+      const int i = r * TPC + (threadIdx.x >> LOGF);  // i in [0,512)
       const int h = i >> 8, j = i & 255;
       X[f * LDST + 256 * h + dif_swz(j)] =
           active ? in[base_f + (uint32_t)i * stride32]
@@ -891,8 +892,8 @@ __global__ void __launch_bounds__(256)
 
   // ---- stage A: radix-2 split, odd half × W512^i ----
   for (int b = threadIdx.x; b < F * 256; b += blockDim.x) {
-    const int f = b & 31;
-    const int i = b >> 5;
+    const int f = b & (F - 1);
+    const int i = b >> LOGF;
     float2* row = X + f * LDST;
     const int e = dif_swz(i);
     const float2 u = row[e];
@@ -909,8 +910,8 @@ __global__ void __launch_bounds__(256)
     const int M = L >> 2;
     const int m_log2 = __builtin_ctz((unsigned)M);
     for (int b = threadIdx.x; b < F * 128; b += blockDim.x) {
-      const int f = b & 31;
-      const int rest = b >> 5;        // [0,128): quartet index + half
+      const int f = b & (F - 1);
+      const int rest = b >> LOGF;     // [0,128): quartet index + half
       const int h = rest >> 6;
       const int bb = rest & 63;
       const int j = bb & (M - 1);
@@ -944,12 +945,13 @@ __global__ void __launch_bounds__(256)
 
   // ---- store: natural k, inter-pass twiddle W^(q0*f0*k) ----
   {
-    const int f = threadIdx.x & 31;
+    const int f = threadIdx.x & (F - 1);
     unsigned long long q0, q1, q2;
     digits(col0 + f, d, q0, q1, q2);
     const bool active = (col0 + f) < n_ffts;
-    for (int r = 0; r < 64; ++r) {
-      const int k = r * 8 + (threadIdx.x >> 5);
+    constexpr int TPC = 256 / F;
+    for (int r = 0; r < 512 / TPC; ++r) {
+      const int k = r * TPC + (threadIdx.x >> LOGF);
       const unsigned rk = rev4_bits((unsigned)(k >> 1), 8);
       const float2 v = X[f * LDST + 256 * (k & 1) + dif_swz((int)rk)];
       float2 rr = v;
@@ -1303,14 +1305,28 @@ hipError_t fft_mid512_pass(const float2* in, float2* out,
   d.tw_mask = hd.tw_mod ? hd.tw_mod - 1 : 0;
   d.tw_lo_bits = hd.tw_lo_bits;
   d.tw_angle = hd.tw_angle;
-  const uint32_t grid = (uint32_t)((n_ffts + 31) / 32);
-  const size_t lds_bytes = (512 + 32 * 513) * sizeof(float2);
-  if (sign < 0)
-    hipLaunchKernelGGL((k_fft_mid512<-1>), dim3(grid), dim3(256), lds_bytes,
-                       stream, in, out, d, n_ffts, tw_n);
-  else
-    hipLaunchKernelGGL((k_fft_mid512<1>), dim3(grid), dim3(256), lds_bytes,
-                       stream, in, out, d, n_ffts, tw_n);
+  // SRTB_FFT_MIDF=16 halves LDS per workgroup (2 WG/CU) at 128 B runs
+  static const int Fw = [] {
+    const char* e = std::getenv("SRTB_FFT_MIDF");
+    return (e && std::atoi(e) == 16) ? 16 : 32;
+  }();
+  const uint32_t grid = (uint32_t)((n_ffts + Fw - 1) / Fw);
+  const size_t lds_bytes = (512 + (size_t)Fw * 513) * sizeof(float2);
+  if (Fw == 16) {
+    if (sign < 0)
+      hipLaunchKernelGGL((k_fft_mid512<-1, 16>), dim3(grid), dim3(256),
+                         lds_bytes, stream, in, out, d, n_ffts, tw_n);
+    else
+      hipLaunchKernelGGL((k_fft_mid512<1, 16>), dim3(grid), dim3(256),
+                         lds_bytes, stream, in, out, d, n_ffts, tw_n);
+  } else {
+    if (sign < 0)
+      hipLaunchKernelGGL((k_fft_mid512<-1, 32>), dim3(grid), dim3(256),
+                         lds_bytes, stream, in, out, d, n_ffts, tw_n);
+    else
+      hipLaunchKernelGGL((k_fft_mid512<1, 32>), dim3(grid), dim3(256),
+                         lds_bytes, stream, in, out, d, n_ffts, tw_n);
+  }
   SRTB_CHECK_LAUNCH();
   return hipSuccess;
 }
