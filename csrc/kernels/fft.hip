@@ -858,7 +858,7 @@ __global__ void __launch_bounds__(256)
                  FftPassDescDev d, unsigned long long n_ffts,
                  const float2* __restrict__ tw_n) {
   extern __shared__ float2 lds[];
-  constexpr int LOGF = (F == 16) ? 4 : 5;  // columns per workgroup
+  constexpr int LOGF = col_ilog2(F);  // columns per workgroup
   constexpr int LDST = 513;   // row stride (float2)
   float2* ltw = lds;          // 512-entry twiddle table
   float2* X = lds + 512;
@@ -1305,14 +1305,23 @@ hipError_t fft_mid512_pass(const float2* in, float2* out,
   d.tw_mask = hd.tw_mod ? hd.tw_mod - 1 : 0;
   d.tw_lo_bits = hd.tw_lo_bits;
   d.tw_angle = hd.tw_angle;
-  // SRTB_FFT_MIDF=16 halves LDS per workgroup (2 WG/CU) at 128 B runs
+  // SRTB_FFT_MIDF ∈ {8,16,32}: fewer columns per workgroup = less LDS =
+  // more workgroups/CU (measured: occupancy, not run length, binds)
   static const int Fw = [] {
     const char* e = std::getenv("SRTB_FFT_MIDF");
-    return (e && std::atoi(e) == 16) ? 16 : 32;
+    const int v = e ? std::atoi(e) : 32;
+    return (v == 8 || v == 16) ? v : 32;
   }();
   const uint32_t grid = (uint32_t)((n_ffts + Fw - 1) / Fw);
   const size_t lds_bytes = (512 + (size_t)Fw * 513) * sizeof(float2);
-  if (Fw == 16) {
+  if (Fw == 8) {
+    if (sign < 0)
+      hipLaunchKernelGGL((k_fft_mid512<-1, 8>), dim3(grid), dim3(256),
+                         lds_bytes, stream, in, out, d, n_ffts, tw_n);
+    else
+      hipLaunchKernelGGL((k_fft_mid512<1, 8>), dim3(grid), dim3(256),
+                         lds_bytes, stream, in, out, d, n_ffts, tw_n);
+  } else if (Fw == 16) {
     if (sign < 0)
       hipLaunchKernelGGL((k_fft_mid512<-1, 16>), dim3(grid), dim3(256),
                          lds_bytes, stream, in, out, d, n_ffts, tw_n);

@@ -95,3 +95,16 @@ def test_fft_forced_column_widths(C, maxcol, monkeypatch):
     out = C.native_fft(xt, -1).cpu().numpy()
     ref_t = torch.fft.fft(torch.from_numpy(x).cuda(), dim=1).cpu().numpy()
     assert rel_err(out, ref_t) < 1e-4
+
+
+def test_fft_mid512_pass(C, monkeypatch):
+    """The experimental 512-point strided LDS middle pass (reachable only
+    through the factorization override) must match torch.fft; this exact
+    plan/config was GPU-measured during round 1."""
+    monkeypatch.setenv("SRTB_FFT_FACTORS", "2,512,256")
+    n = 1 << 18
+    x = rand_c64((4, n), seed=99)
+    xt = torch.from_numpy(x).cuda()
+    out = C.native_fft(xt, -1).cpu().numpy()
+    ref = torch.fft.fft(torch.from_numpy(x).cuda(), dim=1).cpu().numpy()
+    assert rel_err(out, ref) < 1e-4
