@@ -50,8 +50,9 @@ def test_fft_two_pass(C, n, sign):
     assert rel_err(out, ref) < 1e-4
 
 
-# three-pass lengths (forward 2^25..2^29 class; test smaller for speed)
-@pytest.mark.parametrize("n", [1 << 25, 1 << 27])
+# multi-pass lengths (forward 2^25..2^29 class); 2^25 = balanced planner,
+# 2^26/2^27/2^28 = the greedy-64 planner with remainder 0/2/4
+@pytest.mark.parametrize("n", [1 << 25, 1 << 26, 1 << 27, 1 << 28])
 def test_fft_three_pass(C, n):
     x = rand_c64(n, seed=5)
     xt = torch.from_numpy(x).cuda()
