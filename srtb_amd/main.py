@@ -180,7 +180,8 @@ def main(argv=None) -> int:
     agg = DetectionAggregator()
     writer = SignalWriteScheduler(
         cfg.baseband_output_file_prefix, cfg.baseband_input_count,
-        cfg.baseband_sample_rate, real_time=(cfg.input_file_path == ""))
+        cfg.baseband_sample_rate, real_time=(cfg.input_file_path == ""),
+        async_writes=True)
 
     t0 = time.time()
     n_blocks = 0
@@ -273,6 +274,7 @@ def main(argv=None) -> int:
     if write_all_f is not None:
         write_all_f.close()
     elapsed = time.time() - t0
+    writer.close()  # flush the async write pool before the summary
     stats = agg.reduce()
     if rank == 0:
         sps = stats.blocks * cfg.baseband_input_count / max(elapsed, 1e-9)
