@@ -367,30 +367,24 @@ hipError_t unpack(const uint8_t* in, float* out, size_t out_count, int nbits,
       }
       break;
     }
-    case 16:
-      hipLaunchKernelGGL((k_unpack_cast<uint16_t, false>),
-                         grid_for(out_count), dim3(kBlock), 0, stream,
-                         reinterpret_cast<const uint16_t*>(in), out, out_count,
-                         window);
+#define SRTB_CAST_CASE(BITS, T)                                              \
+    case BITS:                                                               \
+      if (w)                                                                 \
+        hipLaunchKernelGGL((k_unpack_cast<T, true>), grid_for(out_count),    \
+                           dim3(kBlock), 0, stream,                          \
+                           reinterpret_cast<const T*>(in), out, out_count,   \
+                           window);                                          \
+      else                                                                   \
+        hipLaunchKernelGGL((k_unpack_cast<T, false>), grid_for(out_count),   \
+                           dim3(kBlock), 0, stream,                          \
+                           reinterpret_cast<const T*>(in), out, out_count,   \
+                           window);                                          \
       break;
-    case -16:
-      hipLaunchKernelGGL((k_unpack_cast<int16_t, false>), grid_for(out_count),
-                         dim3(kBlock), 0, stream,
-                         reinterpret_cast<const int16_t*>(in), out, out_count,
-                         window);
-      break;
-    case 32:
-      hipLaunchKernelGGL((k_unpack_cast<uint32_t, false>),
-                         grid_for(out_count), dim3(kBlock), 0, stream,
-                         reinterpret_cast<const uint32_t*>(in), out, out_count,
-                         window);
-      break;
-    case -32:
-      hipLaunchKernelGGL((k_unpack_cast<int32_t, false>), grid_for(out_count),
-                         dim3(kBlock), 0, stream,
-                         reinterpret_cast<const int32_t*>(in), out, out_count,
-                         window);
-      break;
+    SRTB_CAST_CASE(16, uint16_t)
+    SRTB_CAST_CASE(-16, int16_t)
+    SRTB_CAST_CASE(32, uint32_t)
+    SRTB_CAST_CASE(-32, int32_t)
+#undef SRTB_CAST_CASE
     default:
       return hipErrorInvalidValue;
   }

@@ -58,13 +58,17 @@ def test_unpack_16bit(C):
         np.testing.assert_array_equal(out, expect)
 
 
-def test_unpack_with_window(C):
+@pytest.mark.parametrize("nbits", [2, 8, -16, 32])
+def test_unpack_with_window(C, nbits):
+    """Window fusion must apply to EVERY sample format (the 16/32-bit cast
+    cases once hardcoded it off)."""
     rng = np.random.default_rng(3)
     raw = rng.integers(0, 256, 4096, dtype=np.uint8)
-    w = ref.window_coefficients("hamming", 4096 * 4)
-    expect = ref.unpack(raw, 2, window=w)
-    out = C.unpack(to_gpu(raw), 2, expect.size, to_gpu(w)).cpu().numpy()
-    np.testing.assert_allclose(out, expect, rtol=1e-6)
+    expect_plain = ref.unpack(raw, nbits)
+    w = ref.window_coefficients("hamming", expect_plain.size)
+    expect = ref.unpack(raw, nbits, window=w)
+    out = C.unpack(to_gpu(raw), nbits, expect.size, to_gpu(w)).cpu().numpy()
+    np.testing.assert_allclose(out, expect, rtol=1e-5, atol=1e-3)
 
 
 def test_unpack_2pol_kinds(C):
