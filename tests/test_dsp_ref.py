@@ -300,3 +300,64 @@ def test_sk_v1_time_major_layout():
     outn = ref.rfi_mitigate_sk_v1(wf, 1.05, normalize=True)
     surv = np.abs(outn[:, 8].astype(np.complex128)) ** 2
     np.testing.assert_allclose(surv.mean(), 1.0, rtol=0.05)
+
+
+def test_unpack_hand_computed_goldens():
+    """Hand-computed bit tables, independent of the vectorized oracle
+    implementation (mirrors reference tests/test-unpack.cpp:59-256 which
+    uses 0b01100011-style literals)."""
+    b = np.array([0b01100011], dtype=np.uint8)
+    # 1-bit MSB-first: 0,1,1,0,0,0,1,1
+    np.testing.assert_array_equal(ref.unpack(b, 1),
+                                  [0, 1, 1, 0, 0, 0, 1, 1])
+    # 2-bit MSB-first: 01 10 00 11 -> 1,2,0,3
+    np.testing.assert_array_equal(ref.unpack(b, 2), [1, 2, 0, 3])
+    # 4-bit: 0110 0011 -> 6,3
+    np.testing.assert_array_equal(ref.unpack(b, 4), [6, 3])
+    # 8-bit unsigned / signed
+    np.testing.assert_array_equal(ref.unpack(b, 8), [0x63])
+    nb = np.array([0x9C], dtype=np.uint8)  # -100 as int8
+    np.testing.assert_array_equal(ref.unpack(nb, -8), [-100])
+    # 16-bit little-endian: 0x0163
+    b2 = np.array([0x63, 0x01], dtype=np.uint8)
+    np.testing.assert_array_equal(ref.unpack(b2, 16), [0x0163])
+
+
+def test_window_hamming_numpy_golden():
+    """Hamming coefficients vs numpy.hamming(16) (reference
+    tests/test-fft_window.cpp:23-60 embeds the same literals).  The
+    reference uses a0 = 25/46 exactly, numpy uses 0.54 — compare against
+    the exact-a0 formula evaluated independently here."""
+    n = 16
+    w = ref.window_coefficients("hamming", n)
+    a0 = 25.0 / 46.0
+    k = np.arange(n)
+    expect = a0 - (1 - a0) * np.cos(2 * np.pi * k / (n - 1))
+    np.testing.assert_allclose(w, expect, rtol=1e-6)
+    # and hann
+    w = ref.window_coefficients("hann", n)
+    expect = 0.5 - 0.5 * np.cos(2 * np.pi * k / (n - 1))
+    np.testing.assert_allclose(w, expect, rtol=1e-6)
+
+
+def test_dedispersion_phase_extended_precision():
+    """The delta-phase reaches ~1e9 cycles at the J1644 configuration; the
+    fp64 integer-part cancellation must still leave the FRACTIONAL phase
+    accurate.  Compare against float128 (80-bit x87) evaluation (mirrors
+    the reference's df64-vs-double + mpmath checks, tests/test-df64.*)."""
+    n = 1 << 12
+    f_min, bw, dm = 1437.0, -64.0, -478.80
+    f_c, df = f_min + bw, bw / n
+    fac64 = ref.dedisp_phase_factors(n, f_min, f_c, df, dm)
+
+    D = np.float128(4.148808e3) * np.float128(1e6)
+    i = np.arange(n, dtype=np.float128)
+    f = np.float128(f_min) + np.float128(df) * i
+    k = (D * np.float128(dm) / f *
+         ((f - np.float128(f_c)) / np.float128(f_c)) ** 2)
+    frac = k - np.floor(k)
+    expect = np.exp(-2j * np.pi * frac.astype(np.float64))
+    # |k| peaks near 1.3e6 cycles here; fp64 keeps ~1e-9 of the fraction,
+    # but the ref path uses float32 sincos downstream — require 1e-5
+    err = np.abs(fac64.astype(np.complex128) - expect).max()
+    assert err < 1e-5, err
