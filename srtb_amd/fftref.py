@@ -252,3 +252,42 @@ def rfft_packed(x: np.ndarray, factors: list[int] | None = None) -> np.ndarray:
     else:
         zf = fft_deep(z.astype(np.complex64), factors, -1).astype(np.complex128)
     return r2c_post(zf, -1)
+
+
+def plan_factors(t: int, maxcol_log2: int = 5, final_log2: int = 8) -> list[int]:
+    """Mirror of the native planner's factorization policy
+    (csrc/fft/native_fft.h plan()): the column factors + final DIF length
+    chosen for a 2**t transform.  Kept as the executable SPEC of the
+    policy — the C++ is the implementation of record.
+
+    t <= 12 -> [2**t] (single LDS Stockham pass).
+    Else: final DIF length 256 (pure 4^k), and the residual bits go to
+    register-column factors: greedy 64s for rest >= 18 (the N=64 lane-pair
+    kernel is the fastest pass; measured fft_factor_sweep.py), balanced
+    <=32 columns otherwise (measured best at the 2^18 waterfall).
+    """
+    if t <= 12:
+        return [1 << t]
+    if final_log2 % 2:
+        final_log2 += 1
+    final_log2 = min(max(final_log2, 8), 12)
+    if final_log2 >= t:
+        final_log2 = t - 1 if t % 2 else t - 2
+    rest = t - final_log2
+    f: list[int] = []
+    if 18 <= rest <= 24:
+        left = rest
+        while left >= 6 and len(f) < 4:
+            f.append(64)
+            left -= 6
+        if left:
+            f.append(1 << left)
+    else:
+        ncols = min((rest + maxcol_log2 - 1) // maxcol_log2, 4)
+        base, extra = divmod(rest, ncols)
+        for i in range(ncols):
+            b = base + (1 if i < extra else 0)
+            assert b <= 6, "factor too large"
+            f.append(1 << b)
+    f.append(1 << final_log2)
+    return f
