@@ -6,7 +6,7 @@ provider yields raw packets; the block worker places payloads at
 (counter - begin) * payload_size inside a block buffer, zero-filling lost
 packets and counting the loss rate.  The production-rate native path
 (recvmmsg batching, pinned hugepage buffers, core pinning) lives in
-csrc/io/udp_receiver.cpp and the srtb-baseband-receiver tool; this module is
+csrc/app/udp_receiver.h and the srtb-backend / srtb-baseband-receiver tools; this module is
 the protocol logic, unit-testable without sockets, plus a socket provider for
 integration tests and moderate-rate use.
 """
