@@ -207,8 +207,14 @@ int main(int argc, char** argv) {
           const int got = prov.receive();
           for (int i = 0; i < got; ++i) {
             if (assembler.push(prov.packet(i), prov.packet_len(i))) {
-              q_in.push(BlockMsg{bi, assembler.first_timestamp(), false}, stopped);
+              q_in.push(BlockMsg{bi, assembler.block_begin_counter(), false},
+                        stopped);
               ++count;
+              const auto& st = assembler.stats();
+              SRTB_APP_LOGI("udp block " << assembler.block_begin_counter()
+                            << ": received=" << st.received
+                            << " lost=" << st.lost << " loss_rate="
+                            << st.loss_rate());
               if (max_blocks >= 0 && count >= max_blocks) {
                 q_in.push(BlockMsg{-1, 0, true}, stopped);
                 return;

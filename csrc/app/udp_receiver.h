@@ -38,6 +38,10 @@ struct PacketFormat {
 
   size_t packet_size() const { return header_size + payload_size; }
 
+  // 'simple' (reference backend_registry.hpp:36-39) is a headerless linear
+  // sample stream with no counter: packets append sequentially, any size
+  bool headerless = false;
+
   static PacketFormat for_backend(const std::string& name) {
     PacketFormat f;
     if (name == "fastmb_roach2" || name == "naocpsr_roach2" ||
@@ -50,9 +54,10 @@ struct PacketFormat {
       f.payload_size = 8192;
       f.counter_offset = 24;  // VDIF words 6..7
     } else if (name == "simple") {
-      f.header_size = 8;
-      f.payload_size = 4096;
+      f.header_size = 0;
+      f.payload_size = 4096;  // nominal; any size accepted (headerless)
       f.counter_offset = 0;
+      f.headerless = true;
     } else {
       throw std::runtime_error("unknown backend format: " + name);
     }
@@ -118,6 +123,7 @@ class RecvmmsgProvider {
 
 struct LossStats {
   uint64_t received = 0, lost = 0, out_of_order = 0, wrong_size = 0;
+  uint64_t duplicate = 0;
   double loss_rate() const {
     const uint64_t t = received + lost;
     return t ? (double)lost / (double)t : 0.0;
@@ -130,15 +136,18 @@ class BlockAssembler {
  public:
   BlockAssembler(PacketFormat fmt, size_t block_bytes, uint8_t* block_buf)
       : fmt_(fmt), block_bytes_(block_bytes), buf_(block_buf) {
-    if (block_bytes % fmt.payload_size != 0)
+    if (!fmt.headerless && block_bytes % fmt.payload_size != 0)
       throw std::runtime_error("block not a multiple of payload");
-    packets_per_block_ = block_bytes / fmt.payload_size;
+    packets_per_block_ =
+        fmt.headerless ? 0 : block_bytes / fmt.payload_size;
+    occupied_.assign(packets_per_block_, 0);
     std::memset(buf_, 0, block_bytes_);
   }
 
   // feed one packet; returns true when the current block completed (caller
   // consumes buf_ and MUST call begin_next() before pushing more)
   bool push(const uint8_t* pkt, size_t len) {
+    if (fmt_.headerless) return push_headerless(pkt, len);
     if (len != fmt_.packet_size()) {
       ++stats_.wrong_size;
       return false;
@@ -160,9 +169,18 @@ class BlockAssembler {
       stats_.lost += packets_per_block_ - filled_;
       const uint64_t skip = (uint64_t)idx / packets_per_block_;
       stats_.lost += (skip - 1) * packets_per_block_;
+      // the counter that identifies the block being handed to the caller
+      // (reference udp_receiver.hpp block_first_counter — NOT the stream's
+      // first counter ever: product files are named with this value)
+      completed_begin_ = begin_;
       next_begin_ = begin_ + skip * packets_per_block_;
       return true;
     }
+    if (occupied_[(size_t)idx]) {
+      ++stats_.duplicate;
+      return false;
+    }
+    occupied_[(size_t)idx] = 1;
     std::memcpy(buf_ + (size_t)idx * fmt_.payload_size,
                 pkt + fmt_.header_size, fmt_.payload_size);
     ++filled_;
@@ -173,6 +191,7 @@ class BlockAssembler {
   void begin_next() {
     begin_ = next_begin_;
     filled_ = 0;
+    std::fill(occupied_.begin(), occupied_.end(), 0);
     std::memset(buf_, 0, block_bytes_);
     if (!pending_.empty()) {
       std::vector<uint8_t> p;
@@ -187,15 +206,40 @@ class BlockAssembler {
   }
 
   uint64_t first_timestamp() const { return first_timestamp_; }
+  // begin counter of the most recently COMPLETED block (valid after push()
+  // returned true, until the next completion) — use this to stamp/name the
+  // block's products, matching the reference's per-block first counter
+  uint64_t block_begin_counter() const { return completed_begin_; }
   const LossStats& stats() const { return stats_; }
 
  private:
+  // 'simple' headerless mode: sequential append of raw sample bytes; block
+  // completes when full, remainder carries into the next block.  The block
+  // counter is the byte offset of the block start in the stream.
+  bool push_headerless(const uint8_t* pkt, size_t len) {
+    const size_t take = std::min(len, block_bytes_ - fill_bytes_);
+    std::memcpy(buf_ + fill_bytes_, pkt, take);
+    fill_bytes_ += take;
+    stats_.received += 1;
+    if (fill_bytes_ >= block_bytes_) {
+      completed_begin_ = stream_offset_;
+      stream_offset_ += block_bytes_;
+      if (take < len) pending_.assign(pkt + take, pkt + len);
+      fill_bytes_ = 0;
+      return true;
+    }
+    return false;
+  }
+
   PacketFormat fmt_;
   size_t block_bytes_, packets_per_block_;
   uint8_t* buf_;
   uint64_t begin_ = 0, next_begin_ = 0, filled_ = 0, first_timestamp_ = 0;
+  uint64_t completed_begin_ = 0;
+  uint64_t fill_bytes_ = 0, stream_offset_ = 0;  // headerless mode
   bool started_ = false;
   std::vector<uint8_t> pending_;
+  std::vector<uint8_t> occupied_;
   LossStats stats_;
 };
 

@@ -42,6 +42,7 @@ class LossStats:
     lost: int = 0
     out_of_order: int = 0
     wrong_size: int = 0
+    duplicate: int = 0
 
     @property
     def loss_rate(self) -> float:
@@ -59,21 +60,36 @@ class BlockAssembler:
     """
 
     def __init__(self, backend: type[Backend], block_bytes: int):
-        assert backend.packet_payload_size > backend.packet_header_size > 0, \
-            "BlockAssembler needs a counter-stamped backend format"
+        # 'simple' (reference backend_registry.hpp:36-39) is a headerless
+        # linear sample stream: sequential append, no counter
+        self.headerless = backend.packet_payload_size == 0
         self.backend = backend
-        self.payload = backend.packet_payload_size - backend.packet_header_size
-        if block_bytes % self.payload != 0:
-            raise ValueError(
-                f"block_bytes {block_bytes} not a multiple of payload "
-                f"{self.payload}")
+        if self.headerless:
+            self.payload = 0
+            self.packets_per_block = 0
+            self.fill_bytes = 0
+            self.stream_offset = 0
+            self._pending = b""
+        else:
+            assert backend.packet_payload_size > backend.packet_header_size > 0, \
+                "BlockAssembler needs a counter-stamped backend format"
+            self.payload = (backend.packet_payload_size -
+                            backend.packet_header_size)
+            if block_bytes % self.payload != 0:
+                raise ValueError(
+                    f"block_bytes {block_bytes} not a multiple of payload "
+                    f"{self.payload}")
+            self.packets_per_block = block_bytes // self.payload
         self.block_bytes = block_bytes
-        self.packets_per_block = block_bytes // self.payload
         self.begin_counter: int | None = None
         self.buf = np.zeros(block_bytes, dtype=np.uint8)
         self.filled = np.zeros(self.packets_per_block, dtype=bool)
         self.stats = LossStats()
         self.first_timestamp = 0
+        # counter/timestamp identifying the most recently COMPLETED block
+        # (reference block_first_counter — set when push() returns a block)
+        self.last_block_counter = 0
+        self.last_block_timestamp = 0
 
     def _reset(self, begin: int):
         self.begin_counter = begin
@@ -84,8 +100,12 @@ class BlockAssembler:
         """Feed one packet; returns a completed block or None.
 
         A block completes when a packet at/after the end arrives; missing
-        packets stay zero (counted as lost).
+        packets stay zero (counted as lost).  When a block is returned,
+        `last_block_counter`/`last_block_timestamp` identify THAT block
+        (its begin counter — reference block_first_counter semantics).
         """
+        if self.headerless:
+            return self._push_headerless(packet)
         if len(packet) != self.backend.packet_payload_size:
             self.stats.wrong_size += 1
             return None
@@ -98,7 +118,10 @@ class BlockAssembler:
             self.stats.out_of_order += 1
             return None
         if idx >= self.packets_per_block:
-            # complete current block (zero-fill the tail as lost)
+            # complete current block (zero-fill the tail as lost); stamp it
+            # with ITS begin counter, not the triggering packet's
+            self.last_block_counter = self.begin_counter
+            self.last_block_timestamp = self.first_timestamp
             out = self.finish()
             # advance begin by whole blocks so this packet lands in the new
             # block; fully-lost intermediate blocks are accounted as lost
@@ -110,11 +133,34 @@ class BlockAssembler:
             assert res is None
             return out
         off = idx * self.payload
+        if self.filled[idx]:
+            self.stats.duplicate += 1
+            return None
         self.buf[off:off + self.payload] = np.frombuffer(
             packet, dtype=np.uint8)[self.backend.packet_header_size:]
-        if not self.filled[idx]:
-            self.filled[idx] = True
-            self.stats.received += 1
+        self.filled[idx] = True
+        self.stats.received += 1
+        return None
+
+    def _push_headerless(self, packet: bytes) -> np.ndarray | None:
+        take = min(len(packet), self.block_bytes - self.fill_bytes)
+        self.buf[self.fill_bytes:self.fill_bytes + take] = np.frombuffer(
+            packet[:take], dtype=np.uint8)
+        self.fill_bytes += take
+        self.stats.received += 1
+        if self.fill_bytes >= self.block_bytes:
+            self.last_block_counter = self.stream_offset
+            self.last_block_timestamp = self.stream_offset
+            self.stream_offset += self.block_bytes
+            out = self.buf.copy()
+            self.buf[:] = 0
+            self.fill_bytes = 0
+            rem = packet[take:]
+            if rem:
+                r = self._push_headerless(rem)
+                assert r is None
+                self.stats.received -= 1  # same packet, counted once
+            return out
         return None
 
     def finish(self) -> np.ndarray:
@@ -126,11 +172,15 @@ class BlockAssembler:
 
 def run_receiver(provider, assembler: BlockAssembler, on_block,
                  stop_flag) -> None:
-    """Receive loop: provider → assembler → on_block(block_bytes)."""
+    """Receive loop: provider → assembler → on_block(block, block_counter).
+
+    The counter passed to on_block is the completed block's OWN begin
+    counter (reference block_first_counter), not the stream's first one.
+    """
     while not stop_flag():
         pkt = provider.receive()
         if not pkt:
             continue
         blk = assembler.push(pkt)
         if blk is not None:
-            on_block(blk, assembler.first_timestamp)
+            on_block(blk, assembler.last_block_timestamp)

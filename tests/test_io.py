@@ -113,6 +113,62 @@ def test_assembler_big_jump_counts_lost_blocks():
     assert (out2[4096:2 * 4096] == 13 % 256).all()
 
 
+def test_assembler_block_counter_labels_completed_block():
+    # each completed block is stamped with ITS OWN begin counter
+    # (reference block_first_counter), not the triggering packet's nor the
+    # stream's first counter ever
+    asm = BlockAssembler(bk.FastmbRoach2, 4096 * 4)
+    for c in range(4):
+        assert asm.push(mk_packet(c)) is None
+    out = asm.push(mk_packet(4))  # completes block [0..4)
+    assert out is not None
+    assert asm.last_block_counter == 0
+    for c in range(5, 8):
+        asm.push(mk_packet(c))
+    out = asm.push(mk_packet(8))  # completes block [4..8)
+    assert out is not None
+    assert asm.last_block_counter == 4
+    # loss-crossing completion: jump into a far block — completed block
+    # is still labeled with its own begin (8), next begin aligned to 16
+    out = asm.push(mk_packet(17))
+    assert out is not None
+    assert asm.last_block_counter == 8
+    out = asm.push(mk_packet(20))  # completes block [16..20)
+    assert out is not None
+    assert asm.last_block_counter == 16
+
+
+def test_assembler_duplicates_counted_separately():
+    asm = BlockAssembler(bk.FastmbRoach2, 4096 * 4)
+    asm.push(mk_packet(0, payload_byte=7))
+    asm.push(mk_packet(0, payload_byte=9))  # duplicate, dropped
+    assert asm.stats.received == 1
+    assert asm.stats.duplicate == 1
+    asm.push(mk_packet(1))
+    asm.push(mk_packet(2))
+    asm.push(mk_packet(3))
+    out = asm.push(mk_packet(4))
+    assert out is not None
+    assert (out[:4096] == 7).all()  # first copy wins
+    assert asm.stats.lost == 0  # duplicates must not corrupt loss stats
+
+
+def test_assembler_simple_headerless_append():
+    # 'simple' backend = headerless linear sample stream (reference
+    # backend_registry.hpp:36-39): bytes append sequentially, any size
+    asm = BlockAssembler(bk.Simple, 100)
+    data = bytes(range(250))
+    blocks = []
+    for off in range(0, 250, 60):  # arbitrary chunking
+        blk = asm.push(data[off:off + 60])
+        if blk is not None:
+            blocks.append((asm.last_block_counter, blk))
+    assert len(blocks) == 2
+    assert blocks[0][0] == 0 and blocks[1][0] == 100
+    assert bytes(blocks[0][1]) == data[:100]
+    assert bytes(blocks[1][1]) == data[100:200]
+
+
 # ---------------- file input ----------------
 
 def test_file_reader_overlap(tmp_path):
