@@ -1,8 +1,10 @@
 #!/usr/bin/env python3
 """Flagship benchmark: the J1644-4559 single-pulse pipeline on MI355X.
 
-One step = one full baseband block (default 2^30 samples, 2-bit, the
-J1644-4559 observation config of BASELINE.json) through the native engine:
+One step = --blocks-per-step full baseband blocks (default 2^30 samples
+each, 2-bit, the J1644-4559 observation config of BASELINE.json, with the
+DM-478.8 dedispersion overlap of ~23.5 M samples reserved and reprocessed
+per block as the reference does) through the native engine:
 
   H2D → unpack → R2C FFT (2^30) → RFI s1 + manual zap + coherent dedispersion
   (fused) → waterfall batched C2C (2048 ch) → spectral-kurtosis RFI →
@@ -40,6 +42,13 @@ def main():
                     help="baseband_input_count per block (J1644: 2^30)")
     ap.add_argument("--channels", type=int, default=2**11)
     ap.add_argument("--bits", type=int, default=2)
+    ap.add_argument("--blocks-per-step", type=int, default=4,
+                    help="baseband blocks processed per timed step (one "
+                    "block ~18 ms; >1 keeps the timed region long enough "
+                    "for external GPU-utilization sampling)")
+    ap.add_argument("--no-reserve", action="store_true",
+                    help="disable the dedispersion overlap reservation "
+                    "(nsamps_reserved=0; NOT the honest J1644 config)")
     ap.add_argument("--phase-table", action="store_true",
                     help="cache dedispersion phase factors (fixed DM); "
                     "default recomputes fp64 phase per block like the reference")
@@ -51,6 +60,10 @@ def main():
                     help="capture the per-block chain into hipGraphs")
     ap.add_argument("--slots", type=int, default=4,
                     help="double-buffered engine slots (streams)")
+    ap.add_argument("--backend", choices=["nccl", "gloo"], default="nccl",
+                    help="torch.distributed backend for world>1 (nccl=RCCL "
+                    "over xGMI, one rank per GPU; gloo validates the "
+                    "distributed path with several ranks sharing one GPU)")
     args = ap.parse_args()
 
     import torch
@@ -62,12 +75,15 @@ def main():
     distributed = world_size > 1
 
     assert torch.cuda.is_available(), "bench.py requires a GPU"
-    torch.cuda.set_device(local_rank)
+    torch.cuda.set_device(local_rank % torch.cuda.device_count())
 
     if distributed:
-        torch.distributed.init_process_group(backend="nccl")
+        torch.distributed.init_process_group(backend=args.backend)
+    # collective tensors live on the GPU for RCCL, on the host for gloo
+    coll_dev = "cuda" if args.backend == "nccl" else "cpu"
 
     from srtb_amd.ops import native
+    from srtb_amd import ref
     C = native()
 
     # J1644-4559 observation config (reference userspace/srtb_config_1644-4559.cfg)
@@ -87,11 +103,18 @@ def main():
     hi = round((1418.0 - freq_low) / bandwidth * (nc - 1))
     zap_ranges = [[int(lo), int(hi)]]
 
+    # dedispersion overlap: adjacent blocks share `reserved` samples which
+    # are reprocessed next block (reference coherent_dedispersion.hpp:87-128);
+    # at DM 478.8 on this band that is ~23.5 M of 2^30 (~2.2%).  Throughput
+    # below counts only the (n - reserved) NEW samples per block.
+    reserved = 0 if args.no_reserve else ref.nsamps_reserved(
+        n, s, freq_low, bandwidth, sample_rate, dm, True)
+
     eng = C.PipelineEngine(
         n=n, nbits=args.bits, channels=s, freq_low=freq_low,
         bandwidth=bandwidth, sample_rate=sample_rate, dm=dm,
         rfi_threshold=rfi_threshold, sk_threshold=sk_threshold,
-        snr_threshold=snr, max_boxcar=max_boxcar, nsamps_reserved=0,
+        snr_threshold=snr, max_boxcar=max_boxcar, nsamps_reserved=reserved,
         zap_ranges=zap_ranges, use_phase_table=args.phase_table,
         enable_rfi_s1=not args.no_rfi, enable_sk=True, n_slots=args.slots,
         fft_backend={"native": 0, "hipfft": 1, "auto": 2}[args.fft],
@@ -105,26 +128,48 @@ def main():
         t = torch.from_numpy(rng.integers(0, 256, raw_bytes, dtype=np.uint8))
         pinned.append(t.pin_memory())
 
-    counts_dev = torch.zeros(4, dtype=torch.int64, device="cuda")
+    # Detection-stat aggregation over RCCL: per-block counters go out as
+    # ASYNC all-reduces (async_op=True — RCCL runs them on its own stream,
+    # overlapped with the next blocks' compute); handles are only waited
+    # when their rotating buffer is reused and at drain.  Nothing blocking
+    # sits on the submit path.
+    n_cbuf = max(args.slots, 2)
+    counts_dev = [torch.zeros(4, dtype=torch.int64, device=coll_dev)
+                  for _ in range(n_cbuf)]
+    pending_handles = [None] * n_cbuf
+    agg_i = 0
 
     def agg(res):
+        nonlocal agg_i
+        b = agg_i % n_cbuf
+        agg_i += 1
+        if pending_handles[b] is not None:
+            pending_handles[b].wait()
+            pending_handles[b] = None
         c = [res["zero_count"]] + [cnt for _, cnt in res["counts"][:3]]
-        counts_dev.copy_(torch.tensor(c, dtype=torch.int64))
+        counts_dev[b].copy_(torch.tensor(c, dtype=torch.int64))
         if distributed:
-            torch.distributed.all_reduce(counts_dev)
-        return counts_dev
+            pending_handles[b] = torch.distributed.all_reduce(
+                counts_dev[b], async_op=True)
+
+    def drain_handles():
+        for b in range(n_cbuf):
+            if pending_handles[b] is not None:
+                pending_handles[b].wait()
+                pending_handles[b] = None
 
     def run_steps(k):
         # keep ALL slots in flight (submit-ahead depth = n_slots); waiting
         # with only one block queued leaves the GPU idle during the host-side
         # stat aggregation (~20% measured on the J1644 config)
         inflight = []
-        for i in range(k):
+        for i in range(k * args.blocks_per_step):
             inflight.append(eng.submit(pinned[i % len(pinned)]))
             if len(inflight) >= args.slots:
                 agg(eng.wait(inflight.pop(0)))
         while inflight:
             agg(eng.wait(inflight.pop(0)))
+        drain_handles()
         eng.synchronize()
 
     # warmup
@@ -143,14 +188,17 @@ def main():
     elapsed = t1 - t0
     # max over ranks
     if distributed:
-        e = torch.tensor([elapsed], dtype=torch.float64, device="cuda")
+        e = torch.tensor([elapsed], dtype=torch.float64, device=coll_dev)
         torch.distributed.all_reduce(e, op=torch.distributed.ReduceOp.MAX)
         elapsed = float(e.item())
 
-    samples_total = float(n) * args.steps * n_gpus
+    # valid (new-sky) samples only: the overlap tail is reprocessed work
+    n_valid = n - reserved
+    blocks = args.steps * args.blocks_per_step
+    samples_total = float(n_valid) * blocks * n_gpus
     msamps = samples_total / elapsed / 1e6
     ms_per_step = elapsed / args.steps * 1e3
-    per_gpu_sps = float(n) * args.steps / elapsed
+    per_gpu_sps = float(n_valid) * blocks / elapsed
     real_time_ratio = per_gpu_sps / sample_rate
 
     if rank == 0:
@@ -173,6 +221,8 @@ def main():
             "config": {
                 "model": "J1644-4559 coherent-dedispersion single-pulse pipeline",
                 "baseband_input_count": n,
+                "nsamps_reserved": reserved,
+                "blocks_per_step": args.blocks_per_step,
                 "baseband_input_bits": args.bits,
                 "spectrum_channel_count": s,
                 "dm": dm,
