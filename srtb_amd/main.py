@@ -49,6 +49,29 @@ def write_ppm(path: str, argb: np.ndarray) -> None:
         f.write(rgb.tobytes())
 
 
+def _engine_products(eng, cfg: Config, slot: int, res: dict,
+                     raw: np.ndarray | None, counter: int) -> BlockProducts:
+    """Build the dumpable products of one processed block from a native
+    engine slot (detection gate + waterfall/time-series extraction)."""
+    # raw baseband is attached even without a detection: the reference's
+    # write_signal works always carry baseband_data, so a cross-pol
+    # coincidence write can dump the negative stream's baseband too
+    products = BlockProducts(counter=counter, timestamp=counter, raw=raw)
+    gate = res["zero_count"] < (cfg.signal_detect_channel_threshold *
+                                cfg.spectrum_channel_count)
+    detected = [(L, c) for L, c in res["counts"] if c > 0]
+    if gate and detected:
+        products.waterfall = eng.waterfall(slot).cpu().numpy()
+        ts = eng.time_series(slot).cpu().numpy()
+        for L, _count in detected:
+            if L == 1:
+                products.time_series.append((1, ts.copy()))
+            else:
+                products.time_series.append(
+                    (L, eng.boxcar_series(slot, L).cpu().numpy()))
+    return products
+
+
 class GpuStreamPipeline:
     """One data stream on one GPU via the native engine."""
 
@@ -78,27 +101,12 @@ class GpuStreamPipeline:
 
     def process_block(self, raw: np.ndarray, counter: int) -> BlockProducts:
         torch = self.torch
-        cfg = self.cfg
         t = torch.from_numpy(np.ascontiguousarray(raw))
         slot = self.eng.submit(t)
         res = self.eng.wait(slot)
-        products = BlockProducts(counter=counter, timestamp=counter)
-        gate = res["zero_count"] < (cfg.signal_detect_channel_threshold *
-                                    cfg.spectrum_channel_count)
-        detected = [(L, c) for L, c in res["counts"] if c > 0]
         self.last_result = res
         self.last_slot = slot
-        if gate and detected:
-            products.raw = raw
-            products.waterfall = self.eng.waterfall(slot).cpu().numpy()
-            ts = self.eng.time_series(slot).cpu().numpy()
-            for L, _count in detected:
-                if L == 1:
-                    products.time_series.append((1, ts.copy()))
-                else:
-                    products.time_series.append(
-                        (L, self.eng.boxcar_series(slot, L).cpu().numpy()))
-        return products
+        return _engine_products(self.eng, self.cfg, slot, res, raw, counter)
 
     def waterfall_frame(self, width: int, height: int) -> np.ndarray:
         wf = self.eng.waterfall(self.last_slot)
@@ -107,6 +115,98 @@ class GpuStreamPipeline:
         pix = self.C.generate_pixmap(img, ref.COLOR_0, ref.COLOR_1,
                                      ref.COLOR_OVERFLOW)
         return pix.cpu().numpy().view(np.uint32)
+
+
+class GpuMultiPolPipeline:
+    """One PACKED packet stream fanning out into per-polarization sample
+    streams, each through its own engine (reference unpack_pipe.hpp:146-390
+    fans one naocpsr_snap1 / gznupsr_a1 stream into data_stream_id works)."""
+
+    def __init__(self, cfg: Config, nsamps_reserved: int, fmt_name: str):
+        import torch
+        from .ops import native
+        from .pipeline.gpu import _make_engine
+        self.torch = torch
+        self.C = native()
+        self.cfg = cfg
+        self.fmt = bk.resolve_alias(fmt_name)
+        self.n_streams = bk.get_data_stream_count(fmt_name)
+        self.engines = [_make_engine(self.C, cfg, nsamps_reserved, nbits=-8)
+                        for _ in range(self.n_streams)]
+
+    def _fanout(self, raw_t):
+        if self.fmt == "gznupsr_a1":
+            return self.C.unpack_gznupsr_a1(raw_t, self.n_streams)
+        return self.C.unpack_2pol(raw_t, self.fmt)
+
+    def process_blocks(self, raw: np.ndarray,
+                       counter: int) -> list[BlockProducts]:
+        torch = self.torch
+        raw_t = torch.from_numpy(np.ascontiguousarray(raw)).cuda()
+        pols = self._fanout(raw_t)
+        slots = [eng.submit_samples(pol)
+                 for eng, pol in zip(self.engines, pols)]
+        out = []
+        results = []
+        for eng, slot in zip(self.engines, slots):
+            res = eng.wait(slot)
+            results.append(res)
+            # both pols share the same raw packet block (as the reference's
+            # fanned-out works share baseband_data): attach it to each so a
+            # coincidence write from either stream dumps it
+            out.append(_engine_products(eng, self.cfg, slot, res, raw,
+                                        counter))
+        self.last_result = results[0]
+        self.last_slot = slots[0]
+        return out
+
+    def waterfall_frame(self, width: int, height: int) -> np.ndarray:
+        wf = self.engines[0].waterfall(self.last_slot)
+        img = self.C.resample_power(wf, height, width)
+        self.C.normalize_by_mean(img)
+        pix = self.C.generate_pixmap(img, ref.COLOR_0, ref.COLOR_1,
+                                     ref.COLOR_OVERFLOW)
+        return pix.cpu().numpy().view(np.uint32)
+
+
+class CpuMultiPolPipeline:
+    """CPU/NumPy twin of GpuMultiPolPipeline (oracle + plumbing runs)."""
+
+    def __init__(self, cfg: Config, nsamps_reserved: int, fmt_name: str):
+        self.cfg = cfg
+        self.fmt = bk.resolve_alias(fmt_name)
+        self.n_streams = bk.get_data_stream_count(fmt_name)
+        self.pipes = [CpuPipeline(cfg) for _ in range(self.n_streams)]
+
+    def _fanout(self, raw: np.ndarray) -> list[np.ndarray]:
+        if self.fmt == "gznupsr_a1":
+            return ref.unpack_gznupsr_a1(raw, self.n_streams)
+        return list(ref.unpack_naocpsr_snap1(raw))
+
+    def process_blocks(self, raw: np.ndarray,
+                       counter: int) -> list[BlockProducts]:
+        cfg = self.cfg
+        out = []
+        for pipe, samples in zip(self.pipes, self._fanout(raw)):
+            res = pipe.process_samples(samples)
+            self.last_result = res
+            products = BlockProducts(counter=counter, timestamp=counter,
+                                     raw=raw)
+            gate = res["zero_count"] < (cfg.signal_detect_channel_threshold *
+                                        cfg.spectrum_channel_count)
+            if gate and res["detections"]:
+                products.waterfall = res["waterfall"]
+                for L, _cnt, series in res["detections"]:
+                    products.time_series.append((L, series))
+            out.append(products)
+        return out
+
+    def waterfall_frame(self, width: int, height: int) -> np.ndarray:
+        wf = self.last_result["waterfall"]
+        p = np.abs(wf.astype(np.complex64)) ** 2
+        img = ref.resample_power_2d(p, height, width)
+        img = ref.normalize_by_mean(img)
+        return ref.generate_pixmap(img)
 
 
 class CpuStreamPipeline:
@@ -120,11 +220,10 @@ class CpuStreamPipeline:
         cfg = self.cfg
         res = self.pipe.process_block(raw)
         self.last_result = res
-        products = BlockProducts(counter=counter, timestamp=counter)
+        products = BlockProducts(counter=counter, timestamp=counter, raw=raw)
         gate = res["zero_count"] < (cfg.signal_detect_channel_threshold *
                                     cfg.spectrum_channel_count)
         if gate and res["detections"]:
-            products.raw = raw
             products.waterfall = res["waterfall"]
             for L, _cnt, series in res["detections"]:
                 products.time_series.append((L, series))
@@ -176,7 +275,22 @@ def main(argv=None) -> int:
         cfg.baseband_freq_low, cfg.baseband_bandwidth,
         cfg.baseband_sample_rate, cfg.dm, cfg.baseband_reserve_sample)
 
-    make = GpuStreamPipeline if use_gpu else CpuStreamPipeline
+    # polarization fan-out: formats with >1 data stream run one packed
+    # packet stream through per-pol pipelines + cross-pol coincidence writes
+    n_streams = bk.get_data_stream_count(cfg.baseband_format_type)
+
+    def make(cfg_, reserved_):
+        if n_streams > 1:
+            cls = GpuMultiPolPipeline if use_gpu else CpuMultiPolPipeline
+            return cls(cfg_, reserved_, cfg.baseband_format_type)
+        return (GpuStreamPipeline if use_gpu else CpuStreamPipeline)(
+            cfg_, reserved_)
+
+    def process(pipe, raw, counter) -> list[BlockProducts]:
+        if hasattr(pipe, "process_blocks"):
+            return pipe.process_blocks(raw, counter)
+        return [pipe.process_block(raw, counter)]
+
     agg = DetectionAggregator()
     writer = SignalWriteScheduler(
         cfg.baseband_output_file_prefix, cfg.baseband_input_count,
@@ -194,15 +308,16 @@ def main(argv=None) -> int:
         reserved_bytes = reserved * abs(cfg.baseband_input_bits) // 8
 
     if cfg.input_file_path:
-        # file replay: single stream (the reference also replays one stream)
+        # file replay: one packet/sample stream (multi-pol formats carry
+        # n_streams interleaved sample streams per block)
         pipe = make(cfg, reserved)
         reader = FileBlockReader(cfg.input_file_path,
-                                 cfg.baseband_input_count,
-                                 cfg.baseband_input_bits, reserved,
+                                 cfg.baseband_input_count * n_streams,
+                                 cfg.baseband_input_bits, reserved * n_streams,
                                  cfg.input_file_offset_bytes)
         for sample_index, raw in reader:
-            products = pipe.process_block(raw, sample_index)
-            writer.push(products)
+            for products in process(pipe, raw, sample_index):
+                writer.push(products)
             if write_all_f is not None:
                 end = raw.size - reserved_bytes if reserved_bytes else raw.size
                 write_all_f.write(raw[:end].tobytes())
@@ -265,8 +380,8 @@ def main(argv=None) -> int:
                 ep, blk, ts = q.get(timeout=0.2)
             except _queue.Empty:
                 continue
-            products = pipes[ep].process_block(blk, ts)
-            writer.push(products)
+            for products in process(pipes[ep], blk, ts):
+                writer.push(products)
             state["n"] += 1
         state["stop"] = True
         n_blocks = state["n"]

@@ -50,7 +50,12 @@ class CpuPipeline:
         if samples.size != c.baseband_input_count:
             raise ValueError(
                 f"block has {samples.size} samples, expected {c.baseband_input_count}")
+        return self.process_samples(samples)
 
+    def process_samples(self, samples: np.ndarray) -> dict:
+        """Run the chain from already-unpacked float samples (the entry used
+        by multi-polarization fan-out, reference unpack_pipe.hpp:146-390)."""
+        c = self.cfg
         spec = ref.fft_r2c_drop_nyquist(samples)  # Nc bins
         spec = ref.rfi_mitigate_s1(spec, c.mitigate_rfi_average_method_threshold,
                                    c.spectrum_channel_count)

@@ -91,6 +91,81 @@ baseband_output_file_prefix = {tmp_path}/q_
     assert not glob.glob(str(tmp_path / "q_*.bin"))
 
 
+def make_2pol_recording(tmp_path, n_blocks=2):
+    """naocpsr_snap1-style '1 1 2 2' int8 interleave with a dispersed pulse
+    in BOTH polarizations of block 1."""
+    cfg = Config()
+    cfg.baseband_input_count = 1 << 16
+    cfg.spectrum_channel_count = 1 << 6
+    cfg.baseband_input_bits = -8
+    cfg.baseband_freq_low = 1400.0
+    cfg.baseband_bandwidth = 64.0
+    cfg.baseband_sample_rate = 128e6
+    cfg.dm = 40.0
+    cfg.baseband_reserve_sample = False
+    rng = np.random.default_rng(11)
+    n = cfg.baseband_input_count
+    blocks = []
+    for b in range(n_blocks):
+        pols = []
+        for _pol in range(2):
+            if b == 1:
+                t = 0.5 * n / cfg.baseband_sample_rate
+                raw = synthesize_dispersed_pulse(cfg, t, pulse_amp=40.0,
+                                                 noise_sigma=2.0, rng=rng)
+            else:
+                raw = np.clip(np.round(rng.normal(0, 2, n)),
+                              -128, 127).astype(np.int8).view(np.uint8)
+            pols.append(raw.view(np.int8))
+        # "1 1 2 2": 2 samples of pol0 then 2 of pol1, repeating
+        inter = np.empty((n // 2, 4), dtype=np.int8)
+        inter[:, 0:2] = pols[0].reshape(-1, 2)
+        inter[:, 2:4] = pols[1].reshape(-1, 2)
+        blocks.append(inter.reshape(-1).view(np.uint8))
+    path = tmp_path / "rec2pol.bin"
+    np.concatenate(blocks).tofile(path)
+    return cfg, str(path)
+
+
+def test_main_2pol_fanout_detects_both_pols(tmp_path):
+    """A 2-pol naocpsr_snap1 recording fans out into two per-pol pipelines;
+    the dispersed pulse is detected in BOTH pols and coincident products are
+    written (reference unpack_pipe.hpp:146-390 + write_signal_pipe.hpp:81-140)."""
+    cfg, rec = make_2pol_recording(tmp_path)
+    cfg_file = tmp_path / "p2.cfg"
+    cfg_file.write_text(f"""
+baseband_format_type = naocpsr_snap1
+baseband_input_count = 2 ** 16
+spectrum_channel_count = 2 ** 6
+baseband_input_bits = -8
+baseband_freq_low = 1400
+baseband_bandwidth = 64
+baseband_sample_rate = 128 * 1e6
+dm = 40.0
+baseband_reserve_sample = 0
+mitigate_rfi_average_method_threshold = 1e30
+mitigate_rfi_spectral_kurtosis_threshold = 1e30
+signal_detect_signal_noise_threshold = 6
+signal_detect_max_boxcar_length = 16
+input_file_path = {rec}
+baseband_output_file_prefix = {tmp_path}/p2_
+""")
+    rc = main(["--config_file_name", str(cfg_file), "--device", "cpu"])
+    assert rc == 0
+    npys = sorted(glob.glob(str(tmp_path / "p2_*.npy")))
+    # both polarizations of the pulse block wrote a waterfall:
+    # ${prefix}${counter}.0.npy and .1.npy
+    # the block counter is the RAW file sample index: block 1 of a 2-pol
+    # interleaved file starts at raw sample 2 * 2^16
+    counter = 2 << 16
+    assert str(tmp_path / f"p2_{counter}.0.npy") in npys
+    assert str(tmp_path / f"p2_{counter}.1.npy") in npys
+    tims = glob.glob(str(tmp_path / "p2_*.tim"))
+    assert tims, "no time-series product written"
+    bins = glob.glob(str(tmp_path / "p2_*.bin"))
+    assert any(str(counter) in b for b in bins)
+
+
 def test_main_waterfall_ppm(tmp_path):
     cfg, rec = make_recording(tmp_path, n_blocks=1)
     cfg_file = tmp_path / "w.cfg"
