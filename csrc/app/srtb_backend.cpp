@@ -1,10 +1,14 @@
 // srtb-backend — the native MI355X real-time radio-telescope backend
 // executable (reference src/main.cpp:61-333 equivalent).
 //
-// Pipeline: input thread (file replay with dedispersion-overlap seek-back,
-// or recvmmsg UDP ingest pinned to the configured core) → bounded SPSC queue
-// of pinned block buffers → GPU engine (double-buffered HIP streams) →
-// detection gate → product writers (.bin/.npy/.tim).
+// Pipeline: input thread(s) (file replay with dedispersion-overlap seek-back,
+// or one recvmmsg UDP ingest thread per configured endpoint, pinned to the
+// configured cores) → bounded queue of pinned block buffers → GPU engines
+// (double-buffered HIP streams; formats with 2 data streams fan one packed
+// block out into per-polarization engines on a device-side unpack stream,
+// reference unpack_pipe.hpp:146-390) → detection gate + cross-polarization
+// coincidence (reference write_signal_pipe.hpp:81-140) → async product
+// dumps (D2H staged on the slot stream, files written on a pool).
 //
 // Extra (non-reference) flags: --max-blocks N, --dry-run (parse config and
 // print the resolved values — used by CPU tests), --print-config.
@@ -14,8 +18,13 @@
 #include <atomic>
 #include <cinttypes>
 #include <cmath>
+#include <condition_variable>
 #include <cstring>
+#include <deque>
 #include <fstream>
+#include <memory>
+#include <mutex>
+#include <numeric>
 #include <thread>
 #include <vector>
 
@@ -85,6 +94,48 @@ struct BlockMsg {
   bool eof = false;
 };
 
+// multi-producer free list of pinned block buffers (writer threads return
+// buffers too, so the SPSC queue is not enough here)
+class FreeList {
+ public:
+  void put(int i) {
+    {
+      std::lock_guard<std::mutex> l(m_);
+      v_.push_back(i);
+    }
+    cv_.notify_one();
+  }
+  template <typename Stop>
+  bool get(int& out, Stop stop) {
+    std::unique_lock<std::mutex> l(m_);
+    while (v_.empty()) {
+      if (stop()) return false;
+      cv_.wait_for(l, std::chrono::milliseconds(2));
+    }
+    out = v_.back();
+    v_.pop_back();
+    return true;
+  }
+
+ private:
+  std::mutex m_;
+  std::condition_variable cv_;
+  std::vector<int> v_;
+};
+
+// Per-(engine,slot) pinned staging for async detection dumps: waterfall,
+// time series and cumulative sum go D2H on the SLOT stream (ordered before
+// any next block's kernels on that stream), an event marks completion, and
+// the file writes happen on the WritePool — the submit thread never blocks
+// on a 4 GB hipMemcpy (reference posts product writes to asio pools).
+struct DumpStaging {
+  std::complex<float>* wf = nullptr;
+  float* ts = nullptr;
+  float* cumsum = nullptr;
+  hipEvent_t ev = nullptr;
+  std::atomic<bool> busy{false};
+};
+
 }  // namespace
 
 int main(int argc, char** argv) {
@@ -116,16 +167,46 @@ int main(int argc, char** argv) {
   }
   log_level().store(cfg.log_level);
 
-  const size_t reserved = nsamps_reserved(cfg);
+  // polarization fan-out: 2-stream formats run one packed packet stream
+  // through per-pol engines (reference backend_registry get_data_stream_count)
+  const std::string fmt_name = cfg.baseband_format_type;
+  const int n_streams =
+      (fmt_name == "naocpsr_snap1" || fmt_name == "gznupsr_a1") ? 2 : 1;
+
+  const size_t reserved_ref = nsamps_reserved(cfg);
+  size_t reserved = reserved_ref;
+  const bool udp_mode = cfg.input_file_path.empty();
+  if (udp_mode && reserved > 0) {
+    // UDP overlap reservation (the reference left this as a TODO,
+    // udp_receiver_pipe.hpp:38): each block keeps the previous block's tail.
+    // The reserved sample count must align to both the packet payload and
+    // the 2*S valid-window rule, so round it UP to the lcm.
+    const auto fmt = PacketFormat::for_backend(fmt_name);
+    const int bits = std::abs(cfg.baseband_input_bits);
+    const size_t samples_per_payload =
+        fmt.payload_size * 8 / (size_t)bits / (size_t)n_streams;
+    const size_t step =
+        std::lcm(samples_per_payload, 2 * cfg.spectrum_channel_count);
+    const size_t adj = (reserved + step - 1) / step * step;
+    if (adj != reserved)
+      SRTB_APP_LOGI("UDP overlap: nsamps_reserved rounded " << reserved
+                    << " -> " << adj << " (payload/2S alignment)");
+    reserved = adj;
+  }
+
   if (dry_run) {
     std::printf("%s", cfg.dump().c_str());
-    std::printf("nsamps_reserved = %zu\n", reserved);
+    std::printf("data_stream_count = %d\n", n_streams);
+    if (reserved != reserved_ref)
+      std::printf("nsamps_reserved_udp = %zu\n", reserved);
+    std::printf("nsamps_reserved = %zu\n", reserved_ref);
     return 0;
   }
 
   EngineConfig ec;
   ec.baseband_input_count = cfg.baseband_input_count;
-  ec.baseband_input_bits = cfg.baseband_input_bits;
+  // fan-out formats carry int8 per-pol sample streams
+  ec.baseband_input_bits = (n_streams > 1) ? -8 : cfg.baseband_input_bits;
   ec.spectrum_channel_count = cfg.spectrum_channel_count;
   ec.freq_low = cfg.baseband_freq_low;
   ec.bandwidth = cfg.baseband_bandwidth;
@@ -142,26 +223,63 @@ int main(int argc, char** argv) {
                                    << " bits=" << cfg.baseband_input_bits
                                    << " S=" << cfg.spectrum_channel_count
                                    << " dm=" << cfg.dm
-                                   << " reserved=" << reserved);
+                                   << " reserved=" << reserved
+                                   << " streams=" << n_streams);
 
-  PipelineEngine engine(ec, 2);
-  const size_t raw_bytes = engine.raw_bytes();
+  std::vector<std::unique_ptr<PipelineEngine>> engines;
+  for (int i = 0; i < n_streams; ++i)
+    engines.emplace_back(std::make_unique<PipelineEngine>(ec, 2));
+  PipelineEngine& engine = *engines[0];
+  const size_t n_per_pol = cfg.baseband_input_count;
+  const size_t raw_bytes = (n_streams > 1)
+                               ? n_per_pol * (size_t)n_streams
+                               : engine.raw_bytes();
 
-  // ring of pinned block buffers feeding the engine
-  constexpr int kBufs = 4;
+  // device-side fan-out resources: the packed block uploads + unpacks on a
+  // dedicated stream; engines wait on the recorded event (no host sync)
+  hipStream_t fan_stream = nullptr;
+  struct FanRing {
+    uint8_t* d_raw = nullptr;
+    float* pol0 = nullptr;
+    float* pol1 = nullptr;
+    hipEvent_t ev = nullptr;
+  };
+  std::array<FanRing, 2> fan{};
+  int fan_next = 0;
+  if (n_streams > 1) {
+    srtb_hip::check_hip(
+        hipStreamCreateWithFlags(&fan_stream, hipStreamNonBlocking),
+        "fan stream");
+    for (auto& r : fan) {
+      srtb_hip::check_hip(hipMalloc(&r.d_raw, raw_bytes), "fan raw");
+      srtb_hip::check_hip(hipMalloc(&r.pol0, n_per_pol * sizeof(float)),
+                          "fan pol0");
+      srtb_hip::check_hip(hipMalloc(&r.pol1, n_per_pol * sizeof(float)),
+                          "fan pol1");
+      srtb_hip::check_hip(
+          hipEventCreateWithFlags(&r.ev, hipEventDisableTiming), "fan ev");
+    }
+  }
+
+  // ring of pinned block buffers feeding the engines; writer threads return
+  // buffers after async dumps, so release is multi-producer
+  const int n_endpoints =
+      udp_mode ? std::max<int>(1, (int)cfg.udp_receiver_address.size()) : 1;
+  const int kBufs = 4 + 2 * (n_endpoints - 1);
   std::vector<uint8_t*> bufs(kBufs);
   for (auto& b : bufs)
     srtb_hip::check_hip(hipHostMalloc(&b, raw_bytes), "pinned block");
-  SpscQueue<BlockMsg, kBufs> q_in;
-  SpscQueue<int, kBufs> q_free;  // returned buffer indices
-  for (int i = 0; i < kBufs; ++i) q_free.try_push(i);
+  SpscQueue<BlockMsg, 64> q_in;
+  FreeList q_free;
+  for (int i = 0; i < kBufs; ++i) q_free.put(i);
   std::atomic<bool> stop{false};
+  std::atomic<long long> blocks_in{0};
   auto stopped = [&] { return stop.load(std::memory_order_relaxed); };
 
-  std::thread input_thread;
-  if (!cfg.input_file_path.empty()) {
+  std::vector<std::thread> input_threads;
+  if (!udp_mode) {
     // ---- file replay with overlap seek-back (read_file_pipe.hpp:58-126) ----
-    input_thread = std::thread([&] {
+    input_threads.emplace_back([&] {
       set_thread_name("srtb_input");
       std::ifstream f(cfg.input_file_path, std::ios::binary);
       if (!f) {
@@ -170,17 +288,20 @@ int main(int argc, char** argv) {
         return;
       }
       const int bits = std::abs(cfg.baseband_input_bits);
-      size_t res_bytes = reserved * bits / 8;
+      size_t res_bytes = reserved * (size_t)n_streams * bits / 8;
       if (res_bytes >= raw_bytes) res_bytes = 0;
       uint64_t pos = cfg.input_file_offset_bytes;
       long long count = 0;
       while (!stopped()) {
         if (max_blocks >= 0 && count >= max_blocks) break;
         int bi;
-        if (!q_free.pop(bi, stopped)) break;
+        if (!q_free.get(bi, stopped)) break;
         f.seekg((std::streamoff)pos);
         f.read(reinterpret_cast<char*>(bufs[bi]), (std::streamsize)raw_bytes);
-        if ((size_t)f.gcount() < raw_bytes) break;  // EOF
+        if ((size_t)f.gcount() < raw_bytes) {
+          q_free.put(bi);
+          break;  // EOF
+        }
         const uint64_t sample_index = pos * 8 / bits;
         q_in.push(BlockMsg{bi, sample_index, false}, stopped);
         pos += raw_bytes - res_bytes;
@@ -189,140 +310,286 @@ int main(int argc, char** argv) {
       q_in.push(BlockMsg{-1, 0, true}, stopped);
     });
   } else {
-    // ---- UDP ingest (recvmmsg, pinned core) ----
-    input_thread = std::thread([&] {
-      set_thread_name("srtb_udp");
-      if (!cfg.udp_receiver_cpu_preferred.empty())
-        set_thread_affinity(cfg.udp_receiver_cpu_preferred[0]);
-      try {
-        auto fmt = PacketFormat::for_backend(cfg.baseband_format_type);
-        RecvmmsgProvider prov(cfg.udp_receiver_address.at(0),
-                              cfg.udp_receiver_port.at(0),
-                              fmt.packet_size());
-        int bi;
-        if (!q_free.pop(bi, stopped)) return;
-        BlockAssembler assembler(fmt, raw_bytes, bufs[bi]);
-        long long count = 0;
-        while (!stopped()) {
-          const int got = prov.receive();
-          for (int i = 0; i < got; ++i) {
-            if (assembler.push(prov.packet(i), prov.packet_len(i))) {
-              q_in.push(BlockMsg{bi, assembler.block_begin_counter(), false},
-                        stopped);
-              ++count;
-              const auto& st = assembler.stats();
-              SRTB_APP_LOGI("udp block " << assembler.block_begin_counter()
-                            << ": received=" << st.received
-                            << " lost=" << st.lost << " loss_rate="
-                            << st.loss_rate());
-              if (max_blocks >= 0 && count >= max_blocks) {
-                q_in.push(BlockMsg{-1, 0, true}, stopped);
-                return;
+    // ---- UDP ingest: one recvmmsg receiver per endpoint (reference
+    // main.cpp:230-272 spawns N receiver pipes), each pinned to its core ----
+    const int bits = std::abs(cfg.baseband_input_bits);
+    size_t res_bytes = reserved * (size_t)n_streams * bits / 8;
+    if (res_bytes >= raw_bytes) res_bytes = 0;
+    std::atomic<int> eof_count{0};
+    for (int ep = 0; ep < n_endpoints; ++ep) {
+      input_threads.emplace_back([&, ep, res_bytes, bits] {
+        set_thread_name(("srtb_udp" + std::to_string(ep)).c_str());
+        if ((size_t)ep < cfg.udp_receiver_cpu_preferred.size())
+          set_thread_affinity(cfg.udp_receiver_cpu_preferred[ep]);
+        try {
+          auto fmt = PacketFormat::for_backend(fmt_name);
+          RecvmmsgProvider prov(cfg.udp_receiver_address.at(ep),
+                                cfg.udp_receiver_port.at(ep),
+                                fmt.packet_size());
+          int bi;
+          if (!q_free.get(bi, stopped)) return;
+          // assembler fills the NEW region after the overlap head; the
+          // previous block's tail is copied in front on completion
+          BlockAssembler assembler(fmt, raw_bytes - res_bytes,
+                                   bufs[bi] + res_bytes);
+          std::vector<uint8_t> prev_tail(res_bytes, 0);
+          long long count = 0;
+          uint64_t next_log = 1;
+          bool done = false;
+          while (!stopped() && !done) {
+            const int got = prov.receive();
+            for (int i = 0; i < got && !done; ++i) {
+              if (assembler.push(prov.packet(i), prov.packet_len(i))) {
+                if (res_bytes) {
+                  std::memcpy(bufs[bi], prev_tail.data(), res_bytes);
+                  std::memcpy(prev_tail.data(),
+                              bufs[bi] + raw_bytes - res_bytes, res_bytes);
+                }
+                q_in.push(BlockMsg{bi, assembler.block_begin_counter(), false},
+                          stopped);
+                ++count;
+                const auto& st = assembler.stats();
+                if ((uint64_t)count >= next_log) {
+                  SRTB_APP_LOGI("udp[" << ep << "] block "
+                                << assembler.block_begin_counter()
+                                << ": received=" << st.received
+                                << " lost=" << st.lost << " dup="
+                                << st.duplicate << " loss_rate="
+                                << st.loss_rate());
+                  next_log *= 2;  // log blocks 1,2,4,8,... then every 256
+                  if (next_log > 256) next_log = (uint64_t)count + 256;
+                }
+                if (max_blocks >= 0 && count >= max_blocks) {
+                  done = true;  // per-endpoint block budget reached
+                  break;
+                }
+                if (!q_free.get(bi, stopped)) return;
+                assembler.set_block_buffer(bufs[bi] + res_bytes);
+                assembler.begin_next();
               }
-              if (!q_free.pop(bi, stopped)) return;
-              assembler.set_block_buffer(bufs[bi]);
-              assembler.begin_next();
             }
           }
+          blocks_in.fetch_add(count);
+        } catch (const std::exception& e) {
+          SRTB_APP_LOGE("udp[" << ep << "]: " << e.what());
+          stop.store(true);
         }
-      } catch (const std::exception& e) {
-        SRTB_APP_LOGE("udp: " << e.what());
-        stop.store(true);
-      }
-      q_in.push(BlockMsg{-1, 0, true}, stopped);
-    });
+        if (eof_count.fetch_add(1) + 1 == n_endpoints)
+          q_in.push(BlockMsg{-1, 0, true}, stopped);
+      });
+    }
   }
 
   // ---- GPU pipeline + writer loop ----
-  struct InFlight { int slot; int buf_index; uint64_t counter; };
+  struct InFlight {
+    std::array<int, 2> slots{-1, -1};
+    int buf_index;
+    uint64_t counter;
+  };
   std::vector<InFlight> inflight;
   const size_t S = engine.n_channels(), Lw = engine.waterfall_len();
+  const size_t ts_count = engine.ts_count();
   uint64_t blocks = 0, detections = 0;
-  // product writes go to a small pool (reference posts them to asio
-  // thread_pools) so a detection dump never stalls block submission;
-  // device buffers are staged to host HERE (they recycle on next submit),
-  // file IO + fdatasync happen on the pool
   srtb_app::WritePool writers(2);
   const std::string out_prefix = cfg.baseband_output_file_prefix;
+
+  // async dump staging, lazily pinned on first detection
+  std::vector<std::array<DumpStaging, 2>> staging(n_streams);
+
+  // cross-pol / cross-block coincidence state (write_signal_pipe.hpp:81-140):
+  // counters of recent positive blocks; one held-back negative block buffer
+  std::deque<uint64_t> recent_positive;
+  uint64_t block_delta = 0, last_counter = 0;
+  bool have_last = false;
+  int held_buf = -1;
+  uint64_t held_counter = 0;
+
+  auto coincident = [&](uint64_t c) {
+    const double window = 0.45 * (double)block_delta;
+    for (uint64_t t : recent_positive)
+      if (std::abs((double)(int64_t)(c - t)) <= window) return true;
+    return false;
+  };
+
+  auto submit_block = [&](int bi) -> std::array<int, 2> {
+    std::array<int, 2> slots{-1, -1};
+    if (n_streams == 1) {
+      slots[0] = engine.submit(bufs[bi], raw_bytes);
+      return slots;
+    }
+    FanRing& r = fan[fan_next];
+    fan_next ^= 1;
+    srtb_hip::check_hip(hipMemcpyAsync(r.d_raw, bufs[bi], raw_bytes,
+                                       hipMemcpyHostToDevice, fan_stream),
+                        "fan h2d");
+    if (fmt_name == "gznupsr_a1")
+      srtb_hip::check_hip(
+          srtb_hip::unpack_gznupsr_a1(r.d_raw, r.pol0, r.pol1, nullptr,
+                                      nullptr, 2, n_per_pol, nullptr,
+                                      fan_stream),
+          "fan gznupsr");
+    else
+      srtb_hip::check_hip(
+          srtb_hip::unpack_naocpsr_snap1(
+              reinterpret_cast<const int8_t*>(r.d_raw), r.pol0, r.pol1,
+              n_per_pol, nullptr, fan_stream),
+          "fan snap1");
+    srtb_hip::check_hip(hipEventRecord(r.ev, fan_stream), "fan ev rec");
+    slots[0] = engines[0]->submit_samples_device(r.pol0, n_per_pol, NAN, r.ev);
+    slots[1] = engines[1]->submit_samples_device(r.pol1, n_per_pol, NAN, r.ev);
+    return slots;
+  };
+  auto dump_stream = [&](int si, int slot, const srtb_hip::BlockResult& res,
+                         uint64_t counter) {
+    PipelineEngine& eng = *engines[si];
+    DumpStaging& st = staging[si][slot];
+    while (st.busy.load(std::memory_order_acquire))
+      std::this_thread::yield();  // previous dump of this slot still writing
+    if (!st.wf) {
+      srtb_hip::check_hip(
+          hipHostMalloc(&st.wf, S * Lw * sizeof(std::complex<float>)),
+          "stage wf");
+      srtb_hip::check_hip(hipHostMalloc(&st.ts, ts_count * sizeof(float)),
+                          "stage ts");
+      srtb_hip::check_hip(hipHostMalloc(&st.cumsum, ts_count * sizeof(float)),
+                          "stage cumsum");
+      srtb_hip::check_hip(hipEventCreateWithFlags(&st.ev,
+                                                  hipEventDisableTiming),
+                          "stage ev");
+    }
+    hipStream_t strm = eng.stream(slot);
+    srtb_hip::check_hip(hipMemcpyAsync(st.wf, eng.waterfall_ptr(slot),
+                                       S * Lw * sizeof(float2),
+                                       hipMemcpyDeviceToHost, strm),
+                        "wf d2h");
+    srtb_hip::check_hip(hipMemcpyAsync(st.ts, eng.time_series_ptr(slot),
+                                       ts_count * sizeof(float),
+                                       hipMemcpyDeviceToHost, strm),
+                        "ts d2h");
+    srtb_hip::check_hip(hipMemcpyAsync(st.cumsum, eng.cumsum_ptr(slot),
+                                       ts_count * sizeof(float),
+                                       hipMemcpyDeviceToHost, strm),
+                        "cumsum d2h");
+    srtb_hip::check_hip(hipEventRecord(st.ev, strm), "stage ev rec");
+    st.busy.store(true, std::memory_order_release);
+    auto counts = res.counts;
+    DumpStaging* stp = &st;
+    const size_t S_ = S, Lw_ = Lw, tsc = ts_count;
+    const std::string prefix = out_prefix;
+    writers.post([stp, counts, counter, S_, Lw_, tsc, prefix] {
+      srtb_hip::check_hip(hipEventSynchronize(stp->ev), "stage ev sync");
+      write_spectrum_npy(prefix, counter, stp->wf, S_, Lw_);
+      for (const auto& [len, cnt] : counts) {
+        if (cnt == 0) continue;
+        if (len == 1) {
+          write_time_series_tim(prefix, counter, 1, stp->ts, tsc);
+        } else {
+          const size_t n_out = tsc - len;
+          std::vector<float> box(n_out);
+          for (size_t i = 0; i < n_out; ++i)
+            box[i] = stp->cumsum[i + len] - stp->cumsum[i];
+          write_time_series_tim(prefix, counter, len, box.data(), n_out);
+        }
+      }
+      stp->busy.store(false, std::memory_order_release);
+    });
+  };
+
+  // write a block's raw baseband async and release its pinned buffer after
+  auto write_raw_and_release = [&](int bi, uint64_t counter) {
+    uint8_t* p = bufs[bi];
+    const size_t nb = raw_bytes;
+    const std::string prefix = out_prefix;
+    writers.post([&q_free, p, nb, bi, counter, prefix] {
+      write_baseband_bin(prefix, counter, p, nb);
+      q_free.put(bi);
+    });
+  };
 
   auto drain_one = [&] {
     const InFlight w = inflight.front();
     inflight.erase(inflight.begin());
-    auto res = engine.wait(w.slot);
+    std::vector<srtb_hip::BlockResult> res(n_streams);
+    for (int si = 0; si < n_streams; ++si)
+      res[si] = engines[si]->wait(w.slots[si]);
     ++blocks;
-    const bool gate =
-        res.zero_count <
-        cfg.signal_detect_channel_threshold * (double)S;
-    uint64_t positive = 0;
-    for (auto& [len, cnt] : res.counts) positive += cnt;
-    if (gate && positive > 0) {
-      ++detections;
-      SRTB_APP_LOGI("detection in block " << w.counter << " ("
-                                          << positive << " samples over "
-                                          << "threshold)");
-      // raw baseband lives in a pinned buffer that recycles after the
-      // q_free push below — copy it out for the async write
-      {
-        std::vector<uint8_t> bb(bufs[w.buf_index],
-                                bufs[w.buf_index] + raw_bytes);
-        writers.post([out_prefix, counter = w.counter,
-                      bb = std::move(bb)] {
-          write_baseband_bin(out_prefix, counter, bb.data(), bb.size());
-        });
-      }
-      std::vector<std::complex<float>> h_wf(S * Lw);
-      srtb_hip::check_hip(
-          hipMemcpy(h_wf.data(), engine.waterfall_ptr(w.slot),
-                    S * Lw * sizeof(float2), hipMemcpyDeviceToHost),
-          "wf d2h");
-      writers.post([out_prefix, counter = w.counter, S, Lw,
-                    wf = std::move(h_wf)] {
-        write_spectrum_npy(out_prefix, counter, wf.data(), S, Lw);
-      });
-      std::vector<float> h_ts(engine.ts_count());
-      srtb_hip::check_hip(
-          hipMemcpy(h_ts.data(), engine.time_series_ptr(w.slot),
-                    engine.ts_count() * sizeof(float),
-                    hipMemcpyDeviceToHost),
-          "ts d2h");
-      for (auto& [len, cnt] : res.counts) {
-        if (cnt == 0) continue;
-        if (len == 1) {
-          writers.post([out_prefix, counter = w.counter, ts = h_ts] {
-            write_time_series_tim(out_prefix, counter, 1, ts.data(),
-                                  ts.size());
-          });
-        } else {
-          float* box = engine.compute_boxcar(w.slot, len);
-          std::vector<float> h_box(engine.ts_count() - len);
-          srtb_hip::check_hip(hipMemcpy(h_box.data(), box,
-                                        h_box.size() * sizeof(float),
-                                        hipMemcpyDeviceToHost),
-                              "box d2h");
-          writers.post([out_prefix, counter = w.counter, len = len,
-                        bx = std::move(h_box)] {
-            write_time_series_tim(out_prefix, counter, len, bx.data(),
-                                  bx.size());
-          });
-        }
-      }
+    if (have_last && w.counter > last_counter)
+      block_delta = w.counter - last_counter;
+    last_counter = w.counter;
+    have_last = true;
+
+    bool any_positive = false;
+    std::vector<bool> positive(n_streams, false);
+    for (int si = 0; si < n_streams; ++si) {
+      const bool gate =
+          res[si].zero_count <
+          cfg.signal_detect_channel_threshold * (double)S;
+      uint64_t pos = 0;
+      for (auto& [len, cnt] : res[si].counts) pos += cnt;
+      positive[si] = gate && pos > 0;
+      any_positive |= positive[si];
     }
-    q_free.push(w.buf_index, stopped);
+
+    if (any_positive) {
+      ++detections;
+      for (int si = 0; si < n_streams; ++si) {
+        if (!positive[si]) continue;
+        uint64_t pos = 0;
+        for (auto& [len, cnt] : res[si].counts) pos += cnt;
+        SRTB_APP_LOGI("detection in block " << w.counter << " stream " << si
+                                            << " (" << pos
+                                            << " samples over threshold)");
+        dump_stream(si, w.slots[si], res[si], w.counter);
+      }
+      recent_positive.push_back(w.counter);
+      while (recent_positive.size() > 5) recent_positive.pop_front();
+      // a held-back negative block coincident with this positive gets its
+      // baseband dumped too (reference pending-negative re-check)
+      if (held_buf >= 0 && coincident(held_counter)) {
+        write_raw_and_release(held_buf, held_counter);
+        held_buf = -1;
+      }
+      write_raw_and_release(w.buf_index, w.counter);
+    } else if (coincident(w.counter)) {
+      write_raw_and_release(w.buf_index, w.counter);
+    } else {
+      // hold this negative block's buffer back one round so a positive in
+      // the NEXT block (±0.45 window) can still dump it
+      if (held_buf >= 0) q_free.put(held_buf);
+      held_buf = w.buf_index;
+      held_counter = w.counter;
+    }
   };
 
   while (true) {
     BlockMsg msg;
     if (!q_in.pop(msg, stopped)) break;
     if (msg.eof) break;
-    const int slot = engine.submit(bufs[msg.buf_index], raw_bytes);
-    inflight.push_back({slot, msg.buf_index, msg.counter});
+    inflight.push_back({submit_block(msg.buf_index), msg.buf_index,
+                        msg.counter});
     if (inflight.size() >= 2) drain_one();
   }
   while (!inflight.empty()) drain_one();
+  if (held_buf >= 0) q_free.put(held_buf);
   stop.store(true);
-  if (input_thread.joinable()) input_thread.join();
-  engine.synchronize();
+  for (auto& t : input_threads)
+    if (t.joinable()) t.join();
+  for (auto& e : engines) e->synchronize();
   writers.drain();  // all product files on disk before the summary line
+  for (auto& st_arr : staging)
+    for (auto& st : st_arr) {
+      if (st.wf) (void)hipHostFree(st.wf);
+      if (st.ts) (void)hipHostFree(st.ts);
+      if (st.cumsum) (void)hipHostFree(st.cumsum);
+      if (st.ev) (void)hipEventDestroy(st.ev);
+    }
+  for (auto& r : fan) {
+    if (r.d_raw) (void)hipFree(r.d_raw);
+    if (r.pol0) (void)hipFree(r.pol0);
+    if (r.pol1) (void)hipFree(r.pol1);
+    if (r.ev) (void)hipEventDestroy(r.ev);
+  }
+  if (fan_stream) (void)hipStreamDestroy(fan_stream);
   for (auto b : bufs) (void)hipHostFree(b);
 
   SRTB_APP_LOGI("done: " << blocks << " blocks, " << detections

@@ -393,7 +393,8 @@ int PipelineEngine::submit_device(const void* dev_bytes, size_t nbytes,
 }
 
 int PipelineEngine::submit_samples_device(const float* dev_samples,
-                                          size_t count, double dm_override) {
+                                          size_t count, double dm_override,
+                                          hipEvent_t wait_event) {
   if (count != n_) throw std::runtime_error("submit_samples: wrong count");
   const int id = next_slot_;
   next_slot_ = (next_slot_ + 1) % n_slots_;
@@ -402,6 +403,8 @@ int PipelineEngine::submit_samples_device(const float* dev_samples,
     check_hip(hipEventSynchronize(s.done), "slot wait");
     s.busy = false;
   }
+  if (wait_event)
+    check_hip(hipStreamWaitEvent(s.stream, wait_event, 0), "fanout wait");
   enqueue_chain(s, nullptr, dev_samples, dm_override);
   return id;
 }
@@ -444,10 +447,15 @@ float* PipelineEngine::cumsum_ptr(int slot) { return slots_.at(slot)->cumsum; }
 hipStream_t PipelineEngine::stream(int slot) { return slots_.at(slot)->stream; }
 
 float* PipelineEngine::compute_boxcar(int slot, size_t L) {
+  float* box = compute_boxcar_async(slot, L);
+  check_hip(hipStreamSynchronize(slots_.at(slot)->stream), "boxcar sync");
+  return box;
+}
+
+float* PipelineEngine::compute_boxcar_async(int slot, size_t L) {
   Slot& s = *slots_.at(slot);
   const size_t n_out = ts_count_ - L;
   check_hip(boxcar(s.cumsum, s.box, n_out, L, s.stream), "boxcar recompute");
-  check_hip(hipStreamSynchronize(s.stream), "boxcar sync");
   return s.box;
 }
 

@@ -92,8 +92,12 @@ class PipelineEngine {
   // samples on the device (count = baseband_input_count).  Used for
   // multi-polarization formats whose unpack fans one packet stream out into
   // several sample streams (reference unpack_pipe.hpp:146-390).
+  // wait_event (optional): the slot's stream waits on it before the chain —
+  // lets an app-owned fan-out stream (H2D + unpack of the packed block)
+  // feed several engines without host synchronization.
   int submit_samples_device(const float* dev_samples, size_t count,
-                            double dm_override = NAN);
+                            double dm_override = NAN,
+                            hipEvent_t wait_event = nullptr);
 
   // Wait for a slot's chain and return its detection counters.
   BlockResult wait(int slot);
@@ -119,6 +123,10 @@ class PipelineEngine {
   // Recompute one boxcar series into the slot's boxcar buffer and return it
   // (for dumping a detected series).
   float* compute_boxcar(int slot, size_t boxcar_length);
+  // Async variant: enqueue the recompute on the slot's stream and return the
+  // device pointer WITHOUT synchronizing (follow with hipMemcpyAsync on the
+  // same stream + an event for a fully async detection dump).
+  float* compute_boxcar_async(int slot, size_t boxcar_length);
 
   hipStream_t stream(int slot);
 

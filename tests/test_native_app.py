@@ -165,6 +165,133 @@ def test_srtb_backend_file_replay_detects(tmp_path):
 
 
 @pytest.mark.gpu
+def test_srtb_backend_2pol_fanout(tmp_path):
+    """naocpsr_snap1 file replay: one packed 2-pol stream fans out on the
+    GPU into per-pol engines; the dispersed pulse present in both pols is
+    detected in both and products for both streams are written."""
+    from srtb_amd.config import Config
+    from srtb_amd.pipeline.cpu import synthesize_dispersed_pulse
+
+    cfg = Config()
+    cfg.baseband_input_count = 1 << 17
+    cfg.spectrum_channel_count = 1 << 6
+    cfg.baseband_input_bits = -8
+    cfg.baseband_freq_low = 1400.0
+    cfg.baseband_bandwidth = 64.0
+    cfg.baseband_sample_rate = 128e6
+    cfg.dm = 60.0
+    rng = np.random.default_rng(5)
+    n = cfg.baseband_input_count
+    blocks = []
+    for b in range(2):
+        pols = []
+        for _pol in range(2):
+            if b == 1:
+                t = 0.4 * n / cfg.baseband_sample_rate
+                raw = synthesize_dispersed_pulse(cfg, t, pulse_amp=40.0,
+                                                 noise_sigma=2.0, rng=rng)
+            else:
+                raw = np.clip(np.round(rng.normal(0, 2, n)), -128,
+                              127).astype(np.int8).view(np.uint8)
+            pols.append(raw.view(np.int8))
+        inter = np.empty((n // 2, 4), dtype=np.int8)  # "1 1 2 2"
+        inter[:, 0:2] = pols[0].reshape(-1, 2)
+        inter[:, 2:4] = pols[1].reshape(-1, 2)
+        blocks.append(inter.reshape(-1).view(np.uint8))
+    rec = tmp_path / "rec2.bin"
+    np.concatenate(blocks).tofile(rec)
+
+    out = subprocess.run(
+        [BACKEND, "--input_file_path", str(rec),
+         "--baseband_format_type", "naocpsr_snap1",
+         "--baseband_input_count", str(n),
+         "--baseband_input_bits", "8",
+         "--spectrum_channel_count", str(cfg.spectrum_channel_count),
+         "--baseband_freq_low", "1400", "--baseband_bandwidth", "64",
+         "--baseband_sample_rate", "128e6", "--dm", "60.0",
+         "--baseband_reserve_sample", "0",
+         "--mitigate_rfi_average_method_threshold", "1e30",
+         "--mitigate_rfi_spectral_kurtosis_threshold", "1e30",
+         "--signal_detect_signal_noise_threshold", "6",
+         "--signal_detect_max_boxcar_length", "16",
+         "--baseband_output_file_prefix", str(tmp_path) + "/p2_"],
+        capture_output=True, text=True, timeout=300)
+    assert out.returncode == 0, out.stderr
+    assert "blocks=2 detections=1" in out.stdout, (out.stdout, out.stderr)
+    import glob
+    counter = 2 * n  # pulse block starts at raw sample 2*N
+    npys = sorted(glob.glob(str(tmp_path / "p2_*.npy")))
+    # both polarizations dumped a waterfall: .0.npy and .1.npy
+    assert str(tmp_path / f"p2_{counter}.0.npy") in npys, npys
+    assert str(tmp_path / f"p2_{counter}.1.npy") in npys, npys
+    wf = np.load(npys[0])
+    assert wf.shape == (cfg.spectrum_channel_count,
+                        n // 2 // cfg.spectrum_channel_count)
+    assert glob.glob(str(tmp_path / "p2_*.bin"))
+    assert glob.glob(str(tmp_path / "p2_*.tim"))
+
+
+@pytest.mark.gpu
+@pytest.mark.timeout(120)
+def test_srtb_backend_udp_ingest_with_overlap(tmp_path):
+    """Native UDP ingest end-to-end on the GPU: counter-stamped packets with
+    an injected gap; blocks keep the dedispersion-overlap tail of the
+    previous block (UDP overlap reservation) and loss stats are logged."""
+    import socket
+    import threading
+    import time as _time
+
+    n = 1 << 16
+    s_chan = 1 << 5
+    payload = 4096
+    port = 29955
+
+    proc = subprocess.Popen(
+        [BACKEND,
+         "--baseband_format_type", "fastmb_roach2",
+         "--baseband_input_count", str(n),
+         "--baseband_input_bits", "8",
+         "--spectrum_channel_count", str(s_chan),
+         "--baseband_freq_low", "1400", "--baseband_bandwidth", "64",
+         "--baseband_sample_rate", "128e6", "--dm", "0.5",
+         "--baseband_reserve_sample", "1",
+         "--mitigate_rfi_average_method_threshold", "1e30",
+         "--mitigate_rfi_spectral_kurtosis_threshold", "1e30",
+         "--signal_detect_signal_noise_threshold", "1e30",
+         "--udp_receiver_address", "127.0.0.1",
+         "--udp_receiver_port", str(port),
+         "--baseband_output_file_prefix", str(tmp_path) + "/u_",
+         "--max-blocks", "2"],
+        stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True)
+
+    def sender():
+        sock = socket.socket(socket.AF_INET, socket.SOCK_DGRAM)
+        _time.sleep(2.0)  # engine + socket setup
+        rng = np.random.default_rng(9)
+        # enough packets for >2 blocks even after the overlap rounding;
+        # drop packet 7 to exercise gap zero-fill + loss accounting
+        for c in range(3 * n // payload + 8):
+            if c == 7:
+                continue
+            pay = rng.integers(0, 256, payload, dtype=np.uint8).tobytes()
+            sock.sendto(c.to_bytes(8, "little") + pay, ("127.0.0.1", port))
+            _time.sleep(0.001)
+        sock.close()
+
+    t = threading.Thread(target=sender)
+    t.start()
+    try:
+        out, _ = proc.communicate(timeout=90)
+    finally:
+        t.join()
+        if proc.poll() is None:
+            proc.kill()
+    assert proc.returncode == 0, out
+    assert "blocks=2" in out, out
+    assert "loss_rate=" in out, out
+
+
+@pytest.mark.gpu
 def test_srtb_correlator_native(tmp_path):
     rng = np.random.default_rng(1)
     n = 1 << 14
