@@ -38,7 +38,10 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=8)
     ap.add_argument("--warmup", type=int, default=3)
-    ap.add_argument("--n", type=int, default=2**30,
+    # NOTE: under torchrun pass --input-count, not --n: torchrun's argparse
+    # greedily matches --n as an ambiguous prefix of its own --nnodes/... even
+    # after the script positional.
+    ap.add_argument("--n", "--input-count", dest="n", type=int, default=2**30,
                     help="baseband_input_count per block (J1644: 2^30)")
     ap.add_argument("--channels", type=int, default=2**11)
     ap.add_argument("--bits", type=int, default=2)

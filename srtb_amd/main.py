@@ -34,7 +34,8 @@ from .parallel.sharding import (DetectionAggregator, broadcast_config,
                                 init_distributed, shard_streams)
 
 # extra, non-reference options handled by the runner itself
-RUNNER_FLAGS = {"--max-blocks", "--device", "--waterfall-ppm"}
+RUNNER_FLAGS = {"--max-blocks", "--device", "--waterfall-ppm", "--gui-port",
+                "--gui-linger"}
 
 
 def write_ppm(path: str, argb: np.ndarray) -> None:
@@ -158,15 +159,25 @@ class GpuMultiPolPipeline:
                                         counter))
         self.last_result = results[0]
         self.last_slot = slots[0]
+        self.last_slots = slots
         return out
 
-    def waterfall_frame(self, width: int, height: int) -> np.ndarray:
-        wf = self.engines[0].waterfall(self.last_slot)
+    def _frame(self, eng, slot, width: int, height: int) -> np.ndarray:
+        wf = eng.waterfall(slot)
         img = self.C.resample_power(wf, height, width)
         self.C.normalize_by_mean(img)
         pix = self.C.generate_pixmap(img, ref.COLOR_0, ref.COLOR_1,
                                      ref.COLOR_OVERFLOW)
         return pix.cpu().numpy().view(np.uint32)
+
+    def waterfall_frame(self, width: int, height: int) -> np.ndarray:
+        return self._frame(self.engines[0], self.last_slot, width, height)
+
+    def waterfall_frames(self, width: int, height: int) -> list[np.ndarray]:
+        """One pixmap per data stream (the reference opens one waterfall
+        window per stream, src/main.qml:14-28)."""
+        return [self._frame(e, s, width, height)
+                for e, s in zip(self.engines, self.last_slots)]
 
 
 class CpuMultiPolPipeline:
@@ -187,9 +198,11 @@ class CpuMultiPolPipeline:
                        counter: int) -> list[BlockProducts]:
         cfg = self.cfg
         out = []
+        self.last_results = []
         for pipe, samples in zip(self.pipes, self._fanout(raw)):
             res = pipe.process_samples(samples)
             self.last_result = res
+            self.last_results.append(res)
             products = BlockProducts(counter=counter, timestamp=counter,
                                      raw=raw)
             gate = res["zero_count"] < (cfg.signal_detect_channel_threshold *
@@ -207,6 +220,15 @@ class CpuMultiPolPipeline:
         img = ref.resample_power_2d(p, height, width)
         img = ref.normalize_by_mean(img)
         return ref.generate_pixmap(img)
+
+    def waterfall_frames(self, width: int, height: int) -> list[np.ndarray]:
+        out = []
+        for res in self.last_results:
+            p = np.abs(res["waterfall"].astype(np.complex64)) ** 2
+            img = ref.normalize_by_mean(ref.resample_power_2d(p, height,
+                                                              width))
+            out.append(ref.generate_pixmap(img))
+        return out
 
 
 class CpuStreamPipeline:
@@ -243,6 +265,8 @@ def main(argv=None) -> int:
     max_blocks = None
     device = None
     waterfall_every = 0
+    gui_port = 8265
+    gui_linger = 0.0
     cfg_argv = []
     i = 0
     while i < len(argv):
@@ -255,6 +279,12 @@ def main(argv=None) -> int:
             i += 1 if "=" in a else 2
         elif a.startswith("--waterfall-ppm"):
             waterfall_every = int(a.split("=", 1)[1] if "=" in a else argv[i + 1])
+            i += 1 if "=" in a else 2
+        elif a.startswith("--gui-port"):
+            gui_port = int(a.split("=", 1)[1] if "=" in a else argv[i + 1])
+            i += 1 if "=" in a else 2
+        elif a.startswith("--gui-linger"):
+            gui_linger = float(a.split("=", 1)[1] if "=" in a else argv[i + 1])
             i += 1 if "=" in a else 2
         else:
             cfg_argv.append(a)
@@ -290,6 +320,30 @@ def main(argv=None) -> int:
         if hasattr(pipe, "process_blocks"):
             return pipe.process_blocks(raw, counter)
         return [pipe.process_block(raw, counter)]
+
+    # live waterfall GUI (reference Qt windows → built-in browser viewer);
+    # only rank 0 serves when running under torchrun
+    gui = None
+    if cfg.gui_enable and rank == 0:
+        from .gui import WaterfallServer
+        gui = WaterfallServer(port=gui_port).start()
+        print(f"[srtb_amd] live waterfall: http://127.0.0.1:{gui.port}/")
+
+    def gui_update(pipe, blocks_done, written):
+        if gui is None:
+            return
+        try:
+            if hasattr(pipe, "waterfall_frames"):
+                frames = pipe.waterfall_frames(cfg.gui_pixmap_width,
+                                               cfg.gui_pixmap_height)
+            else:
+                frames = [pipe.waterfall_frame(cfg.gui_pixmap_width,
+                                               cfg.gui_pixmap_height)]
+        except Exception:
+            return  # no block processed yet
+        for sid, frame in enumerate(frames):
+            gui.push_frame(sid, frame)
+        gui.update_status(blocks=blocks_done, written=written)
 
     agg = DetectionAggregator()
     writer = SignalWriteScheduler(
@@ -330,6 +384,7 @@ def main(argv=None) -> int:
                 zc, counts = 0, []
             agg.update(int(zc), [(int(a), int(b)) for a, b in counts])
             n_blocks += 1
+            gui_update(pipe, n_blocks, len(writer.written))
             if waterfall_every and n_blocks % waterfall_every == 0:
                 frame = pipe.waterfall_frame(cfg.gui_pixmap_width,
                                              cfg.gui_pixmap_height)
@@ -383,12 +438,19 @@ def main(argv=None) -> int:
             for products in process(pipes[ep], blk, ts):
                 writer.push(products)
             state["n"] += 1
+            gui_update(pipes[ep], state["n"], len(writer.written))
         state["stop"] = True
         n_blocks = state["n"]
 
     if write_all_f is not None:
         write_all_f.close()
     elapsed = time.time() - t0
+    if gui is not None:
+        # keep serving the last frames briefly (the reference GUI keeps its
+        # windows open after the pipeline drains)
+        if gui_linger > 0:
+            time.sleep(gui_linger)
+        gui.stop()
     writer.close()  # flush the async write pool before the summary
     stats = agg.reduce()
     if rank == 0:

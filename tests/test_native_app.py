@@ -265,17 +265,18 @@ def test_srtb_backend_udp_ingest_with_overlap(tmp_path):
         stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True)
 
     def sender():
-        sock = socket.socket(socket.AF_INET, socket.SOCK_DGRAM)
-        _time.sleep(2.0)  # engine + socket setup
-        rng = np.random.default_rng(9)
-        # enough packets for >2 blocks even after the overlap rounding;
+        # stream packets continuously until the backend exits (engine init
+        # on a fresh box can take tens of seconds before the socket binds);
         # drop packet 7 to exercise gap zero-fill + loss accounting
-        for c in range(3 * n // payload + 8):
-            if c == 7:
-                continue
-            pay = rng.integers(0, 256, payload, dtype=np.uint8).tobytes()
-            sock.sendto(c.to_bytes(8, "little") + pay, ("127.0.0.1", port))
-            _time.sleep(0.001)
+        sock = socket.socket(socket.AF_INET, socket.SOCK_DGRAM)
+        rng = np.random.default_rng(9)
+        pay = rng.integers(0, 256, payload, dtype=np.uint8).tobytes()
+        c = 0
+        while proc.poll() is None and c < 500000:
+            if c != 7:
+                sock.sendto(c.to_bytes(8, "little") + pay, ("127.0.0.1", port))
+            c += 1
+            _time.sleep(0.0005)
         sock.close()
 
     t = threading.Thread(target=sender)
