@@ -132,6 +132,13 @@ PipelineEngine::PipelineEngine(const EngineConfig& cfg, int n_slots)
     check_hip(hipMalloc(&window_, n_ * sizeof(float)), "window alloc");
     check_hip(build_window(window_, n_, cfg.window_kind, slots_[0]->stream),
               "window build");
+    // K21: watfft window de-apply table of length l_ (reference
+    // fft_pipe.hpp:350-358); only non-rectangle windows pay the pass
+    check_hip(hipMalloc(&watfft_window_, l_ * sizeof(float)),
+              "watfft window alloc");
+    check_hip(build_window(watfft_window_, l_, cfg.window_kind,
+                           slots_[0]->stream),
+              "watfft window build");
     check_hip(hipStreamSynchronize(slots_[0]->stream), "window sync");
   }
   if (cfg.use_phase_table) {
@@ -174,6 +181,7 @@ PipelineEngine::~PipelineEngine() {
   }
   if (phase_table_) (void)hipFree(phase_table_);
   if (window_) (void)hipFree(window_);
+  if (watfft_window_) (void)hipFree(watfft_window_);
 }
 
 void PipelineEngine::enqueue_chain(Slot& s, const uint8_t* dev_raw,
@@ -247,7 +255,7 @@ void PipelineEngine::enqueue_chain(Slot& s, const uint8_t* dev_raw,
     pre.dm = dm;
     pre.table = table;
     s.nbwd.exec(s.spec, reinterpret_cast<float2*>(s.samples), st, &pre,
-                s.sk_dif_partials);
+                watfft_window_ ? nullptr : s.sk_dif_partials);
     wf = reinterpret_cast<float2*>(s.samples);
   } else {
     check_hip(rfi_dedisperse_fused(
@@ -264,6 +272,9 @@ void PipelineEngine::enqueue_chain(Slot& s, const uint8_t* dev_raw,
     }
   }
   s.wf = wf;
+  if (watfft_window_)
+    check_hip(window_deapply(wf, watfft_window_, s_ * l_, l_, st),
+              "watfft deapply");
 
   const int ncnt = 2 + n_boxcars_;
   check_hip(hipMemsetAsync(s.counters, 0, ncnt * sizeof(unsigned), st),

@@ -147,6 +147,7 @@ class PipelineEngine {
 
   float2* phase_table_ = nullptr;  // shared across slots (read-only)
   float* window_ = nullptr;        // fused FFT window table (null = rect)
+  float* watfft_window_ = nullptr;  // K21 de-apply table, length l_
   bool native_fft_ = false;        // hand-written FORWARD FFT active
   bool fused_unpack_off_ = false;  // SRTB_NO_FUSED_UNPACK kill switch
   bool native_bwd_ = false;        // hand-written BACKWARD (waterfall) FFT

@@ -108,6 +108,11 @@ hipError_t sk_flags(const float2* wf, const float2* s2s4, size_t rows,
                     uint8_t* flags, unsigned* zero_count, hipStream_t stream);
 
 // Zero out flagged rows.
+// K21: waterfall window de-apply — wf[i] /= coef[i mod len] (reference
+// fft_pipe.hpp:350-358; only for non-rectangle FFT windows)
+hipError_t window_deapply(float2* wf, const float* coef, size_t total,
+                          size_t len, hipStream_t stream);
+
 hipError_t sk_zap_rows(float2* wf, const uint8_t* flags, size_t rows,
                        size_t len, hipStream_t stream);
 

@@ -611,6 +611,29 @@ hipError_t sk_flags(const float2* wf, const float2* s2s4, size_t rows,
   return hipSuccess;
 }
 
+// K21: waterfall FFT-window de-apply (reference fft_pipe.hpp:350-358):
+// wf[i] /= coef[i mod len] after the backward waterfall FFT, only when the
+// configured FFT window is not the rectangle.  One elementwise pass.
+__global__ void k_window_deapply(float2* __restrict__ wf,
+                                 const float* __restrict__ coef, size_t total,
+                                 size_t len) {
+  const size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += stride) {
+    const float inv = 1.0f / coef[i % len];
+    const float2 x = wf[i];
+    wf[i] = make_float2(x.x * inv, x.y * inv);
+  }
+}
+
+hipError_t window_deapply(float2* wf, const float* coef, size_t total,
+                          size_t len, hipStream_t stream) {
+  hipLaunchKernelGGL(k_window_deapply, grid_for(total), dim3(kBlock), 0,
+                     stream, wf, coef, total, len);
+  SRTB_CHECK_LAUNCH();
+  return hipSuccess;
+}
+
 hipError_t sk_zap_rows(float2* wf, const uint8_t* flags, size_t rows,
                        size_t len, hipStream_t stream) {
   if (rows > 65535) return hipErrorInvalidValue;

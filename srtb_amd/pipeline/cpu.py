@@ -71,6 +71,11 @@ class CpuPipeline:
 
         n_channels = min(c.spectrum_channel_count, nc)
         wf = ref.waterfall_ifft(spec, n_channels)  # [S][L]
+        if self.window_kind != "rectangle":
+            # K21: de-apply the FFT window after the backward waterfall FFT
+            # (reference fft_pipe.hpp:350-358; rectangle default skips it)
+            coef = ref.window_coefficients(self.window_kind, wf.shape[1])
+            wf = wf / coef[None, :]
         wf = ref.rfi_mitigate_sk(wf, c.mitigate_rfi_spectral_kurtosis_threshold)
 
         det = ref.detect_signals(

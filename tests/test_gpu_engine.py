@@ -210,6 +210,33 @@ def test_engine_fused_window(C):
     np.testing.assert_allclose(ts_gpu / scale, ts_cpu / scale, atol=2e-3)
 
 
+def test_engine_watfft_window_deapply(C):
+    """K21: with a non-rectangle window the waterfall is divided by the
+    length-L window after the backward FFT (reference fft_pipe.hpp:350-358).
+    Compare the engine waterfall against the CPU oracle with and without
+    the window — the two must differ exactly by the de-apply+window chain."""
+    cfg = small_cfg(dm=0.0)
+    rng = np.random.default_rng(23)
+    raw = np.clip(np.round(rng.normal(0, 16, cfg.baseband_input_count)),
+                  -128, 127).astype(np.int8).view(np.uint8)
+    pipe = CpuPipeline(cfg)
+    pipe.window_kind = "hamming"
+    res_cpu = pipe.process_block(raw)
+    eng = make_engine(C, cfg, window_kind=2)
+    slot = eng.submit(torch.from_numpy(raw.copy()))
+    eng.wait(slot)
+    wf_gpu = eng.waterfall(slot).cpu().numpy()
+    wf_cpu = res_cpu["waterfall"].astype(np.complex64)
+    scale = max(np.abs(wf_cpu).max(), 1e-9)
+    np.testing.assert_allclose(wf_gpu / scale, wf_cpu / scale, atol=2e-3)
+    # sanity: the de-apply actually changed the waterfall (window != rect)
+    pipe2 = CpuPipeline(cfg)
+    res_rect = pipe2.process_block(raw)
+    assert not np.allclose(wf_cpu / scale,
+                           res_rect["waterfall"].astype(np.complex64) / scale,
+                           atol=2e-3)
+
+
 def test_engine_hip_graph_replay(C):
     """Graph-captured replay produces identical results to direct enqueue."""
     cfg = small_cfg()
