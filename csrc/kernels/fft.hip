@@ -54,7 +54,42 @@ struct FftPassDescDev {
   unsigned long long tw_mask;  // modulus-1; twiddle enabled via template
   int tw_lo_bits;
   double tw_angle;  // sign * 2*pi / modulus
+  // tuning experiments (host env switches): bit0 = XCD-aware workgroup
+  // swizzle (8 XCDs dispatch round-robin by blockIdx; remapping gives each
+  // XCD a contiguous column range), bit1 = non-temporal loads/stores
+  uint32_t tuning = 0;
 };
+
+// XCD-aware workgroup remap: blockIdx b runs on XCD b%8, so b' =
+// (b&7)*(G/8) + b>>3 hands XCD k the contiguous index range [k*G/8, ...).
+__device__ inline unsigned int xcd_swizzle(unsigned int bx, unsigned int g,
+                                           uint32_t tuning) {
+  if ((tuning & 1u) && (g & 7u) == 0u)
+    return (bx & 7u) * (g >> 3) + (bx >> 3);
+  return bx;
+}
+
+// non-temporal (streaming) access experiment: every element of a column
+// pass is touched exactly once, so L2 retention buys nothing within a pass
+typedef float fvec2_t __attribute__((ext_vector_type(2)));
+__device__ inline float2 ld_stream(const float2* p, uint32_t tuning) {
+  if (tuning & 2u) {
+    const fvec2_t v =
+        __builtin_nontemporal_load(reinterpret_cast<const fvec2_t*>(p));
+    return make_float2(v.x, v.y);
+  }
+  return *p;
+}
+__device__ inline void st_stream(float2* p, float2 v, uint32_t tuning) {
+  if (tuning & 2u) {
+    fvec2_t w;
+    w.x = v.x;
+    w.y = v.y;
+    __builtin_nontemporal_store(w, reinterpret_cast<fvec2_t*>(p));
+  } else {
+    *p = v;
+  }
+}
 
 namespace {
 
@@ -374,7 +409,8 @@ __global__ void __launch_bounds__(256)
               const float2* __restrict__ tw_lo, FftPreopDev pre,
               const uint8_t* __restrict__ raw2) {
   const unsigned long long id =
-      (unsigned long long)blockIdx.x * blockDim.x + threadIdx.x;
+      (unsigned long long)xcd_swizzle(blockIdx.x, gridDim.x, d.tuning) *
+          blockDim.x + threadIdx.x;
   if (id >= n_ffts) return;
   unsigned long long q0, q1, q2;
   digits(id, d, q0, q1, q2);
@@ -394,7 +430,7 @@ __global__ void __launch_bounds__(256)
     const unsigned long long flat = base + off;
     float2 x;
     if constexpr (DEC != 0) x = decN_load<DEC>(raw2, flat);
-    else x = colbase[off];
+    else x = ld_stream(colbase + off, d.tuning);
     if constexpr (PREOP) {
       // fused rfi_dedisperse (spectrum.hip k_rfi_dedisp_fused semantics);
       // `flat` IS the spectrum bin index for in-place column passes
@@ -425,7 +461,7 @@ __global__ void __launch_bounds__(256)
           (q0 * d.tw_f0 * (unsigned long long)k) & d.tw_mask;
       r = cmulf(r, tw_eval(m_, d.tw_angle));
     }
-    ocolbase[(uint32_t)k * stride32] = r;
+    st_stream(ocolbase + (uint32_t)k * stride32, r, d.tuning);
   }
 }
 
@@ -461,7 +497,8 @@ __global__ void __launch_bounds__(256)
                      const float2* __restrict__ tw_lo, FftPreopDev pre,
                      const uint8_t* __restrict__ raw2) {
   const unsigned long long tid =
-      (unsigned long long)blockIdx.x * blockDim.x + threadIdx.x;
+      (unsigned long long)xcd_swizzle(blockIdx.x, gridDim.x, d.tuning) *
+          blockDim.x + threadIdx.x;
   const unsigned long long id = tid >> 1;  // column index
   const int p = (int)(tid & 1);            // which half of the column
   if (id >= n_ffts) return;
@@ -483,7 +520,7 @@ __global__ void __launch_bounds__(256)
     const uint32_t off = (uint32_t)i * stride32;
     float2 x;
     if constexpr (DEC != 0) x = decN_load<DEC>(raw2, base + off);
-    else x = colbase[off];
+    else x = ld_stream(colbase + off, d.tuning);
     if constexpr (PREOP) {
       const unsigned long long flat = base + off;
       bool zap = pre.mean_power && (norm2(x) > thr_mean);
@@ -538,7 +575,7 @@ __global__ void __launch_bounds__(256)
           (q0 * d.tw_f0 * (unsigned long long)k) & d.tw_mask;
       rr = cmulf(rr, tw_eval(m_, d.tw_angle));
     }
-    ocolbase[(uint32_t)k * stride32] = rr;
+    st_stream(ocolbase + (uint32_t)k * stride32, rr, d.tuning);
   }
 }
 
@@ -567,7 +604,8 @@ __global__ void __launch_bounds__(256)
                      const float2* __restrict__ tw_lo, FftPreopDev pre,
                      const uint8_t* __restrict__ raw2) {
   const unsigned long long tid =
-      (unsigned long long)blockIdx.x * blockDim.x + threadIdx.x;
+      (unsigned long long)xcd_swizzle(blockIdx.x, gridDim.x, d.tuning) *
+          blockDim.x + threadIdx.x;
   const unsigned long long id = tid >> 1;  // column index
   const int p = (int)(tid & 1);            // which half of the column
   if (id >= n_ffts) return;
@@ -587,7 +625,7 @@ __global__ void __launch_bounds__(256)
     const uint32_t off = (uint32_t)i * stride32;
     float2 x;
     if constexpr (DEC != 0) x = decN_load<DEC>(raw2, base + off);
-    else x = colbase[off];
+    else x = ld_stream(colbase + off, d.tuning);
     if constexpr (PREOP) {
       const unsigned long long flat = base + off;
       bool zap = pre.mean_power && (norm2(x) > thr_mean);
@@ -631,7 +669,7 @@ __global__ void __launch_bounds__(256)
           (q0 * d.tw_f0 * (unsigned long long)k) & d.tw_mask;
       rr = cmulf(rr, tw_eval(m_, d.tw_angle));
     }
-    ocolbase[(uint32_t)k * stride32] = rr;
+    st_stream(ocolbase + (uint32_t)k * stride32, rr, d.tuning);
   }
 }
 
@@ -1090,6 +1128,20 @@ static bool use_pair32() {
   return v;
 }
 
+// column-pass tuning experiments: SRTB_FFT_SWIZZLE=1 (XCD-aware workgroup
+// remap), SRTB_FFT_NT=1 (non-temporal loads/stores)
+static uint32_t fft_tuning() {
+  static const uint32_t v = [] {
+    uint32_t t = 0;
+    if (const char* e = std::getenv("SRTB_FFT_SWIZZLE"))
+      if (std::atoi(e)) t |= 1u;
+    if (const char* e = std::getenv("SRTB_FFT_NT"))
+      if (std::atoi(e)) t |= 2u;
+    return t;
+  }();
+  return v;
+}
+
 hipError_t fft_col_pass(const float2* in, float2* out, const FftPassDesc& hd,
                         size_t n_ffts, int sign, const float2* tw_n,
                         const float2* tw_hi, const float2* tw_lo,
@@ -1126,6 +1178,7 @@ hipError_t fft_col_pass(const float2* in, float2* out, const FftPassDesc& hd,
   d.tw_mask = hd.tw_mod ? hd.tw_mod - 1 : 0;
   d.tw_lo_bits = hd.tw_lo_bits;
   d.tw_angle = hd.tw_angle;
+  d.tuning = fft_tuning();
   const bool twiddle = hd.tw_mod != 0;
   const uint32_t grid = (uint32_t)((n_ffts + 255) / 256);
 
