@@ -29,6 +29,7 @@
 #include <vector>
 
 #include "../engine/engine.h"
+#include "comm.h"
 #include "config.h"
 #include "runtime.h"
 #include "udp_receiver.h"
@@ -219,6 +220,16 @@ int main(int argc, char** argv) {
   ec.nsamps_reserved = reserved;
   parse_zap_ranges(cfg, ec);
 
+  // multi-GPU: one process per GPU (torchrun or shell-loop launched); the
+  // communicator binds this rank's device BEFORE any engine allocation
+  Comm comm;
+  if (comm.world() > 1) {
+    SRTB_APP_LOGI("multi-GPU: rank " << comm.rank() << "/" << comm.world()
+                                     << " on device " << comm.local_rank());
+    // per-rank product prefix so counters from different ranks never collide
+    cfg.baseband_output_file_prefix += "r" + std::to_string(comm.rank()) + "_";
+  }
+
   SRTB_APP_LOGI("srtb-backend: N=" << cfg.baseband_input_count
                                    << " bits=" << cfg.baseband_input_bits
                                    << " S=" << cfg.spectrum_channel_count
@@ -290,7 +301,8 @@ int main(int argc, char** argv) {
       const int bits = std::abs(cfg.baseband_input_bits);
       size_t res_bytes = reserved * (size_t)n_streams * bits / 8;
       if (res_bytes >= raw_bytes) res_bytes = 0;
-      uint64_t pos = cfg.input_file_offset_bytes;
+      uint64_t pos = cfg.input_file_offset_bytes +
+                     (uint64_t)comm.rank() * (raw_bytes - res_bytes);
       long long count = 0;
       while (!stopped()) {
         if (max_blocks >= 0 && count >= max_blocks) break;
@@ -304,7 +316,8 @@ int main(int argc, char** argv) {
         }
         const uint64_t sample_index = pos * 8 / bits;
         q_in.push(BlockMsg{bi, sample_index, false}, stopped);
-        pos += raw_bytes - res_bytes;
+        // multi-GPU file replay: rank r processes blocks r, r+W, 2W, ...
+        pos += (uint64_t)(raw_bytes - res_bytes) * (uint64_t)comm.world();
         ++count;
       }
       q_in.push(BlockMsg{-1, 0, true}, stopped);
@@ -316,7 +329,14 @@ int main(int argc, char** argv) {
     size_t res_bytes = reserved * (size_t)n_streams * bits / 8;
     if (res_bytes >= raw_bytes) res_bytes = 0;
     std::atomic<int> eof_count{0};
-    for (int ep = 0; ep < n_endpoints; ++ep) {
+    const std::vector<int> my_eps =
+        shard_indices(n_endpoints, comm.world(), comm.rank());
+    const int n_my_eps = (int)my_eps.size();
+    if (n_my_eps == 0) {
+      SRTB_APP_LOGW("rank " << comm.rank() << ": no UDP endpoints assigned");
+      q_in.push(BlockMsg{-1, 0, true}, stopped);
+    }
+    for (const int ep : my_eps) {
       input_threads.emplace_back([&, ep, res_bytes, bits] {
         set_thread_name(("srtb_udp" + std::to_string(ep)).c_str());
         if ((size_t)ep < cfg.udp_receiver_cpu_preferred.size())
@@ -374,7 +394,7 @@ int main(int argc, char** argv) {
           SRTB_APP_LOGE("udp[" << ep << "]: " << e.what());
           stop.store(true);
         }
-        if (eof_count.fetch_add(1) + 1 == n_endpoints)
+        if (eof_count.fetch_add(1) + 1 == n_my_eps)
           q_in.push(BlockMsg{-1, 0, true}, stopped);
       });
     }
@@ -594,6 +614,15 @@ int main(int argc, char** argv) {
 
   SRTB_APP_LOGI("done: " << blocks << " blocks, " << detections
                          << " with detections");
+  if (comm.active()) {
+    // cluster-wide detection statistics over RCCL (xGMI)
+    std::vector<uint64_t> stats{blocks, detections};
+    comm.allreduce_sum(stats);
+    if (comm.rank() == 0)
+      std::printf("[srtb-backend] world=%d blocks=%" PRIu64
+                  " detections=%" PRIu64 " (all ranks)\n",
+                  comm.world(), stats[0], stats[1]);
+  }
   std::printf("[srtb-backend] blocks=%" PRIu64 " detections=%" PRIu64 "\n",
               blocks, detections);
   return 0;

@@ -80,7 +80,7 @@ def build_apps(objects):
         src_path = os.path.join(ROOT, src)
         deps = [src_path] + objects + [
             os.path.join(ROOT, p) for p in (
-                "csrc/app/config.h", "csrc/app/runtime.h",
+                "csrc/app/config.h", "csrc/app/runtime.h", "csrc/app/comm.h",
                 "csrc/app/udp_receiver.h", "csrc/app/writers.h",
                 "csrc/engine/engine.h", "csrc/fft/native_fft.h")]
         if os.path.exists(out) and all(
@@ -89,7 +89,7 @@ def build_apps(objects):
             continue
         cmd = [HIPCC, *HIPCC_FLAGS, "-x", "hip", src_path, "-x", "none", *objects,
                f"-I{ROOT}/csrc/include", f"-L{ROCM}/lib", "-lhipfft",
-               "-lroctx64", "-o", out]
+               "-lroctx64", "-lrccl", "-o", out]
         print("+", " ".join(cmd), flush=True)
         subprocess.check_call(cmd)
 

@@ -293,6 +293,51 @@ def test_srtb_backend_udp_ingest_with_overlap(tmp_path):
 
 
 @pytest.mark.gpu
+def test_srtb_backend_rccl_comm_single_rank(tmp_path):
+    """Native RCCL path: SRTB_FORCE_COMM=1 initializes a real (1-rank)
+    RCCL communicator, runs the file replay sharded loop and all-reduces
+    the end-of-run detection stats over it."""
+    from srtb_amd.config import Config
+    from srtb_amd.pipeline.cpu import synthesize_dispersed_pulse
+
+    cfg = Config()
+    cfg.baseband_input_count = 1 << 17
+    cfg.spectrum_channel_count = 1 << 6
+    cfg.baseband_input_bits = -8
+    cfg.baseband_freq_low = 1400.0
+    cfg.baseband_bandwidth = 64.0
+    cfg.baseband_sample_rate = 128e6
+    cfg.dm = 60.0
+    t_pulse = 0.4 * cfg.baseband_input_count / cfg.baseband_sample_rate
+    pulse = synthesize_dispersed_pulse(cfg, t_pulse, pulse_amp=40.0,
+                                       noise_sigma=2.0)
+    rec = tmp_path / "rec.bin"
+    pulse.tofile(rec)
+    env = dict(os.environ)
+    env.update({"SRTB_FORCE_COMM": "1", "WORLD_SIZE": "1", "RANK": "0",
+                "LOCAL_RANK": "0",
+                "SRTB_RCCL_ID_FILE": str(tmp_path / "rccl_id")})
+    out = subprocess.run(
+        [BACKEND, "--input_file_path", str(rec),
+         "--baseband_input_count", str(cfg.baseband_input_count),
+         "--baseband_input_bits", "-8",
+         "--spectrum_channel_count", str(cfg.spectrum_channel_count),
+         "--baseband_freq_low", "1400", "--baseband_bandwidth", "64",
+         "--baseband_sample_rate", "128e6", "--dm", "60.0",
+         "--baseband_reserve_sample", "0",
+         "--mitigate_rfi_average_method_threshold", "1e30",
+         "--mitigate_rfi_spectral_kurtosis_threshold", "1e30",
+         "--signal_detect_signal_noise_threshold", "6",
+         "--signal_detect_max_boxcar_length", "16",
+         "--baseband_output_file_prefix", str(tmp_path) + "/rc_"],
+        capture_output=True, text=True, timeout=300, env=env)
+    assert out.returncode == 0, out.stderr
+    # the all-ranks line comes from the RCCL allreduce
+    assert "world=1 blocks=1 detections=1 (all ranks)" in out.stdout, \
+        (out.stdout, out.stderr)
+
+
+@pytest.mark.gpu
 def test_srtb_correlator_native(tmp_path):
     rng = np.random.default_rng(1)
     n = 1 << 14
