@@ -8,6 +8,7 @@
 
 #include <arpa/inet.h>
 #include <netinet/in.h>
+#include <sys/mman.h>
 #include <sys/socket.h>
 #include <unistd.h>
 
@@ -36,6 +37,17 @@ struct PacketFormat {
     return c;
   }
 
+  // VDIF-framed formats (gznupsr_a1): the invalid-data flag is bit 31 of
+  // word 0 (reference io/vdif_header.hpp:28-61); others have no validity
+  // marker.  Invalid frames are zero-filled by the assembler like losses.
+  bool vdif = false;
+  bool packet_valid(const uint8_t* pkt) const {
+    if (!vdif) return true;
+    uint32_t w0 = 0;
+    std::memcpy(&w0, pkt, 4);
+    return (w0 >> 31) == 0;  // invalid_data bit clear
+  }
+
   size_t packet_size() const { return header_size + payload_size; }
 
   // 'simple' (reference backend_registry.hpp:36-39) is a headerless linear
@@ -53,6 +65,7 @@ struct PacketFormat {
       f.header_size = 64;
       f.payload_size = 8192;
       f.counter_offset = 24;  // VDIF words 6..7
+      f.vdif = true;
     } else if (name == "simple") {
       f.header_size = 0;
       f.payload_size = 4096;  // nominal; any size accepted (headerless)
@@ -88,11 +101,23 @@ class RecvmmsgProvider {
     timeval tv{0, 100000};
     ::setsockopt(fd_, SOL_SOCKET, SO_RCVTIMEO, &tv, sizeof(tv));
 
-    buf_.resize((size_t)kBatch * packet_size_);
+    // hugepage-backed receive ring (reference recvmmsg provider uses
+    // 2 MB-aligned hugepage buffers); plain pages when unavailable
+    const size_t ring_bytes = (size_t)kBatch * packet_size_;
+    const size_t huge_bytes = (ring_bytes + (2u << 20) - 1) & ~((size_t)(2u << 20) - 1);
+    void* hp = ::mmap(nullptr, huge_bytes, PROT_READ | PROT_WRITE,
+                      MAP_PRIVATE | MAP_ANONYMOUS | MAP_HUGETLB, -1, 0);
+    if (hp != MAP_FAILED) {
+      huge_ = static_cast<uint8_t*>(hp);
+      huge_bytes_ = huge_bytes;
+    } else {
+      buf_.resize(ring_bytes);
+    }
+    uint8_t* ring = huge_ ? huge_ : buf_.data();
     iov_.resize(kBatch);
     msgs_.resize(kBatch);
     for (int i = 0; i < kBatch; ++i) {
-      iov_[i].iov_base = buf_.data() + (size_t)i * packet_size_;
+      iov_[i].iov_base = ring + (size_t)i * packet_size_;
       iov_[i].iov_len = packet_size_;
       std::memset(&msgs_[i], 0, sizeof(mmsghdr));
       msgs_[i].msg_hdr.msg_iov = &iov_[i];
@@ -102,6 +127,7 @@ class RecvmmsgProvider {
 
   ~RecvmmsgProvider() {
     if (fd_ >= 0) ::close(fd_);
+    if (huge_) ::munmap(huge_, huge_bytes_);
   }
 
   // receive up to kBatch packets; returns count (0 on timeout)
@@ -109,21 +135,24 @@ class RecvmmsgProvider {
                                                      kBatch, 0, nullptr)); }
 
   const uint8_t* packet(int i) const {
-    return buf_.data() + (size_t)i * packet_size_;
+    return (huge_ ? huge_ : buf_.data()) + (size_t)i * packet_size_;
   }
+  bool hugepages() const { return huge_ != nullptr; }
   size_t packet_len(int i) const { return msgs_[i].msg_len; }
 
  private:
   int fd_ = -1;
   size_t packet_size_;
   std::vector<uint8_t> buf_;
+  uint8_t* huge_ = nullptr;
+  size_t huge_bytes_ = 0;
   std::vector<iovec> iov_;
   std::vector<mmsghdr> msgs_;
 };
 
 struct LossStats {
   uint64_t received = 0, lost = 0, out_of_order = 0, wrong_size = 0;
-  uint64_t duplicate = 0;
+  uint64_t duplicate = 0, invalid = 0;
   double loss_rate() const {
     const uint64_t t = received + lost;
     return t ? (double)lost / (double)t : 0.0;
@@ -151,6 +180,10 @@ class BlockAssembler {
     if (len != fmt_.packet_size()) {
       ++stats_.wrong_size;
       return false;
+    }
+    if (!fmt_.packet_valid(pkt)) {
+      ++stats_.invalid;
+      return false;  // frame slot stays zero, counted when the block closes
     }
     const uint64_t counter = fmt_.parse_counter(pkt);
     if (!started_) {
