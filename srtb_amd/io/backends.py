@@ -58,6 +58,11 @@ class Backend:
         """Return (counter, timestamp)."""
         return 0, 0
 
+    @staticmethod
+    def packet_valid(buf: bytes) -> bool:
+        """VDIF-framed formats override this with the invalid-data bit."""
+        return True
+
 
 class Simple(Backend):
     name = "simple"
@@ -101,6 +106,12 @@ class GznupsrA1(Backend):
     @staticmethod
     def parse_vdif(buf: bytes) -> VdifHeader:
         return VdifHeader.parse(buf)
+
+    @staticmethod
+    def packet_valid(buf: bytes) -> bool:
+        # VDIF invalid-data flag: bit 31 of word 0 (io/vdif_header.hpp:28-61)
+        w0 = struct.unpack_from("<I", buf)[0]
+        return (w0 >> 31) == 0
 
 
 _BACKENDS = {b.name: b for b in (Simple, FastmbRoach2, NaocpsrSnap1, GznupsrA1)}

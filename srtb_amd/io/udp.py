@@ -43,6 +43,7 @@ class LossStats:
     out_of_order: int = 0
     wrong_size: int = 0
     duplicate: int = 0
+    invalid: int = 0
 
     @property
     def loss_rate(self) -> float:
@@ -109,6 +110,9 @@ class BlockAssembler:
         if len(packet) != self.backend.packet_payload_size:
             self.stats.wrong_size += 1
             return None
+        if not self.backend.packet_valid(packet):
+            self.stats.invalid += 1
+            return None  # slot stays zero, counted as lost on block close
         counter, ts = self.backend.parse_packet(packet)
         if self.begin_counter is None:
             self._reset(counter)
