@@ -85,10 +85,14 @@ class DetectionAggregator:
         """Global stats over all ranks (identity when not distributed)."""
         if not (dist.is_available() and dist.is_initialized()):
             return self.local
+        # the nccl(=RCCL) backend reduces GPU tensors only; gloo wants CPU
+        device = self.device
+        if dist.get_backend() == "nccl" and device.type != "cuda":
+            device = torch.device("cuda")
         t = torch.tensor(
             [self.local.blocks, self.local.detections,
              self.local.zapped_channels, self.local.signal_counts],
-            dtype=torch.int64, device=self.device)
+            dtype=torch.int64, device=device)
         dist.all_reduce(t)
         v = t.tolist()
         return DetectionStats(blocks=v[0], detections=v[1],
