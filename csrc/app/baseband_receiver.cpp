@@ -1,19 +1,33 @@
 // srtb-baseband-receiver — record-only UDP baseband capture
 // (reference src/baseband_receiver.cpp:37-87): recvmmsg ingest → counter-gap
-// block assembly → append blocks to ${prefix}recording.bin.  No GPU.
+// block assembly → composite (stamp ∘ write) pipe appending blocks to
+// ${prefix}recording.bin.  No GPU.  Built on the generic pipe framework
+// (pipe.h) exactly as the reference builds on composite_pipe.
 
 #include <atomic>
 #include <cinttypes>
 #include <cstdio>
 #include <cstring>
 #include <fstream>
+#include <memory>
+#include <optional>
 #include <vector>
 
 #include "config.h"
+#include "pipe.h"
 #include "runtime.h"
 #include "udp_receiver.h"
 
 using namespace srtb_app;
+
+namespace {
+
+struct BlockWork {
+  std::shared_ptr<std::vector<uint8_t>> data;
+  uint64_t counter = 0;
+};
+
+}  // namespace
 
 int main(int argc, char** argv) {
   install_termination_handler();
@@ -39,23 +53,51 @@ int main(int argc, char** argv) {
                         cfg.udp_receiver_port.at(0), fmt.packet_size());
   BlockAssembler assembler(fmt, block_bytes, block.data());
   const std::string out = cfg.baseband_output_file_prefix + "recording.bin";
+
+  // pipeline: [this thread] receive+assemble → queue → [writer pipe thread]
+  // composite (stamp ∘ write_file), the reference's cast∘write composite
+  SpscQueue<BlockWork, 4> q;
+  StopFlag stop;
+  std::atomic<long long> written{0};
+
+  auto stamp = [](const StopFlag&, BlockWork w) -> std::optional<BlockWork> {
+    SRTB_APP_LOGD("block " << w.counter << " (" << w.data->size()
+                           << " bytes) -> writer");
+    return w;
+  };
   std::ofstream f(out, std::ios::binary | std::ios::app);
+  auto write_file = [&f, &written](const StopFlag&,
+                                   BlockWork w) -> std::optional<int> {
+    f.write(reinterpret_cast<const char*>(w.data->data()),
+            (std::streamsize)w.data->size());
+    f.flush();
+    written.fetch_add(1);
+    return 0;
+  };
+  auto writer = start_pipe("srtb-writer", compose(stamp, write_file),
+                           QueueIn<SpscQueue<BlockWork, 4>>{&q},
+                           [](const StopFlag&, int) {}, stop);
+
   long long count = 0;
   while (max_blocks < 0 || count < max_blocks) {
     const int got = prov.receive();
     for (int i = 0; i < got; ++i) {
       if (assembler.push(prov.packet(i), prov.packet_len(i))) {
-        f.write(reinterpret_cast<const char*>(block.data()),
-                (std::streamsize)block_bytes);
-        f.flush();
+        auto data = std::make_shared<std::vector<uint8_t>>(
+            block.begin(), block.end());
+        q.push(BlockWork{std::move(data), assembler.block_begin_counter()},
+               stop);
         ++count;
-        SRTB_APP_LOGI("block " << count << " written (loss_rate="
+        SRTB_APP_LOGI("block " << count << " assembled (loss_rate="
                                << assembler.stats().loss_rate() << ")");
         assembler.begin_next();
         if (max_blocks >= 0 && count >= max_blocks) break;
       }
     }
   }
+  while (written.load() < count) std::this_thread::yield();
+  stop.request_stop();
+  writer.join();
   std::printf("[srtb-baseband-receiver] wrote %lld blocks to %s\n", count,
               out.c_str());
   return 0;

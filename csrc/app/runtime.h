@@ -111,6 +111,7 @@ inline void install_termination_handler() {
 template <typename T, size_t CAP = 2>
 class SpscQueue {
  public:
+  using value_type = T;
   bool try_push(T v) {
     const size_t w = w_.load(std::memory_order_relaxed);
     const size_t r = r_.load(std::memory_order_acquire);

@@ -68,6 +68,7 @@ APP_BINARIES = {
     "srtb-backend": "csrc/app/srtb_backend.cpp",
     "srtb-correlator": "csrc/app/correlator.cpp",
     "srtb-baseband-receiver": "csrc/app/baseband_receiver.cpp",
+    "srtb-pipe-test": "csrc/app/pipe_test.cpp",
 }
 
 
@@ -81,6 +82,7 @@ def build_apps(objects):
         deps = [src_path] + objects + [
             os.path.join(ROOT, p) for p in (
                 "csrc/app/config.h", "csrc/app/runtime.h", "csrc/app/comm.h",
+                "csrc/app/pipe.h",
                 "csrc/app/udp_receiver.h", "csrc/app/writers.h",
                 "csrc/engine/engine.h", "csrc/fft/native_fft.h")]
         if os.path.exists(out) and all(

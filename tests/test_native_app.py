@@ -76,6 +76,20 @@ def test_nsamps_reserved_matches_python_oracle(tmp_path):
     assert got == expect
 
 
+PIPE_TEST = os.path.join(ROOT, "bin", "srtb-pipe-test")
+
+
+@pytest.mark.skipif(not os.path.exists(PIPE_TEST),
+                    reason="native binaries not built")
+def test_native_pipe_framework():
+    """Generic pipe framework (csrc/app/pipe.h): 3-stage bounded pipeline,
+    composite pipes, loose (drop-under-load) push, stop semantics."""
+    out = subprocess.run([PIPE_TEST], capture_output=True, text=True,
+                         timeout=60)
+    assert out.returncode == 0, (out.stdout, out.stderr)
+    assert "PIPE TEST OK" in out.stdout
+
+
 @needs_bins
 def test_native_baseband_receiver_loopback(tmp_path):
     """recvmmsg path over 127.0.0.1 with counter gaps zero-filled."""
