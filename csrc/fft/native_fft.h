@@ -68,6 +68,26 @@ class NativeFft {
     const int t = ilog2z(len);
     passes_.clear();
     if (t <= 12) {
+      // 4^k lengths run as ONE in-place radix-4 DIF pass (n_prefix = 0 →
+      // the kernel's digit addressing degenerates to row*n): swizzled
+      // conflict-free LDS, half the ping-pong Stockham's LDS footprint →
+      // 2 workgroups/CU where Stockham got 1 (measured r02: 1024-point
+      // batched 0.82 → DIF-path target ~0.45 ms; rocFFT 0.41).
+      // Odd log2 lengths (trailing radix-2) keep the Stockham kernel.
+      if ((t & 1) == 0 && t >= 4 && !std::getenv("SRTB_FFT_NOSP_DIF")) {
+        ensure_len_table((uint32_t)len, sign, stream);
+        Pass p;
+        p.kind = PassKind::kDif;
+        p.d.n = (uint32_t)len;
+        p.dif.n = (uint32_t)len;
+        p.dif.out_c2 = L;
+        p.dif.out_elem_coef = 1;
+        p.dif.n_prefix = 0;
+        p.n_ffts = batch;
+        p.tw_n = len_table((uint32_t)len);
+        passes_.push_back(p);
+        return;
+      }
       ensure_len_table((uint32_t)len, sign, stream);
       Pass p;
       p.kind = PassKind::kStockham;
@@ -303,10 +323,12 @@ class NativeFft {
   }
 
   int dif_f(const Pass& p) const {
+    // target <= 80 KiB LDS so 2 workgroups fit per CU (the mid512 lesson:
+    // occupancy dominates run length on these latency-bound LDS kernels)
     int F = 32;
     while (F > 1 &&
            ((size_t)p.dif.n + (size_t)F * (p.dif.n + 2)) * sizeof(float2) >
-               160 * 1024)
+               80 * 1024)
       F >>= 1;
     while (F > 1 && p.n_ffts % F != 0) F >>= 1;
     return F;

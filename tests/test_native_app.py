@@ -297,10 +297,15 @@ def test_srtb_backend_udp_ingest_with_overlap(tmp_path):
     t.start()
     try:
         out, _ = proc.communicate(timeout=90)
-    finally:
+    except subprocess.TimeoutExpired:
+        proc.kill()
+        out, _ = proc.communicate(timeout=10)
         t.join()
+        raise AssertionError(f"backend did not finish; output so far:\n{out}")
+    finally:
         if proc.poll() is None:
             proc.kill()
+        t.join()
     assert proc.returncode == 0, out
     assert "blocks=2" in out, out
     assert "loss_rate=" in out, out
