@@ -45,9 +45,9 @@ def main():
                     help="baseband_input_count per block (J1644: 2^30)")
     ap.add_argument("--channels", type=int, default=2**11)
     ap.add_argument("--bits", type=int, default=2)
-    ap.add_argument("--blocks-per-step", type=int, default=4,
+    ap.add_argument("--blocks-per-step", type=int, default=8,
                     help="baseband blocks processed per timed step (one "
-                    "block ~18 ms; >1 keeps the timed region long enough "
+                    "block ~20 ms; >1 keeps the timed region long enough "
                     "for external GPU-utilization sampling)")
     ap.add_argument("--no-reserve", action="store_true",
                     help="disable the dedispersion overlap reservation "
