@@ -68,13 +68,14 @@ class NativeFft {
     const int t = ilog2z(len);
     passes_.clear();
     if (t <= 12) {
-      // 4^k lengths run as ONE in-place radix-4 DIF pass (n_prefix = 0 →
-      // the kernel's digit addressing degenerates to row*n): swizzled
-      // conflict-free LDS, half the ping-pong Stockham's LDS footprint →
-      // 2 workgroups/CU where Stockham got 1 (measured r02: 1024-point
-      // batched 0.82 → DIF-path target ~0.45 ms; rocFFT 0.41).
-      // Odd log2 lengths (trailing radix-2) keep the Stockham kernel.
-      if ((t & 1) == 0 && t >= 4 && !std::getenv("SRTB_FFT_NOSP_DIF")) {
+      // n = 4096 runs as ONE in-place radix-4 DIF pass (n_prefix = 0 →
+      // the kernel's digit addressing degenerates to row*n): the ping-pong
+      // Stockham at that length needs >160 KB for F=2, so it runs F=1 at
+      // 1 WG/CU — the in-place DIF at 65 KB gets 2 WG/CU and measured
+      // 1.46x faster (1.07 vs 1.56 ms on 2^27 elements).  At 256/1024 the
+      // Stockham kernel measured FASTER than the DIF (0.69/0.77 vs
+      // 0.85/0.84) and keeps the job.
+      if (t == 12 && !std::getenv("SRTB_FFT_NOSP_DIF")) {
         ensure_len_table((uint32_t)len, sign, stream);
         Pass p;
         p.kind = PassKind::kDif;
