@@ -288,6 +288,12 @@ int main(int argc, char** argv) {
   auto stopped = [&] { return stop.load(std::memory_order_relaxed); };
 
   std::vector<std::thread> input_threads;
+  // shared with the input threads, which outlive any branch-local scope
+  // (capturing block-locals by reference here was a use-after-scope that
+  // corrupted main's stack — found by ASan)
+  std::atomic<int> udp_eof_count{0};
+  std::vector<int> udp_my_eps;
+  int udp_n_my_eps = 0;
   if (!udp_mode) {
     // ---- file replay with overlap seek-back (read_file_pipe.hpp:58-126) ----
     input_threads.emplace_back([&] {
@@ -328,15 +334,13 @@ int main(int argc, char** argv) {
     const int bits = std::abs(cfg.baseband_input_bits);
     size_t res_bytes = reserved * (size_t)n_streams * bits / 8;
     if (res_bytes >= raw_bytes) res_bytes = 0;
-    std::atomic<int> eof_count{0};
-    const std::vector<int> my_eps =
-        shard_indices(n_endpoints, comm.world(), comm.rank());
-    const int n_my_eps = (int)my_eps.size();
-    if (n_my_eps == 0) {
+    udp_my_eps = shard_indices(n_endpoints, comm.world(), comm.rank());
+    udp_n_my_eps = (int)udp_my_eps.size();
+    if (udp_n_my_eps == 0) {
       SRTB_APP_LOGW("rank " << comm.rank() << ": no UDP endpoints assigned");
       q_in.push(BlockMsg{-1, 0, true}, stopped);
     }
-    for (const int ep : my_eps) {
+    for (const int ep : udp_my_eps) {
       input_threads.emplace_back([&, ep, res_bytes, bits] {
         set_thread_name(("srtb_udp" + std::to_string(ep)).c_str());
         if ((size_t)ep < cfg.udp_receiver_cpu_preferred.size())
@@ -394,7 +398,7 @@ int main(int argc, char** argv) {
           SRTB_APP_LOGE("udp[" << ep << "]: " << e.what());
           stop.store(true);
         }
-        if (eof_count.fetch_add(1) + 1 == n_my_eps)
+        if (udp_eof_count.fetch_add(1) + 1 == udp_n_my_eps)
           q_in.push(BlockMsg{-1, 0, true}, stopped);
       });
     }

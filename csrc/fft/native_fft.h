@@ -338,9 +338,10 @@ class NativeFft {
   int pick_f(uint32_t n, const Pass& p, size_t n_ffts) const {
     size_t f = (size_t)elems_per_wg() / n;
     if (f < 1) f = 1;
-    // keep LDS under 160 KiB: (n + 2*F*(n+2)) * 8  (tw table + ping-pong)
+    // keep LDS under 160 KiB: (n + 2*F*(n+2) + 16) * 8 (tw tables + ping-pong)
     while (f > 1 &&
-           ((size_t)n + 2ull * f * (n + 2)) * sizeof(float2) > 160 * 1024)
+           ((size_t)n + 2ull * f * (n + 2) + 16) * sizeof(float2) >
+               160 * 1024)
       f >>= 1;
     while (f > 1 && n_ffts % f != 0) f >>= 1;
     // instances in a workgroup must share q1/q2 digits only if... they need

@@ -147,155 +147,6 @@ __device__ inline float2 decN_load(const uint8_t* __restrict__ raw,
   }
 }
 
-// LDS layout: [tw: n][X: F*(n+2)][Y: F*(n+2)] float2s.
-template <bool LOAD_FFAST, bool STORE_FFAST, bool TWIDDLE, int SIGN>
-__global__ void __launch_bounds__(256)
-    k_fft_stockham(const float2* __restrict__ in, float2* __restrict__ out,
-                   FftPassDescDev d, const float2* __restrict__ tw_n,
-                   const float2* __restrict__ tw_hi,
-                   const float2* __restrict__ tw_lo) {
-  extern __shared__ float2 lds[];
-  const int n = d.n;
-  const int nl = d.n_log2;
-  const int F = 1 << d.f_log2;
-  const int ldst = n + 2;
-  float2* ltw = lds;
-  float2* X = lds + n;
-  float2* Y = X + (size_t)F * ldst;
-  const unsigned long long fft0 = (unsigned long long)blockIdx.x << d.f_log2;
-  const int total = F << nl;
-
-  for (int j = threadIdx.x; j < n; j += blockDim.x) ltw[j] = tw_n[j];
-
-  // ---- load (register-staged in chunks of 8 so ≥8 global loads stay in
-  // flight per thread; a naive load→ds_write loop serializes on vmcnt) ----
-  {
-    const int iters = total >> 8;  // blockDim == 256
-    int it = 0;
-    for (; it + 8 <= iters; it += 8) {
-      float2 tmp[8];
-      int lidx[8];
-#pragma unroll
-      for (int k = 0; k < 8; ++k) {
-        const int e = ((it + k) << 8) + threadIdx.x;
-        int f, i;
-        if (LOAD_FFAST) { f = e & (F - 1); i = e >> d.f_log2; }
-        else            { f = e >> nl; i = e & (n - 1); }
-        unsigned long long q0, q1, q2;
-        digits(fft0 + f, d, q0, q1, q2);
-        const unsigned long long base =
-            q0 * d.in_c0 + q1 * d.in_c1 + q2 * d.in_c2;
-        tmp[k] = in[base + (unsigned long long)i * d.in_stride];
-        lidx[k] = f * ldst + i;
-      }
-#pragma unroll
-      for (int k = 0; k < 8; ++k) X[lidx[k]] = tmp[k];
-    }
-    for (int e = (it << 8) + threadIdx.x; e < total; e += blockDim.x) {
-      int f, i;
-      if (LOAD_FFAST) { f = e & (F - 1); i = e >> d.f_log2; }
-      else            { f = e >> nl; i = e & (n - 1); }
-      unsigned long long q0, q1, q2;
-      digits(fft0 + f, d, q0, q1, q2);
-      const unsigned long long base =
-          q0 * d.in_c0 + q1 * d.in_c1 + q2 * d.in_c2;
-      X[f * ldst + i] = in[base + (unsigned long long)i * d.in_stride];
-    }
-  }
-  __syncthreads();
-
-  // ---- radix-4 stages (fftref.fft_small_r4 math) ----
-  int ncur = n, s = 1, tstep_log2 = 0;
-  const int quarter = n >> 2;
-  while ((ncur & 3) == 0 && ncur > 1) {
-    const int m = ncur >> 2;
-    const int s_log2 = tstep_log2;  // s == 1 << tstep_log2 here
-    for (int b = threadIdx.x; b < F * quarter; b += blockDim.x) {
-      const int f = b >> (nl - 2);
-      const int bb = b & (quarter - 1);
-      const int p = bb >> s_log2;
-      const int q = bb & (s - 1);
-      const float2* row = X + f * ldst;
-      const float2 a = row[q + s * p];
-      const float2 bv = row[q + s * (p + m)];
-      const float2 c = row[q + s * (p + 2 * m)];
-      const float2 dv = row[q + s * (p + 3 * m)];
-      const float2 apc = make_float2(a.x + c.x, a.y + c.y);
-      const float2 amc = make_float2(a.x - c.x, a.y - c.y);
-      const float2 bpd = make_float2(bv.x + dv.x, bv.y + dv.y);
-      const float2 bmd = make_float2(bv.x - dv.x, bv.y - dv.y);
-      // si*(b-d) with si = SIGN*i: i*(x,y) = (-y, x)
-      const float2 sibmd = (SIGN > 0) ? make_float2(-bmd.y, bmd.x)
-                                      : make_float2(bmd.y, -bmd.x);
-      const float2 u0 = make_float2(apc.x + bpd.x, apc.y + bpd.y);
-      const float2 u2 = make_float2(apc.x - bpd.x, apc.y - bpd.y);
-      const float2 u1 = make_float2(amc.x + sibmd.x, amc.y + sibmd.y);
-      const float2 u3 = make_float2(amc.x - sibmd.x, amc.y - sibmd.y);
-      const int i1 = p << tstep_log2;
-      const float2 w1 = ltw[i1];
-      const float2 w2 = ltw[i1 * 2];
-      const float2 w3 = ltw[i1 * 3];
-      float2* orow = Y + f * ldst;
-      const int ob = q + ((p * s) << 2);
-      orow[ob] = u0;
-      orow[ob + s] = cmulf(u1, w1);
-      orow[ob + 2 * s] = cmulf(u2, w2);
-      orow[ob + 3 * s] = cmulf(u3, w3);
-    }
-    __syncthreads();
-    float2* t = X; X = Y; Y = t;
-    ncur >>= 2;
-    s <<= 2;
-    tstep_log2 += 2;
-  }
-  // ---- final radix-2 stage (odd log2(n)) ----
-  if (ncur == 2) {
-    const int half = n >> 1;  // == s at this point
-    for (int b = threadIdx.x; b < F * half; b += blockDim.x) {
-      const int f = b >> (nl - 1);
-      const int q = b & (half - 1);
-      const float2* row = X + f * ldst;
-      const float2 a = row[q];
-      const float2 c = row[q + half];
-      Y[f * ldst + q] = make_float2(a.x + c.x, a.y + c.y);
-      Y[f * ldst + q + half] = make_float2(a.x - c.x, a.y - c.y);
-    }
-    __syncthreads();
-    float2* t = X; X = Y; Y = t;
-  }
-
-  // ---- store (+ inter-pass twiddle) ----
-  for (int e = threadIdx.x; e < total; e += blockDim.x) {
-    int f, k;
-    if (STORE_FFAST) { f = e & (F - 1); k = e >> d.f_log2; }
-    else             { f = e >> nl; k = e & (n - 1); }
-    unsigned long long q0, q1, q2;
-    digits(fft0 + f, d, q0, q1, q2);
-    float2 v = X[f * ldst + k];
-    if constexpr (TWIDDLE) {
-      const unsigned long long tf = q0 * d.tw_f0 + q1 * d.tw_f1;
-      const unsigned long long m_ = (tf * (unsigned long long)k) & d.tw_mask;
-      v = cmulf(v, tw_eval(m_, d.tw_angle));
-    }
-    const unsigned long long base =
-        q0 * d.out_c0 + q1 * d.out_c1 + q2 * d.out_c2;
-    out[base + (unsigned long long)k * d.out_stride] = v;
-  }
-}
-
-// ---------------------------------------------------------------------------
-// Register-resident column FFT (one FFT per thread, N ≤ 64 in VGPRs).
-//
-// Used for the strided passes of large composites: a wave's 64 lanes own 64
-// CONSECUTIVE columns, so every load/store instruction touches a contiguous
-// 512-byte run regardless of the column stride — the LDS kernel's F-limited
-// runs (64 B at F=8) were the bandwidth killer on these passes.
-//
-// Algorithm: mixed-radix in-place DIT (radix-4 stages + trailing radix-2),
-// input permutation σ folded into the (compile-time) register indices —
-// validated in srtb_amd/fftref.py and the dit_mixed prototype.
-// ---------------------------------------------------------------------------
-
 constexpr int col_ilog2(int v) { return v <= 1 ? 0 : 1 + col_ilog2(v / 2); }
 
 // schedule: first-to-last radices = [4]*(t/2) + [2 if t odd];
@@ -390,6 +241,195 @@ __device__ inline void col_fft(float2 (&v)[N],
   if constexpr (T == 3) col_stage<N, 8, 2, SIGN>(v, tw_n);
   if constexpr (T == 5) col_stage<N, 32, 2, SIGN>(v, tw_n);
 }
+
+
+// LDS layout: [tw: n][X: F*(n+2)][Y: F*(n+2)] float2s.
+template <bool LOAD_FFAST, bool STORE_FFAST, bool TWIDDLE, int SIGN>
+__global__ void __launch_bounds__(256)
+    k_fft_stockham(const float2* __restrict__ in, float2* __restrict__ out,
+                   FftPassDescDev d, const float2* __restrict__ tw_n,
+                   const float2* __restrict__ tw_hi,
+                   const float2* __restrict__ tw_lo) {
+  extern __shared__ float2 lds[];
+  const int n = d.n;
+  const int nl = d.n_log2;
+  const int F = 1 << d.f_log2;
+  const int ldst = n + 2;
+  float2* ltw = lds;
+  float2* X = lds + n;
+  float2* Y = X + (size_t)F * ldst;
+  float2* ltw16 = Y + (size_t)F * ldst;  // 16-entry W_16 table (radix-16)
+  const unsigned long long fft0 = (unsigned long long)blockIdx.x << d.f_log2;
+  const int total = F << nl;
+
+  for (int j = threadIdx.x; j < n; j += blockDim.x) ltw[j] = tw_n[j];
+  if (nl >= 4 && threadIdx.x < 16)
+    ltw16[threadIdx.x] = tw_n[(size_t)threadIdx.x << (nl - 4)];
+
+  // ---- load (register-staged in chunks of 8 so ≥8 global loads stay in
+  // flight per thread; a naive load→ds_write loop serializes on vmcnt) ----
+  {
+    const int iters = total >> 8;  // blockDim == 256
+    int it = 0;
+    for (; it + 8 <= iters; it += 8) {
+      float2 tmp[8];
+      int lidx[8];
+#pragma unroll
+      for (int k = 0; k < 8; ++k) {
+        const int e = ((it + k) << 8) + threadIdx.x;
+        int f, i;
+        if (LOAD_FFAST) { f = e & (F - 1); i = e >> d.f_log2; }
+        else            { f = e >> nl; i = e & (n - 1); }
+        unsigned long long q0, q1, q2;
+        digits(fft0 + f, d, q0, q1, q2);
+        const unsigned long long base =
+            q0 * d.in_c0 + q1 * d.in_c1 + q2 * d.in_c2;
+        tmp[k] = in[base + (unsigned long long)i * d.in_stride];
+        lidx[k] = f * ldst + i;
+      }
+#pragma unroll
+      for (int k = 0; k < 8; ++k) X[lidx[k]] = tmp[k];
+    }
+    for (int e = (it << 8) + threadIdx.x; e < total; e += blockDim.x) {
+      int f, i;
+      if (LOAD_FFAST) { f = e & (F - 1); i = e >> d.f_log2; }
+      else            { f = e >> nl; i = e & (n - 1); }
+      unsigned long long q0, q1, q2;
+      digits(fft0 + f, d, q0, q1, q2);
+      const unsigned long long base =
+          q0 * d.in_c0 + q1 * d.in_c1 + q2 * d.in_c2;
+      X[f * ldst + i] = in[base + (unsigned long long)i * d.in_stride];
+    }
+  }
+  __syncthreads();
+
+  int ncur = n, s = 1, tstep_log2 = 0;
+  // ---- radix-16 stages: one 16-point register FFT per butterfly (the
+  // col_fft<16> machinery) — halves the LDS round trips + barriers vs
+  // pure radix-4 (1024: 5 stages -> 16,16,4 = 3) ----
+  if (d.tuning & 4u) {
+    const int sixteenth = n >> 4;
+    while ((ncur & 15) == 0 && ncur >= 16) {
+      const int m = ncur >> 4;
+      const int s_log2 = tstep_log2;
+      for (int b = threadIdx.x; b < F * sixteenth; b += blockDim.x) {
+        const int f = b >> (nl - 4);
+        const int bb = b & (sixteenth - 1);
+        const int p = bb >> s_log2;
+        const int q = bb & (s - 1);
+        const float2* row = X + f * ldst;
+        float2 v[16];
+#pragma unroll
+        for (int j = 0; j < 16; ++j)
+          v[col_sigma<16>(j)] = row[q + s * (p + j * m)];
+        col_fft<16, SIGN>(v, ltw16);
+        float2* orow = Y + f * ldst;
+        const int i1 = p << tstep_log2;
+        const int ob = q + ((p * s) << 4);
+        orow[ob] = v[0];
+#pragma unroll
+        for (int j = 1; j < 16; ++j)
+          orow[ob + j * s] = cmulf(v[j], ltw[i1 * j]);
+      }
+      __syncthreads();
+      float2* tswp = X;
+      X = Y;
+      Y = tswp;
+      ncur >>= 4;
+      s <<= 4;
+      tstep_log2 += 4;
+    }
+  }
+  // ---- radix-4 stages (fftref.fft_small_r4 math) ----
+  const int quarter = n >> 2;
+  while ((ncur & 3) == 0 && ncur > 1) {
+    const int m = ncur >> 2;
+    const int s_log2 = tstep_log2;  // s == 1 << tstep_log2 here
+    for (int b = threadIdx.x; b < F * quarter; b += blockDim.x) {
+      const int f = b >> (nl - 2);
+      const int bb = b & (quarter - 1);
+      const int p = bb >> s_log2;
+      const int q = bb & (s - 1);
+      const float2* row = X + f * ldst;
+      const float2 a = row[q + s * p];
+      const float2 bv = row[q + s * (p + m)];
+      const float2 c = row[q + s * (p + 2 * m)];
+      const float2 dv = row[q + s * (p + 3 * m)];
+      const float2 apc = make_float2(a.x + c.x, a.y + c.y);
+      const float2 amc = make_float2(a.x - c.x, a.y - c.y);
+      const float2 bpd = make_float2(bv.x + dv.x, bv.y + dv.y);
+      const float2 bmd = make_float2(bv.x - dv.x, bv.y - dv.y);
+      // si*(b-d) with si = SIGN*i: i*(x,y) = (-y, x)
+      const float2 sibmd = (SIGN > 0) ? make_float2(-bmd.y, bmd.x)
+                                      : make_float2(bmd.y, -bmd.x);
+      const float2 u0 = make_float2(apc.x + bpd.x, apc.y + bpd.y);
+      const float2 u2 = make_float2(apc.x - bpd.x, apc.y - bpd.y);
+      const float2 u1 = make_float2(amc.x + sibmd.x, amc.y + sibmd.y);
+      const float2 u3 = make_float2(amc.x - sibmd.x, amc.y - sibmd.y);
+      const int i1 = p << tstep_log2;
+      const float2 w1 = ltw[i1];
+      const float2 w2 = ltw[i1 * 2];
+      const float2 w3 = ltw[i1 * 3];
+      float2* orow = Y + f * ldst;
+      const int ob = q + ((p * s) << 2);
+      orow[ob] = u0;
+      orow[ob + s] = cmulf(u1, w1);
+      orow[ob + 2 * s] = cmulf(u2, w2);
+      orow[ob + 3 * s] = cmulf(u3, w3);
+    }
+    __syncthreads();
+    float2* t = X; X = Y; Y = t;
+    ncur >>= 2;
+    s <<= 2;
+    tstep_log2 += 2;
+  }
+  // ---- final radix-2 stage (odd log2(n)) ----
+  if (ncur == 2) {
+    const int half = n >> 1;  // == s at this point
+    for (int b = threadIdx.x; b < F * half; b += blockDim.x) {
+      const int f = b >> (nl - 1);
+      const int q = b & (half - 1);
+      const float2* row = X + f * ldst;
+      const float2 a = row[q];
+      const float2 c = row[q + half];
+      Y[f * ldst + q] = make_float2(a.x + c.x, a.y + c.y);
+      Y[f * ldst + q + half] = make_float2(a.x - c.x, a.y - c.y);
+    }
+    __syncthreads();
+    float2* t = X; X = Y; Y = t;
+  }
+
+  // ---- store (+ inter-pass twiddle) ----
+  for (int e = threadIdx.x; e < total; e += blockDim.x) {
+    int f, k;
+    if (STORE_FFAST) { f = e & (F - 1); k = e >> d.f_log2; }
+    else             { f = e >> nl; k = e & (n - 1); }
+    unsigned long long q0, q1, q2;
+    digits(fft0 + f, d, q0, q1, q2);
+    float2 v = X[f * ldst + k];
+    if constexpr (TWIDDLE) {
+      const unsigned long long tf = q0 * d.tw_f0 + q1 * d.tw_f1;
+      const unsigned long long m_ = (tf * (unsigned long long)k) & d.tw_mask;
+      v = cmulf(v, tw_eval(m_, d.tw_angle));
+    }
+    const unsigned long long base =
+        q0 * d.out_c0 + q1 * d.out_c1 + q2 * d.out_c2;
+    out[base + (unsigned long long)k * d.out_stride] = v;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Register-resident column FFT (one FFT per thread, N ≤ 64 in VGPRs).
+//
+// Used for the strided passes of large composites: a wave's 64 lanes own 64
+// CONSECUTIVE columns, so every load/store instruction touches a contiguous
+// 512-byte run regardless of the column stride — the LDS kernel's F-limited
+// runs (64 B at F=8) were the bandwidth killer on these passes.
+//
+// Algorithm: mixed-radix in-place DIT (radix-4 stages + trailing radix-2),
+// input permutation σ folded into the (compile-time) register indices —
+// validated in srtb_amd/fftref.py and the dit_mixed prototype.
+// ---------------------------------------------------------------------------
 
 struct FftPreopDev {
   const double* mean_power;
@@ -1066,6 +1106,33 @@ hipError_t fft_build_twiddle(float2* table, size_t count, double m, int sign,
   return hipSuccess;
 }
 
+static bool use_pair32() {
+  // default ON since the round-2 soak (300 iters on the flagship backward
+  // shape, bit-correct, 5.79 vs 5.92 ms mono; the round-1 memory fault
+  // never reproduced); SRTB_FFT_PAIR32=0 falls back to the mono kernel
+  static const bool v = [] {
+    const char* e = std::getenv("SRTB_FFT_PAIR32");
+    return e ? (std::atoi(e) != 0) : true;
+  }();
+  return v;
+}
+
+// column-pass tuning experiments: SRTB_FFT_SWIZZLE=1 (XCD-aware workgroup
+// remap), SRTB_FFT_NT=1 (non-temporal loads/stores)
+static uint32_t fft_tuning() {
+  static const uint32_t v = [] {
+    uint32_t t = 4u;  // bit2: radix-16 Stockham stages (default ON)
+    if (const char* e = std::getenv("SRTB_FFT_SWIZZLE"))
+      if (std::atoi(e)) t |= 1u;
+    if (const char* e = std::getenv("SRTB_FFT_NT"))
+      if (std::atoi(e)) t |= 2u;
+    if (const char* e = std::getenv("SRTB_FFT_SP16"))
+      if (!std::atoi(e)) t &= ~4u;
+    return t;
+  }();
+  return v;
+}
+
 hipError_t fft_stockham_pass(const float2* in, float2* out,
                              const FftPassDesc& hd, size_t n_ffts, int F,
                              bool load_ffast, bool store_ffast, int sign,
@@ -1091,9 +1158,10 @@ hipError_t fft_stockham_pass(const float2* in, float2* out,
   d.tw_angle = hd.tw_angle;
   const bool twiddle = hd.tw_mod != 0;
   if (n_ffts % F != 0) return hipErrorInvalidValue;
+  d.tuning = fft_tuning();
   const uint32_t grid = (uint32_t)(n_ffts / F);
   const size_t lds_bytes =
-      ((size_t)hd.n + 2ull * F * (hd.n + 2)) * sizeof(float2);
+      ((size_t)hd.n + 2ull * F * (hd.n + 2) + 16) * sizeof(float2);
   if (lds_bytes > 160 * 1024) return hipErrorInvalidValue;
 
 #define DISPATCH4(LF, SF, TW, SG)                                           \
@@ -1123,30 +1191,6 @@ hipError_t fft_stockham_pass(const float2* in, float2* out,
   return hipSuccess;
 }
 
-static bool use_pair32() {
-  // default ON since the round-2 soak (300 iters on the flagship backward
-  // shape, bit-correct, 5.79 vs 5.92 ms mono; the round-1 memory fault
-  // never reproduced); SRTB_FFT_PAIR32=0 falls back to the mono kernel
-  static const bool v = [] {
-    const char* e = std::getenv("SRTB_FFT_PAIR32");
-    return e ? (std::atoi(e) != 0) : true;
-  }();
-  return v;
-}
-
-// column-pass tuning experiments: SRTB_FFT_SWIZZLE=1 (XCD-aware workgroup
-// remap), SRTB_FFT_NT=1 (non-temporal loads/stores)
-static uint32_t fft_tuning() {
-  static const uint32_t v = [] {
-    uint32_t t = 0;
-    if (const char* e = std::getenv("SRTB_FFT_SWIZZLE"))
-      if (std::atoi(e)) t |= 1u;
-    if (const char* e = std::getenv("SRTB_FFT_NT"))
-      if (std::atoi(e)) t |= 2u;
-    return t;
-  }();
-  return v;
-}
 
 hipError_t fft_col_pass(const float2* in, float2* out, const FftPassDesc& hd,
                         size_t n_ffts, int sign, const float2* tw_n,
