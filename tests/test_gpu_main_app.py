@@ -56,6 +56,40 @@ baseband_output_file_prefix = {tmp_path}/g_
     assert glob.glob(str(tmp_path / "g_waterfall_*.ppm"))
 
 
+def test_main_gpu_2pol_fanout(tmp_path):
+    """python -m srtb_amd.main on a 2-pol naocpsr_snap1 recording runs the
+    GpuMultiPolPipeline (fan-out unpack kernel + per-pol engines) and
+    writes per-stream waterfalls."""
+    from tests.test_main_app import make_2pol_recording
+    from srtb_amd.main import main
+
+    cfg, rec = make_2pol_recording(tmp_path)
+    cfg_file = tmp_path / "p2g.cfg"
+    cfg_file.write_text(f"""
+baseband_format_type = naocpsr_snap1
+baseband_input_count = 2 ** 16
+spectrum_channel_count = 2 ** 6
+baseband_input_bits = -8
+baseband_freq_low = 1400
+baseband_bandwidth = 64
+baseband_sample_rate = 128 * 1e6
+dm = 40.0
+baseband_reserve_sample = 0
+mitigate_rfi_average_method_threshold = 1e30
+mitigate_rfi_spectral_kurtosis_threshold = 1e30
+signal_detect_signal_noise_threshold = 6
+signal_detect_max_boxcar_length = 16
+input_file_path = {rec}
+baseband_output_file_prefix = {tmp_path}/p2g_
+""")
+    rc = main(["--config_file_name", str(cfg_file)])
+    assert rc == 0
+    counter = 2 << 16
+    npys = sorted(glob.glob(str(tmp_path / "p2g_*.npy")))
+    assert str(tmp_path / f"p2g_{counter}.0.npy") in npys, npys
+    assert str(tmp_path / f"p2g_{counter}.1.npy") in npys, npys
+
+
 def run_script(path, *args, timeout=600):
     out = subprocess.run([sys.executable, path, *args], capture_output=True,
                          text=True, timeout=timeout, cwd=ROOT)
