@@ -1121,13 +1121,18 @@ static bool use_pair32() {
 // remap), SRTB_FFT_NT=1 (non-temporal loads/stores)
 static uint32_t fft_tuning() {
   static const uint32_t v = [] {
-    uint32_t t = 4u;  // bit2: radix-16 Stockham stages (default ON)
+    uint32_t t = 0;
     if (const char* e = std::getenv("SRTB_FFT_SWIZZLE"))
       if (std::atoi(e)) t |= 1u;
     if (const char* e = std::getenv("SRTB_FFT_NT"))
       if (std::atoi(e)) t |= 2u;
+    // bit2: radix-16 Stockham stages — measured 4-5% SLOWER than radix-4
+    // at every single-pass length (r02 check4: 0.724/0.810/1.468 vs
+    // 0.691/0.773/1.435 ms at 256/1024/2048) despite 5->3 LDS round
+    // trips; the per-butterfly 15-gather inter-stage twiddle reads eat
+    // the barrier savings.  Kept opt-in for future work.
     if (const char* e = std::getenv("SRTB_FFT_SP16"))
-      if (!std::atoi(e)) t &= ~4u;
+      if (std::atoi(e)) t |= 4u;
     return t;
   }();
   return v;
