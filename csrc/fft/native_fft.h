@@ -325,8 +325,14 @@ class NativeFft {
 
   int dif_f(const Pass& p) const {
     // target <= 80 KiB LDS so 2 workgroups fit per CU (the mid512 lesson:
-    // occupancy dominates run length on these latency-bound LDS kernels)
+    // occupancy dominates run length on these latency-bound LDS kernels).
+    // SRTB_FFT_DIF_F overrides (r02 PMC: the DIF pass runs 2.7-3.1 TB/s
+    // vs 5.2-5.6 for the column passes, 38% wave-park — occupancy sweep).
     int F = 32;
+    if (const char* e = std::getenv("SRTB_FFT_DIF_F")) {
+      const int v = std::atoi(e);
+      if (v >= 1 && v <= 64 && (v & (v - 1)) == 0) F = v;
+    }
     while (F > 1 &&
            ((size_t)p.dif.n + (size_t)F * (p.dif.n + 2)) * sizeof(float2) >
                80 * 1024)
