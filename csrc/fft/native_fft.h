@@ -328,7 +328,11 @@ class NativeFft {
     // occupancy dominates run length on these latency-bound LDS kernels).
     // SRTB_FFT_DIF_F overrides (r02 PMC: the DIF pass runs 2.7-3.1 TB/s
     // vs 5.2-5.6 for the column passes, 38% wave-park — occupancy sweep).
-    int F = 32;
+    // F=16 measured fastest (r02 sweep: fwd 2^29 8.89/9.25/9.38/12.96 ms
+    // and bwd 2^18x2048 5.11/5.40/6.01/7.77 at F=16/8/32/4): 35 KB LDS ->
+    // 4 workgroups/CU overlap the per-stage barriers that parked 38% of
+    // wave cycles at F=32 (see profiles/r02_pmc_summary.md).
+    int F = 16;
     if (const char* e = std::getenv("SRTB_FFT_DIF_F")) {
       const int v = std::atoi(e);
       if (v >= 1 && v <= 64 && (v & (v - 1)) == 0) F = v;
