@@ -75,6 +75,18 @@ class NativeFft {
       // 1.46x faster (1.07 vs 1.56 ms on 2^27 elements).  At 256/1024 the
       // Stockham kernel measured FASTER than the DIF (0.69/0.77 vs
       // 0.85/0.84) and keeps the job.
+      // experimental wave-local kernel for 256/512/1024 (one FFT per
+      // wave, no barriers on the data path) — opt-in until measured
+      if (t >= 8 && t <= 10 && std::getenv("SRTB_FFT_WAVE")) {
+        ensure_len_table((uint32_t)len, sign, stream);
+        Pass p;
+        p.kind = PassKind::kWave;
+        p.d.n = (uint32_t)len;
+        p.n_ffts = batch;
+        p.tw_n = len_table((uint32_t)len);
+        passes_.push_back(p);
+        return;
+      }
       if (t == 12 && !std::getenv("SRTB_FFT_NOSP_DIF")) {
         ensure_len_table((uint32_t)len, sign, stream);
         Pass p;
@@ -279,6 +291,11 @@ class NativeFft {
                     "fft_mid512_pass");
           dst = cur;  // in place
           break;
+        case PassKind::kWave:
+          check_hip(fft_wave_pass(cur, dst, p.d.n, p.n_ffts, sign_, p.tw_n,
+                                  stream),
+                    "fft_wave_pass");
+          break;
         case PassKind::kDif: {
           const int F = dif_f(p);
           check_hip(fft_dif_final(cur, dst, p.dif, p.n_ffts, F, sign_,
@@ -298,7 +315,7 @@ class NativeFft {
   }
 
  private:
-  enum class PassKind { kStockham, kCol, kMid, kDif };
+  enum class PassKind { kStockham, kCol, kMid, kDif, kWave };
 
   struct Pass {
     PassKind kind = PassKind::kStockham;

@@ -108,6 +108,12 @@ hipError_t sk_flags(const float2* wf, const float2* s2s4, size_t rows,
                     uint8_t* flags, unsigned* zero_count, hipStream_t stream);
 
 // Zero out flagged rows.
+// wave-local small FFT: n in {256,512,1024}, one FFT per wave (batched,
+// contiguous); tw_n = n-entry twiddle table with the sign baked in
+hipError_t fft_wave_pass(const float2* in, float2* out, uint32_t n,
+                         size_t n_ffts, int sign, const float2* tw_n,
+                         hipStream_t stream);
+
 // K21: waterfall window de-apply — wf[i] /= coef[i mod len] (reference
 // fft_pipe.hpp:350-358; only for non-rectangle FFT windows)
 hipError_t window_deapply(float2* wf, const float* coef, size_t total,

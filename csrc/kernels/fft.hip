@@ -906,6 +906,101 @@ __global__ void k_build_twiddle(float2* __restrict__ t, size_t count,
   }
 }
 
+// ---------------------------------------------------------------------------
+// Wave-local small FFT (n = 64*E, E in {4,8,16} -> n in {256,512,1024}):
+// ONE FFT per wave.  Per-lane E-point register FFT over the stride-64
+// dimension, inter-step twiddle W_n^(lane*k2), then a 64-point DIF across
+// lanes via shfl_xor butterflies (bit-reversed lane output), finished by a
+// padded-row LDS transpose so the global store is coalesced.  No
+// __syncthreads anywhere on the data path (wave-synchronous), one LDS
+// round trip per element vs the ping-pong Stockham's log4(n).
+// Index math validated in scripts/wavefft_proto.py.
+// ---------------------------------------------------------------------------
+
+__device__ inline float2 shfl_xor_f2(float2 x, int mask) {
+  return make_float2(__shfl_xor(x.x, mask, 64), __shfl_xor(x.y, mask, 64));
+}
+
+template <int SIGN, int E>
+__global__ void __launch_bounds__(256)
+    k_fft_wave(const float2* __restrict__ in, float2* __restrict__ out,
+               unsigned long long n_ffts, const float2* __restrict__ tw_n) {
+  constexpr int N = 64 * E;
+  constexpr int EL2 = col_ilog2(E);
+  extern __shared__ float2 lds[];
+  // layout: [ltw: N][w64: 64][per-wave transpose buffers: 4 * 64*(E+1)]
+  float2* ltw = lds;
+  float2* w64 = lds + N;
+  const int wave = (int)(threadIdx.x >> 6);
+  const int lane = (int)(threadIdx.x & 63);
+  float2* buf = w64 + 64 + (size_t)wave * 64 * (E + 1);
+
+  for (int j = (int)threadIdx.x; j < N; j += (int)blockDim.x)
+    ltw[j] = tw_n[j];
+  if (threadIdx.x < 64) w64[threadIdx.x] = tw_n[(size_t)threadIdx.x * E];
+  __syncthreads();  // tables only; the data path below is wave-local
+
+  const unsigned rev = (__brev((unsigned)lane) >> 26);  // bitrev6(lane)
+  const unsigned long long wave0 =
+      (unsigned long long)blockIdx.x * (blockDim.x >> 6) + wave;
+  const unsigned long long wstride =
+      (unsigned long long)gridDim.x * (blockDim.x >> 6);
+
+  for (unsigned long long r = wave0; r < n_ffts; r += wstride) {
+    const float2* __restrict__ src = in + r * N;
+    float2* __restrict__ dst = out + r * N;
+    // step 1: per-lane E-point FFT of src[lane + 64*i2] (sigma-permuted
+    // load, col_fft needs the E-entry table = ltw strided by 64... the
+    // E-point twiddles are W_E^j = W_N^(64 j) = ltw[64*j])
+    float2 v[E];
+#pragma unroll
+    for (int i2 = 0; i2 < E; ++i2)
+      v[col_sigma<E>(i2)] = src[lane + 64 * i2];
+    {
+      // col_fft expects an E-entry table at stride 1: gather through a
+      // small register-resident view is not possible, so give it a lambda-
+      // free path: reuse ltw with TS scaling by calling the stages directly
+      // (col_stage uses tw[j*TS]; our table is ltw with W_E^j at 64*j, so
+      // scale TS by 64)
+      if constexpr (E >= 4) col_stage<E, 4, 4, SIGN, (E / 4) * 64>(v, ltw);
+      if constexpr (E >= 16) col_stage<E, 16, 4, SIGN, 64>(v, ltw);
+      if constexpr (E == 8) col_stage<E, 8, 2, SIGN, 64>(v, ltw);
+    }
+    // step 2: twiddle W_N^(lane*k2)
+#pragma unroll
+    for (int k2 = 1; k2 < E; ++k2)
+      v[k2] = cmulf(v[k2], ltw[lane * k2]);
+    // step 3: 64-point DIF across lanes; output lane holds k1 = rev
+#pragma unroll
+    for (int M = 32; M >= 1; M >>= 1) {
+      const int upper = lane & M;
+      const int j = lane & (M - 1);
+      const float2 tw = w64[j * (32 / M)];
+#pragma unroll
+      for (int k2 = 0; k2 < E; ++k2) {
+        const float2 t = shfl_xor_f2(v[k2], M);
+        if (upper) {
+          v[k2] = cmulf(make_float2(t.x - v[k2].x, t.y - v[k2].y), tw);
+        } else {
+          v[k2] = make_float2(v[k2].x + t.x, v[k2].y + t.y);
+        }
+      }
+    }
+    // step 4: padded-row LDS transpose (row = k1, stride E+1) -> coalesced
+    // store.  Wave-local: the compiler's lgkmcnt waits order write/read.
+#pragma unroll
+    for (int k2 = 0; k2 < E; ++k2) buf[rev * (E + 1) + k2] = v[k2];
+    __builtin_amdgcn_wave_barrier();
+#pragma unroll
+    for (int t = 0; t < E; ++t) {
+      const int k = lane + 64 * t;
+      dst[k] = buf[(k >> EL2) * (E + 1) + (k & (E - 1))];
+    }
+    __builtin_amdgcn_wave_barrier();
+  }
+}
+
+
 // r2c post-process (packed-real trick; fftref.r2c_post):
 //   E = (Z[k]+conj(Z[M-k]))/2,  O = -i/2*(Z[k]-conj(Z[M-k])),
 //   w(k) = exp(-2πi k/(2M)),
@@ -1444,6 +1539,31 @@ hipError_t fft_mid512_pass(const float2* in, float2* out,
       hipLaunchKernelGGL((k_fft_mid512<1, 32>), dim3(grid), dim3(256),
                          lds_bytes, stream, in, out, d, n_ffts, tw_n);
   }
+  SRTB_CHECK_LAUNCH();
+  return hipSuccess;
+}
+
+hipError_t fft_wave_pass(const float2* in, float2* out, uint32_t n,
+                         size_t n_ffts, int sign, const float2* tw_n,
+                         hipStream_t stream) {
+  if (n != 256 && n != 512 && n != 1024) return hipErrorInvalidValue;
+  const int E = (int)(n / 64);
+  const size_t lds_bytes = ((size_t)n + 64 + 4ull * 64 * (E + 1)) *
+                           sizeof(float2);
+  // enough waves to fill the chip ~8x; each wave strides over FFTs
+  uint32_t grid = (uint32_t)((n_ffts + 3) / 4);
+  if (grid > 16384) grid = 16384;
+#define WAVE_LAUNCH(SG, EE)                                                    hipLaunchKernelGGL((k_fft_wave<SG, EE>), dim3(grid), dim3(256),                                 lds_bytes, stream, in, out, n_ffts, tw_n)
+  if (sign < 0) {
+    if (E == 4) WAVE_LAUNCH(-1, 4);
+    else if (E == 8) WAVE_LAUNCH(-1, 8);
+    else WAVE_LAUNCH(-1, 16);
+  } else {
+    if (E == 4) WAVE_LAUNCH(1, 4);
+    else if (E == 8) WAVE_LAUNCH(1, 8);
+    else WAVE_LAUNCH(1, 16);
+  }
+#undef WAVE_LAUNCH
   SRTB_CHECK_LAUNCH();
   return hipSuccess;
 }
