@@ -6,6 +6,7 @@
 #include <fcntl.h>
 #include <unistd.h>
 
+#include <cerrno>
 #include <complex>
 #include <cstdint>
 #include <cstdio>
@@ -38,15 +39,20 @@ inline void write_baseband_bin(const std::string& prefix, uint64_t counter,
   ::close(fd);
 }
 
-// minimal .npy (format 1.0) writer for complex64 [rows][cols]
+// minimal .npy (format 1.0) writer for complex64 [rows][cols]; the
+// per-polarization index is claimed with O_EXCL so concurrent pool writers
+// (both pols of one block) get distinct .i.npy names
 inline void write_spectrum_npy(const std::string& prefix, uint64_t counter,
                                const std::complex<float>* data, size_t rows,
                                size_t cols) {
-  std::string path;
+  int fd = -1;
   for (int i = 0;; ++i) {  // multiple polarizations: first free index
-    path = prefix + std::to_string(counter) + "." + std::to_string(i) +
-           ".npy";
-    if (!file_exists(path)) break;
+    const std::string path = prefix + std::to_string(counter) + "." +
+                             std::to_string(i) + ".npy";
+    fd = ::open(path.c_str(), O_WRONLY | O_CREAT | O_EXCL, 0644);
+    if (fd >= 0) break;
+    if (errno != EEXIST)
+      throw std::runtime_error("cannot create " + path);
   }
   std::ostringstream hd;
   hd << "{'descr': '<c8', 'fortran_order': False, 'shape': (" << rows << ", "
@@ -56,13 +62,27 @@ inline void write_spectrum_npy(const std::string& prefix, uint64_t counter,
   const size_t pad = (64 - total % 64) % 64;
   header += std::string(pad, ' ');
   header += '\n';
-  std::ofstream f(path, std::ios::binary);
-  f.write("\x93NUMPY\x01\x00", 8);
+  std::string head;
+  head += std::string("\x93NUMPY\x01", 8 - 1);
+  head += '\0';
   const uint16_t hlen = (uint16_t)header.size();
-  f.write(reinterpret_cast<const char*>(&hlen), 2);
-  f.write(header.data(), header.size());
-  f.write(reinterpret_cast<const char*>(data),
-          rows * cols * sizeof(std::complex<float>));
+  head.append(reinterpret_cast<const char*>(&hlen), 2);
+  head += header;
+  size_t off = 0;
+  while (off < head.size()) {
+    const ssize_t w = ::write(fd, head.data() + off, head.size() - off);
+    if (w <= 0) break;
+    off += (size_t)w;
+  }
+  const char* body = reinterpret_cast<const char*>(data);
+  const size_t nbytes = rows * cols * sizeof(std::complex<float>);
+  off = 0;
+  while (off < nbytes) {
+    const ssize_t w = ::write(fd, body + off, nbytes - off);
+    if (w <= 0) break;
+    off += (size_t)w;
+  }
+  ::close(fd);
 }
 
 inline void write_time_series_tim(const std::string& prefix, uint64_t counter,

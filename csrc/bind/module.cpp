@@ -535,13 +535,24 @@ class PyEngine {
     eng_ = std::make_unique<PipelineEngine>(c, (int)n_slots);
   }
 
+  // device-tensor submissions: the tensor was produced on torch's current
+  // stream while the engine enqueues on its own slot stream — record a
+  // producer event and have the slot stream wait on it (no host sync)
+  hipEvent_t producer_event() {
+    if (!prod_ev_)
+      check(hipEventCreateWithFlags(&prod_ev_, hipEventDisableTiming),
+            "producer event create");
+    check(hipEventRecord(prod_ev_, cur_stream()), "producer event record");
+    return prod_ev_;
+  }
+
   int64_t submit(torch::Tensor raw, double dm_override) {
     TORCH_CHECK(raw.is_contiguous());
     TORCH_CHECK((size_t)raw.numel() * raw.element_size() == eng_->raw_bytes(),
                 "raw block has wrong byte size");
     if (raw.is_cuda())
       return eng_->submit_device(raw.data_ptr(), eng_->raw_bytes(),
-                                 dm_override);
+                                 dm_override, producer_event());
     return eng_->submit(raw.data_ptr(), eng_->raw_bytes(), dm_override);
   }
 
@@ -549,7 +560,8 @@ class PyEngine {
     TORCH_CHECK(samples.is_cuda() && samples.is_contiguous());
     TORCH_CHECK(samples.scalar_type() == torch::kFloat32);
     return eng_->submit_samples_device(samples.data_ptr<float>(),
-                                       samples.numel(), dm_override);
+                                       samples.numel(), dm_override,
+                                       producer_event());
   }
 
   py::dict wait(int64_t slot) {
@@ -602,6 +614,12 @@ class PyEngine {
 
  private:
   std::unique_ptr<PipelineEngine> eng_;
+  hipEvent_t prod_ev_ = nullptr;
+
+ public:
+  ~PyEngine() {
+    if (prod_ev_) (void)hipEventDestroy(prod_ev_);
+  }
 };
 
 }  // namespace

@@ -389,7 +389,7 @@ int PipelineEngine::submit(const void* host_bytes, size_t nbytes,
 }
 
 int PipelineEngine::submit_device(const void* dev_bytes, size_t nbytes,
-                                  double dm_override) {
+                                  double dm_override, hipEvent_t wait_event) {
   if (nbytes != raw_bytes_) throw std::runtime_error("submit: wrong size");
   const int id = next_slot_;
   next_slot_ = (next_slot_ + 1) % n_slots_;
@@ -398,6 +398,8 @@ int PipelineEngine::submit_device(const void* dev_bytes, size_t nbytes,
     check_hip(hipEventSynchronize(s.done), "slot wait");
     s.busy = false;
   }
+  if (wait_event)
+    check_hip(hipStreamWaitEvent(s.stream, wait_event, 0), "producer wait");
   enqueue_chain(s, static_cast<const uint8_t*>(dev_bytes), nullptr,
                 dm_override);
   return id;

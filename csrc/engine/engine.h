@@ -84,9 +84,11 @@ class PipelineEngine {
   int submit(const void* host_bytes, size_t nbytes, double dm_override = NAN);
 
   // Enqueue the chain reading raw bytes already on the device (e.g. from a
-  // torch tensor); caller guarantees lifetime until wait().
+  // torch tensor); caller guarantees lifetime until wait().  wait_event
+  // (optional): slot stream waits on it first — pass an event recorded on
+  // the stream that produced dev_bytes (e.g. torch's current stream).
   int submit_device(const void* dev_bytes, size_t nbytes,
-                    double dm_override = NAN);
+                    double dm_override = NAN, hipEvent_t wait_event = nullptr);
 
   // Enqueue the chain starting at the R2C FFT from already-unpacked float
   // samples on the device (count = baseband_input_count).  Used for

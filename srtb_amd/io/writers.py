@@ -37,9 +37,19 @@ def spectrum_npy_path(prefix: str, counter: int) -> str:
 
 
 def write_spectrum_npy(prefix: str, counter: int, waterfall: np.ndarray) -> str:
-    path = spectrum_npy_path(prefix, counter)
-    np.save(path, np.asarray(waterfall, dtype=np.complex64))
-    return path
+    """Write the waterfall as ${prefix}${counter}.${i}.npy with the first
+    FREE index i — claimed with O_EXCL so concurrent writers (two pols of
+    one block on the async pool) get distinct indices instead of both
+    scanning to .0.npy."""
+    i = 0
+    while True:
+        path = f"{prefix}{counter}.{i}.npy"
+        try:
+            with open(path, "xb") as f:
+                np.save(f, np.asarray(waterfall, dtype=np.complex64))
+            return path
+        except FileExistsError:
+            i += 1
 
 
 def write_time_series_tim(prefix: str, counter: int, boxcar_length: int,
