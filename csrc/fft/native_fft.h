@@ -75,9 +75,12 @@ class NativeFft {
       // 1.46x faster (1.07 vs 1.56 ms on 2^27 elements).  At 256/1024 the
       // Stockham kernel measured FASTER than the DIF (0.69/0.77 vs
       // 0.85/0.84) and keeps the job.
-      // experimental wave-local kernel for 256/512/1024 (one FFT per
-      // wave, no barriers on the data path) — opt-in until measured
-      if (t >= 8 && t <= 10 && std::getenv("SRTB_FFT_WAVE")) {
+      // wave-local kernel for 256/512/1024 (one FFT per wave, no
+      // barriers on the data path): measured 1.15-1.37x over the
+      // ping-pong Stockham (0.517/0.544/0.673 vs 0.698/0.746/0.771 ms on
+      // 2^27 elements, r02 check6).  SRTB_FFT_WAVE=0 reverts.
+      const char* we = std::getenv("SRTB_FFT_WAVE");
+      if (t >= 8 && t <= 10 && (!we || std::atoi(we) != 0)) {
         ensure_len_table((uint32_t)len, sign, stream);
         Pass p;
         p.kind = PassKind::kWave;

@@ -966,10 +966,19 @@ __global__ void __launch_bounds__(256)
       if constexpr (E >= 16) col_stage<E, 16, 4, SIGN, 64>(v, ltw);
       if constexpr (E == 8) col_stage<E, 8, 2, SIGN, 64>(v, ltw);
     }
-    // step 2: twiddle W_N^(lane*k2)
+    // step 2: twiddle W_N^(lane*k2) by recurrence from W_N^lane — the
+    // direct gather ltw[lane*k2] is a stride-k2 LDS access (up to 8-way
+    // bank conflicts at even k2); one stride-1 gather + E-2 register
+    // cmuls is conflict-free (phase error ~E ulp, within fp32 FFT noise)
+    {
+      const float2 wbase = ltw[lane];
+      float2 w = wbase;
 #pragma unroll
-    for (int k2 = 1; k2 < E; ++k2)
-      v[k2] = cmulf(v[k2], ltw[lane * k2]);
+      for (int k2 = 1; k2 < E; ++k2) {
+        v[k2] = cmulf(v[k2], w);
+        w = cmulf(w, wbase);
+      }
+    }
     // step 3: 64-point DIF across lanes; output lane holds k1 = rev
 #pragma unroll
     for (int M = 32; M >= 1; M >>= 1) {
