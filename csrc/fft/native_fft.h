@@ -80,7 +80,10 @@ class NativeFft {
       // ping-pong Stockham (0.517/0.544/0.673 vs 0.698/0.746/0.771 ms on
       // 2^27 elements, r02 check6).  SRTB_FFT_WAVE=0 reverts.
       const char* we = std::getenv("SRTB_FFT_WAVE");
-      if (t >= 8 && t <= 10 && (!we || std::atoi(we) != 0)) {
+      // t=11 (E=32, 2048): opt-in via SRTB_FFT_WAVE=2 until measured
+      // (LDS 2048*8 + 4*64*33*8 = 84 KB -> 1 WG/CU risk)
+      const int wv = we ? std::atoi(we) : 1;
+      if (((t >= 8 && t <= 10 && wv != 0) || (t == 11 && wv >= 2))) {
         ensure_len_table((uint32_t)len, sign, stream);
         Pass p;
         p.kind = PassKind::kWave;
@@ -144,11 +147,14 @@ class NativeFft {
       // accepted override
     } else {
       const char* mc = std::getenv("SRTB_FFT_MAXCOL");
-      const int maxcol_log2 = mc ? ilog2z(std::atoi(mc)) : 5;  // default 32
+      const int maxcol_log2 = mc ? ilog2z(std::atoi(mc)) : 6;  // default 64
       const char* fe = std::getenv("SRTB_FFT_FINAL");
-      int final_log2 = fe ? ilog2z(std::atoi(fe)) : 8;  // 256/1024/4096
+      // default final DIF length 64 (3 radix-4 stages, F=32 dif -> 17 KB
+      // LDS, 8 WG/CU): fwd 2^29 8.89 -> 8.28 ms, bwd 2^18x2048 5.11 ->
+      // 4.89 (r02 dif64 sweep).  t=13 keeps 256 (one fewer pass there).
+      int final_log2 = fe ? ilog2z(std::atoi(fe)) : (t >= 14 ? 6 : 8);
       if (final_log2 & 1) ++final_log2;                 // pure 4^t only
-      if (final_log2 < 8) final_log2 = 8;
+      if (final_log2 < 6) final_log2 = 6;
       if (final_log2 > 12) final_log2 = 12;
       if (final_log2 >= t) final_log2 = (t % 2) ? t - 1 : t - 2;
       const int rest = t - final_log2;
@@ -352,7 +358,7 @@ class NativeFft {
     // and bwd 2^18x2048 5.11/5.40/6.01/7.77 at F=16/8/32/4): 35 KB LDS ->
     // 4 workgroups/CU overlap the per-stage barriers that parked 38% of
     // wave cycles at F=32 (see profiles/r02_pmc_summary.md).
-    int F = 16;
+    int F = (p.dif.n <= 64) ? 32 : 16;
     if (const char* e = std::getenv("SRTB_FFT_DIF_F")) {
       const int v = std::atoi(e);
       if (v >= 1 && v <= 64 && (v & (v - 1)) == 0) F = v;

@@ -962,9 +962,10 @@ __global__ void __launch_bounds__(256)
       // free path: reuse ltw with TS scaling by calling the stages directly
       // (col_stage uses tw[j*TS]; our table is ltw with W_E^j at 64*j, so
       // scale TS by 64)
-      if constexpr (E >= 4) col_stage<E, 4, 4, SIGN, (E / 4) * 64>(v, ltw);
-      if constexpr (E >= 16) col_stage<E, 16, 4, SIGN, 64>(v, ltw);
-      if constexpr (E == 8) col_stage<E, 8, 2, SIGN, 64>(v, ltw);
+      if constexpr (E >= 4) col_stage<E, 4, 4, SIGN, 16 * E>(v, ltw);
+      if constexpr (E >= 16) col_stage<E, 16, 4, SIGN, 4 * E>(v, ltw);
+      if constexpr (E == 8) col_stage<E, 8, 2, SIGN, 8 * E>(v, ltw);
+      if constexpr (E == 32) col_stage<E, 32, 2, SIGN, 2 * E>(v, ltw);
     }
     // step 2: twiddle W_N^(lane*k2) by recurrence from W_N^lane — the
     // direct gather ltw[lane*k2] is a stride-k2 LDS access (up to 8-way
@@ -1555,7 +1556,8 @@ hipError_t fft_mid512_pass(const float2* in, float2* out,
 hipError_t fft_wave_pass(const float2* in, float2* out, uint32_t n,
                          size_t n_ffts, int sign, const float2* tw_n,
                          hipStream_t stream) {
-  if (n != 256 && n != 512 && n != 1024) return hipErrorInvalidValue;
+  if (n != 256 && n != 512 && n != 1024 && n != 2048)
+    return hipErrorInvalidValue;
   const int E = (int)(n / 64);
   const size_t lds_bytes = ((size_t)n + 64 + 4ull * 64 * (E + 1)) *
                            sizeof(float2);
@@ -1566,11 +1568,13 @@ hipError_t fft_wave_pass(const float2* in, float2* out, uint32_t n,
   if (sign < 0) {
     if (E == 4) WAVE_LAUNCH(-1, 4);
     else if (E == 8) WAVE_LAUNCH(-1, 8);
-    else WAVE_LAUNCH(-1, 16);
+    else if (E == 16) WAVE_LAUNCH(-1, 16);
+    else WAVE_LAUNCH(-1, 32);
   } else {
     if (E == 4) WAVE_LAUNCH(1, 4);
     else if (E == 8) WAVE_LAUNCH(1, 8);
-    else WAVE_LAUNCH(1, 16);
+    else if (E == 16) WAVE_LAUNCH(1, 16);
+    else WAVE_LAUNCH(1, 32);
   }
 #undef WAVE_LAUNCH
   SRTB_CHECK_LAUNCH();

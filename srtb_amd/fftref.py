@@ -254,23 +254,27 @@ def rfft_packed(x: np.ndarray, factors: list[int] | None = None) -> np.ndarray:
     return r2c_post(zf, -1)
 
 
-def plan_factors(t: int, maxcol_log2: int = 5, final_log2: int = 8) -> list[int]:
+def plan_factors(t: int, maxcol_log2: int = 6, final_log2: int | None = None) -> list[int]:
     """Mirror of the native planner's factorization policy
     (csrc/fft/native_fft.h plan()): the column factors + final DIF length
     chosen for a 2**t transform.  Kept as the executable SPEC of the
     policy — the C++ is the implementation of record.
 
     t <= 12 -> [2**t] (single LDS Stockham pass).
-    Else: final DIF length 256 (pure 4^k), and the residual bits go to
-    register-column factors: greedy 64s for rest >= 18 (the N=64 lane-pair
-    kernel is the fastest pass; measured fft_factor_sweep.py), balanced
-    <=32 columns otherwise (measured best at the 2^18 waterfall).
+    Else: final DIF length 64 (pure 4^k; 256 at t=13), and the residual
+    bits go to register-column factors: greedy 64s for rest >= 18 (the
+    N=64 lane-pair kernel is the fastest pass; fft_factor_sweep.py),
+    balanced <=64 columns otherwise.
     """
     if t <= 12:
         return [1 << t]
+    if final_log2 is None:
+        # default final DIF length 64 for t >= 14 (r02 dif64 sweep:
+        # shorter final pass at 8+ WG/CU wins); 256 at t=13 saves a pass
+        final_log2 = 6 if t >= 14 else 8
     if final_log2 % 2:
         final_log2 += 1
-    final_log2 = min(max(final_log2, 8), 12)
+    final_log2 = min(max(final_log2, 6), 12)
     if final_log2 >= t:
         final_log2 = t - 1 if t % 2 else t - 2
     rest = t - final_log2
