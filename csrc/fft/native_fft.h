@@ -80,10 +80,10 @@ class NativeFft {
       // ping-pong Stockham (0.517/0.544/0.673 vs 0.698/0.746/0.771 ms on
       // 2^27 elements, r02 check6).  SRTB_FFT_WAVE=0 reverts.
       const char* we = std::getenv("SRTB_FFT_WAVE");
-      // t=11 (E=32, 2048): opt-in via SRTB_FFT_WAVE=2 until measured
-      // (LDS 2048*8 + 4*64*33*8 = 84 KB -> 1 WG/CU risk)
+      // 256-2048: measured 0.515/0.528/0.598/1.291 ms vs Stockham
+      // 0.698/0.746/0.771/1.42 (r02 checks 6-8); SRTB_FFT_WAVE=0 reverts
       const int wv = we ? std::atoi(we) : 1;
-      if (((t >= 8 && t <= 10 && wv != 0) || (t == 11 && wv >= 2))) {
+      if (t >= 8 && t <= 11 && wv != 0) {
         ensure_len_table((uint32_t)len, sign, stream);
         Pass p;
         p.kind = PassKind::kWave;
