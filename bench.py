@@ -61,8 +61,11 @@ def main():
                     help="FFT backend (hand-written Stockham vs hipFFT)")
     ap.add_argument("--graph", action="store_true",
                     help="capture the per-block chain into hipGraphs")
-    ap.add_argument("--slots", type=int, default=4,
-                    help="double-buffered engine slots (streams)")
+    ap.add_argument("--slots", type=int, default=2,
+                    help="double-buffered engine slots (streams); 2 measured "
+                    "fastest (r02 A/B: 17.4/18.7/20.5/17.8 ms per block at "
+                    "2/3/4/6 slots — deeper concurrency degrades aggregate "
+                    "HBM efficiency)")
     ap.add_argument("--backend", choices=["nccl", "gloo"], default="nccl",
                     help="torch.distributed backend for world>1 (nccl=RCCL "
                     "over xGMI, one rank per GPU; gloo validates the "
