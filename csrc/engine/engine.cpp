@@ -92,6 +92,10 @@ PipelineEngine::PipelineEngine(const EngineConfig& cfg, int n_slots)
                             sizeof(float)),
               "scan alloc");
     check_hip(hipMalloc(&s.partials, np * sizeof(double)), "partials alloc");
+    if (n_boxcars_ > 0)
+      check_hip(hipMalloc(&s.box_partials,
+                          (size_t)n_boxcars_ * 2 * np * sizeof(double)),
+                "box partials alloc");
     check_hip(hipMalloc(&s.mean_power, sizeof(double)), "mean alloc");
     check_hip(hipMalloc(&s.sums, 2 * sizeof(double)), "sums alloc");
     const int ncnt = 1 + 1 + n_boxcars_;  // zero_count + raw + boxcars
@@ -169,6 +173,7 @@ PipelineEngine::~PipelineEngine() {
     (void)hipFree(s.box);
     (void)hipFree(s.scan_scratch);
     (void)hipFree(s.partials);
+    if (s.box_partials) (void)hipFree(s.box_partials);
     (void)hipFree(s.mean_power);
     (void)hipFree(s.sums);
     (void)hipFree(s.counters);
@@ -312,19 +317,16 @@ void PipelineEngine::enqueue_chain(Slot& s, const uint8_t* dev_raw,
   check_hip(count_above(s.ts, ts_count_, s.sums + 1, cfg_.snr_threshold,
                         s.counters + 1, s.thresholds + 0, st),
             "count raw");
-  // 10. boxcar ladder from the inclusive scan
+  // 10. boxcar ladder from the inclusive scan — fused: thresholds and
+  // counts for every length derive directly from the ~1 MB cumulative sum
+  // in 3 launches (was 3 per length; the box series never materialize)
   if (n_boxcars_ > 0) {
     check_hip(inclusive_scan(s.ts, s.cumsum, ts_count_, s.scan_scratch, st),
               "scan");
-    for (int b = 0; b < n_boxcars_; ++b) {
-      const size_t L = boxcar_lengths_[b];
-      const size_t n_out = ts_count_ - L;
-      check_hip(boxcar(s.cumsum, s.box, n_out, L, st), "boxcar");
-      check_hip(sum_sumsq(s.box, n_out, s.partials, s.sums, st), "box var");
-      check_hip(count_above(s.box, n_out, s.sums + 1, cfg_.snr_threshold,
-                            s.counters + 2 + b, s.thresholds + 1 + b, st),
-                "box count");
-    }
+    check_hip(boxcar_ladder(s.cumsum, ts_count_, boxcar_lengths_.data(),
+                            n_boxcars_, s.box_partials, cfg_.snr_threshold,
+                            s.thresholds + 1, s.counters + 2, st),
+              "box ladder");
   }
   // 11. result counters → pinned host
   check_hip(hipMemcpyAsync(s.h_counters, s.counters, ncnt * sizeof(unsigned),

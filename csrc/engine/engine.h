@@ -173,6 +173,7 @@ class PipelineEngine {
     float* box = nullptr;           // [ts_count]
     float* scan_scratch = nullptr;  // [4096]
     double* partials = nullptr;     // reduce partials + out scalars
+    double* box_partials = nullptr; // [n_boxcars][2][reduce_partials]
     double* mean_power = nullptr;   // -> inside partials block
     double* sums = nullptr;         // [2]
     unsigned* counters = nullptr;   // [1 + n_boxcars + 1] zero_count + counts

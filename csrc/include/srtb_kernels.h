@@ -166,6 +166,15 @@ hipError_t inclusive_scan(const float* in, float* out, size_t n,
 
 // box[i] = cumsum[i+L] - cumsum[i], i < n_out (reference
 // signal_detect_pipe.hpp:387-423).
+// fused boxcar ladder: thresholds + counts for up to 12 lengths in 3
+// launches, derived directly from the cumulative sum (no box arrays).
+// partials: device scratch of n_lengths * 2 * reduce_partials() doubles;
+// out_thr/out_counts: [n_lengths] each (counts must be zeroed beforehand).
+hipError_t boxcar_ladder(const float* cumsum, size_t ts_count,
+                         const size_t* lengths, int n_lengths,
+                         double* partials, float snr, float* out_thr,
+                         unsigned* out_counts, hipStream_t stream);
+
 hipError_t boxcar(const float* cumsum, float* out, size_t n_out, size_t L,
                   hipStream_t stream);
 

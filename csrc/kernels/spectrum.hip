@@ -495,6 +495,96 @@ __global__ void k_boxcar(const float* __restrict__ cumsum,
     out[i] = cumsum[i + L] - cumsum[i];
 }
 
+// ---- fused boxcar ladder (3 launches instead of 3 per length) ----
+// The per-length box series box_L[i] = cumsum[i+L] - cumsum[i] never
+// materializes: stats, thresholds and counts all derive from the (L2-
+// resident, ~1 MB) cumulative sum directly.
+
+struct BoxLadderDev {
+  int n;                        // number of ladder lengths (<= 12)
+  unsigned long long L[12];
+};
+
+__global__ void k_box_stats(const float* __restrict__ cumsum, size_t ts_count,
+                            BoxLadderDev lad, double* __restrict__ partials) {
+  double s[12], s2[12];
+#pragma unroll
+  for (int l = 0; l < 12; ++l) s[l] = s2[l] = 0.0;
+  const size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < ts_count; i += stride) {
+    const float c0 = cumsum[i];
+    for (int l = 0; l < lad.n; ++l) {
+      const unsigned long long L = lad.L[l];
+      if (i + L < ts_count) {
+        const float v = cumsum[i + L] - c0;
+        s[l] += v;
+        s2[l] += (double)v * v;
+      }
+    }
+  }
+  for (int l = 0; l < lad.n; ++l) {
+    const double bs = block_reduce_sum(s[l]);
+    const double bs2 = block_reduce_sum(s2[l]);
+    if (threadIdx.x == 0) {
+      partials[(size_t)(2 * l) * kReducePartials + blockIdx.x] = bs;
+      partials[(size_t)(2 * l + 1) * kReducePartials + blockIdx.x] = bs2;
+    }
+  }
+}
+
+__global__ void k_box_finish(const double* __restrict__ partials,
+                             size_t ts_count, BoxLadderDev lad, float snr,
+                             float* __restrict__ out_thr) {
+  for (int l = 0; l < lad.n; ++l) {
+    double s2 = 0.0;
+    for (int i = threadIdx.x; i < kReducePartials; i += blockDim.x)
+      s2 += partials[(size_t)(2 * l + 1) * kReducePartials + i];
+    const double bs2 = block_reduce_sum(s2);
+    if (threadIdx.x == 0) {
+      const double n_out = (double)(ts_count - lad.L[l]);
+      out_thr[l] = snr * (float)sqrt(bs2 / n_out);
+    }
+  }
+}
+
+__global__ void k_box_count(const float* __restrict__ cumsum, size_t ts_count,
+                            BoxLadderDev lad,
+                            const float* __restrict__ thr,
+                            unsigned* __restrict__ out_counts) {
+  float t[12];
+  for (int l = 0; l < lad.n; ++l) t[l] = thr[l];
+  unsigned cnt[12];
+#pragma unroll
+  for (int l = 0; l < 12; ++l) cnt[l] = 0;
+  const size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < ts_count; i += stride) {
+    const float c0 = cumsum[i];
+    for (int l = 0; l < lad.n; ++l) {
+      const unsigned long long L = lad.L[l];
+      if (i + L < ts_count) cnt[l] += ((cumsum[i + L] - c0) > t[l]) ? 1u : 0u;
+    }
+  }
+  __shared__ unsigned lds[kBlock / 64];
+  for (int l = 0; l < lad.n; ++l) {
+    unsigned local = cnt[l];
+    const int lane = threadIdx.x & 63;
+    const int wave = threadIdx.x >> 6;
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1)
+      local += __shfl_down(local, off, 64);
+    __syncthreads();
+    if (lane == 0) lds[wave] = local;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      unsigned b = 0;
+      for (int w = 0; w < (int)(blockDim.x >> 6); ++w) b += lds[w];
+      if (b) atomicAdd(&out_counts[l], b);
+    }
+  }
+}
+
 }  // namespace
 
 // ---------------- host wrappers ----------------
@@ -747,6 +837,26 @@ hipError_t boxcar(const float* cumsum, float* out, size_t n_out, size_t L,
                   hipStream_t stream) {
   hipLaunchKernelGGL(k_boxcar, grid_for(n_out), dim3(kBlock), 0, stream,
                      cumsum, out, n_out, L);
+  SRTB_CHECK_LAUNCH();
+  return hipSuccess;
+}
+
+hipError_t boxcar_ladder(const float* cumsum, size_t ts_count,
+                         const size_t* lengths, int n_lengths,
+                         double* partials, float snr, float* out_thr,
+                         unsigned* out_counts, hipStream_t stream) {
+  if (n_lengths <= 0) return hipSuccess;
+  if (n_lengths > 12) return hipErrorInvalidValue;
+  BoxLadderDev lad{};
+  lad.n = n_lengths;
+  for (int l = 0; l < n_lengths; ++l)
+    lad.L[l] = (unsigned long long)lengths[l];
+  hipLaunchKernelGGL(k_box_stats, dim3(kReducePartials), dim3(kBlock), 0,
+                     stream, cumsum, ts_count, lad, partials);
+  hipLaunchKernelGGL(k_box_finish, dim3(1), dim3(kBlock), 0, stream,
+                     partials, ts_count, lad, snr, out_thr);
+  hipLaunchKernelGGL(k_box_count, grid_for(ts_count), dim3(kBlock), 0,
+                     stream, cumsum, ts_count, lad, out_thr, out_counts);
   SRTB_CHECK_LAUNCH();
   return hipSuccess;
 }
