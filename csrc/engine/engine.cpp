@@ -117,7 +117,11 @@ PipelineEngine::PipelineEngine(const EngineConfig& cfg, int n_slots)
     if (native_fft_) {
       s.nfwd.plan(nc_, 1, -1, s.stream);
       if (native_bwd_) {
-        s.nbwd.plan(l_, s_, +1, s.stream);
+        // the RFI+dedispersion preop fuses into the backward's FIRST pass:
+        // plan it 32-max-column + final-256 so the fp64 phase runs on the
+        // 6-wave/SIMD pair32 kernel (see NativeFft::plan comment)
+        s.nbwd.plan(l_, s_, +1, s.stream, /*maxcol_log2=*/5,
+                    /*final_log2=*/8);
         const int wpr = s.nbwd.dif_sk_wgs_per_row();
         if (cfg.enable_sk && wpr > 0)
           check_hip(hipMalloc(&s.sk_dif_partials,

@@ -53,7 +53,14 @@ class NativeFft {
   ~NativeFft() { destroy(); }
 
   // Plan a batched C2C: `batch` rows of length `len`, contiguous.
-  void plan(size_t len, size_t batch, int sign, hipStream_t stream) {
+  // maxcol_log2_ovr / final_log2_ovr (0 = policy default) override the
+  // factorization defaults for THIS plan — the engine plans its backward
+  // (preop-fused) FFT with 32-max columns + final 256: the fp64
+  // dedispersion fused into the first pass wants the 6-wave/SIMD pair32
+  // kernel, not the 4-wave pair64 (r02 bench profile: 8.96 vs 4.56
+  // ms/block contended).  Explicit SRTB_FFT_* env still wins.
+  void plan(size_t len, size_t batch, int sign, hipStream_t stream,
+            int maxcol_log2_ovr = 0, int final_log2_ovr = 0) {
     destroy();
     if (!supported(len)) throw std::runtime_error("NativeFft: unsupported len");
     len_ = len;
@@ -147,12 +154,16 @@ class NativeFft {
       // accepted override
     } else {
       const char* mc = std::getenv("SRTB_FFT_MAXCOL");
-      const int maxcol_log2 = mc ? ilog2z(std::atoi(mc)) : 6;  // default 64
+      const int maxcol_log2 =
+          mc ? ilog2z(std::atoi(mc))
+             : (maxcol_log2_ovr ? maxcol_log2_ovr : 6);  // default 64
       const char* fe = std::getenv("SRTB_FFT_FINAL");
       // default final DIF length 64 (3 radix-4 stages, F=32 dif -> 17 KB
       // LDS, 8 WG/CU): fwd 2^29 8.89 -> 8.28 ms, bwd 2^18x2048 5.11 ->
       // 4.89 (r02 dif64 sweep).  t=13 keeps 256 (one fewer pass there).
-      int final_log2 = fe ? ilog2z(std::atoi(fe)) : (t >= 14 ? 6 : 8);
+      int final_log2 =
+          fe ? ilog2z(std::atoi(fe))
+             : (final_log2_ovr ? final_log2_ovr : (t >= 14 ? 6 : 8));
       if (final_log2 & 1) ++final_log2;                 // pure 4^t only
       if (final_log2 < 6) final_log2 = 6;
       if (final_log2 > 12) final_log2 = 12;
