@@ -117,11 +117,17 @@ PipelineEngine::PipelineEngine(const EngineConfig& cfg, int n_slots)
     if (native_fft_) {
       s.nfwd.plan(nc_, 1, -1, s.stream);
       if (native_bwd_) {
-        // the RFI+dedispersion preop fuses into the backward's FIRST pass:
-        // plan it 32-max-column + final-256 so the fp64 phase runs on the
-        // 6-wave/SIMD pair32 kernel (see NativeFft::plan comment)
-        s.nbwd.plan(l_, s_, +1, s.stream, /*maxcol_log2=*/5,
-                    /*final_log2=*/8);
+        // the RFI+dedispersion preop fuses into the backward's FIRST
+        // pass; SRTB_FFT_BWD32=1 plans it 32-max-column + final-256 (the
+        // pair32 kernel runs the fp64 phase at 6 waves/SIMD).  A/B on one
+        // box decides the default — kernel-sum favors pair32 (28.1 vs
+        // 28.7 ms/block) but wall favored the wide plan on another box.
+        const char* b32 = std::getenv("SRTB_FFT_BWD32");
+        if (b32 && std::atoi(b32) != 0)
+          s.nbwd.plan(l_, s_, +1, s.stream, /*maxcol_log2=*/5,
+                      /*final_log2=*/8);
+        else
+          s.nbwd.plan(l_, s_, +1, s.stream);
         const int wpr = s.nbwd.dif_sk_wgs_per_row();
         if (cfg.enable_sk && wpr > 0)
           check_hip(hipMalloc(&s.sk_dif_partials,
