@@ -43,8 +43,15 @@ def main():
     t0 = time.perf_counter()
     for c in range(pkts):
         s.sendto(struct.pack("<Q", c) + body, ("127.0.0.1", args.port))
-    recv.wait(timeout=120)
     elapsed = time.perf_counter() - t0
+    # an unpaced burst can drop the tail on loopback; keep nudging with
+    # fresh counters (they only close out blocks) until the receiver exits
+    c = pkts
+    while recv.poll() is None and c < pkts + 200000:
+        s.sendto(struct.pack("<Q", c) + body, ("127.0.0.1", args.port))
+        c += 1
+        time.sleep(0.0005)
+    recv.wait(timeout=30)
     sent_bytes = pkts * payload
     print(json.dumps({
         "metric": "UDP ingest (loopback, recvmmsg)",
