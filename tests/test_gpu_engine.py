@@ -210,6 +210,30 @@ def test_engine_fused_window(C):
     np.testing.assert_allclose(ts_gpu / scale, ts_cpu / scale, atol=2e-3)
 
 
+def test_engine_boxcar_ladder_matches_numpy(C):
+    """The fused boxcar ladder's per-length thresholds and counts must match
+    a NumPy recomputation from the engine's own time series."""
+    cfg = small_cfg()
+    rng = np.random.default_rng(3)
+    raw = np.clip(np.round(rng.normal(0, 16, cfg.baseband_input_count)),
+                  -128, 127).astype(np.int8).view(np.uint8)
+    eng = make_engine(C, cfg, snr_threshold=3.0)  # low snr → nonzero counts
+    slot = eng.submit(torch.from_numpy(raw.copy()))
+    res = eng.wait(slot)
+    ts = eng.time_series(slot).cpu().numpy().astype(np.float64)
+    cum = np.cumsum(ts)
+    snr = 3.0
+    assert len(res["counts"]) >= 3
+    for (L, cnt), thr in zip(res["counts"][1:], res["thresholds"][1:]):
+        box = cum[L:] - cum[:-L]
+        thr_exp = snr * np.sqrt(np.mean(box * box))
+        assert abs(thr - thr_exp) / thr_exp < 1e-3, (L, thr, thr_exp)
+        cnt_exp = int((box > thr_exp).sum())
+        # fp32 cumsum vs fp64 recompute can flip borderline samples
+        assert abs(cnt - cnt_exp) <= max(3, 0.02 * max(cnt_exp, 1)), \
+            (L, cnt, cnt_exp)
+
+
 def test_engine_watfft_window_deapply(C):
     """K21: with a non-rectangle window the waterfall is divided by the
     length-L window after the backward FFT (reference fft_pipe.hpp:350-358).
