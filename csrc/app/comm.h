@@ -13,7 +13,10 @@
 
 #include <hip/hip_runtime.h>
 #include <rccl/rccl.h>
+#include <sys/stat.h>
 #include <unistd.h>
+
+#include <ctime>
 
 #include <cstdint>
 #include <cstdio>
@@ -64,13 +67,18 @@ class Comm {
       if (std::rename(tmp.c_str(), path.c_str()) != 0)
         throw std::runtime_error("cannot publish " + path);
     } else {
-      // poll for rank 0's id (bounded)
+      // poll for rank 0's id (bounded); a file older than 120 s is a
+      // leftover from a crashed earlier run and must not be joined
       for (int tries = 0;; ++tries) {
-        FILE* f = std::fopen(path.c_str(), "rb");
-        if (f) {
-          const size_t got = std::fread(&id, 1, sizeof(id), f);
-          std::fclose(f);
-          if (got == sizeof(id)) break;
+        struct stat st {};
+        if (::stat(path.c_str(), &st) == 0 &&
+            std::time(nullptr) - st.st_mtime < 120) {
+          FILE* f = std::fopen(path.c_str(), "rb");
+          if (f) {
+            const size_t got = std::fread(&id, 1, sizeof(id), f);
+            std::fclose(f);
+            if (got == sizeof(id)) break;
+          }
         }
         if (tries > 3000)
           throw std::runtime_error("timed out waiting for " + path);

@@ -46,8 +46,13 @@ PipelineEngine::PipelineEngine(const EngineConfig& cfg, int n_slots)
   const int bits = std::abs(cfg.baseband_input_bits);
   raw_bytes_ = n_ * (size_t)bits / 8;
 
-  // boxcar ladder: 2, 4, ..., while <= max && < ts_count
-  for (size_t L = 2; L <= cfg.max_boxcar_length && L < ts_count_; L *= 2)
+  // boxcar ladder: 2, 4, ..., while <= max && < ts_count (the fused
+  // ladder kernel handles up to 12 lengths = max boxcar 4096; exotic
+  // configs beyond that are clamped rather than rejected)
+  for (size_t L = 2;
+       L <= cfg.max_boxcar_length && L < ts_count_ &&
+       boxcar_lengths_.size() < 12;
+       L *= 2)
     boxcar_lengths_.push_back(L);
   n_boxcars_ = (int)boxcar_lengths_.size();
 
