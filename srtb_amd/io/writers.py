@@ -106,6 +106,9 @@ class SignalWriteScheduler:
 
     def _write(self, blk: BlockProducts) -> None:
         if self._pool is not None:
+            # prune completed futures so day-long runs don't accumulate them
+            if len(self._futures) > 64:
+                self._futures = [f for f in self._futures if not f.done()]
             self._futures.append(self._pool.submit(self._write_sync, blk))
             return
         self._write_sync(blk)
