@@ -105,8 +105,14 @@ class DmTrialSweep:
         torch = self.torch
         raw_t = torch.from_numpy(np.ascontiguousarray(raw)).cuda()
         trials = []
-        for dm in dms:
-            slot = self.eng.submit(raw_t, dm_override=float(dm))
+        # keep both engine slots in flight: trial i+1 uploads/computes while
+        # trial i drains (the serial submit→wait loop left the GPU idle
+        # during each host-side readback)
+        inflight: list[tuple[float, int]] = []
+        n_slots = self.eng.n_slots
+
+        def drain_one():
+            dm, slot = inflight.pop(0)
             res = self.eng.wait(slot)
             ts = self.eng.time_series(slot)
             std = float(ts.std())
@@ -114,6 +120,14 @@ class DmTrialSweep:
             trials.append(DmTrial(dm=float(dm), counts=res["counts"],
                                   zero_count=res["zero_count"],
                                   peak_snr=peak))
+
+        for dm in dms:
+            inflight.append((float(dm),
+                             self.eng.submit(raw_t, dm_override=float(dm))))
+            if len(inflight) >= n_slots:
+                drain_one()
+        while inflight:
+            drain_one()
         return trials
 
     def best(self, trials: list[DmTrial]) -> DmTrial:
