@@ -127,6 +127,44 @@ def test_native_baseband_receiver_loopback(tmp_path):
     assert (data[3 * payload:4 * payload] == 4).all()
 
 
+@needs_bins
+def test_native_baseband_receiver_simple_headerless(tmp_path):
+    """baseband_format_type=simple is a headerless sequential stream
+    (reference backend_registry.hpp:36-39): bytes append in arrival order,
+    no counter header is consumed."""
+    port = 29903
+    block_bytes = 4096 * 2
+
+    def sender():
+        s = socket.socket(socket.AF_INET, socket.SOCK_DGRAM)
+        time.sleep(0.5)
+        # arbitrary-size datagrams summing to 2 blocks + margin
+        data = bytes(range(256)) * 70  # 17920 bytes
+        for off in range(0, len(data), 1500):
+            s.sendto(data[off:off + 1500], ("127.0.0.1", port))
+            time.sleep(0.005)
+        s.close()
+
+    t = threading.Thread(target=sender)
+    t.start()
+    out = subprocess.run(
+        [RECEIVER, "--max-blocks", "2",
+         "--baseband_format_type", "simple",
+         "--baseband_input_count", str(block_bytes),
+         "--baseband_input_bits", "8",
+         "--udp_receiver_address", "127.0.0.1",
+         "--udp_receiver_port", str(port),
+         "--baseband_output_file_prefix", str(tmp_path) + "/simp_"],
+        capture_output=True, text=True, timeout=30)
+    t.join()
+    assert out.returncode == 0, out.stderr
+    data = np.fromfile(tmp_path / "simp_recording.bin", dtype=np.uint8)
+    assert data.size == 2 * block_bytes
+    expect = np.frombuffer((bytes(range(256)) * 70)[:2 * block_bytes],
+                           dtype=np.uint8)
+    np.testing.assert_array_equal(data, expect)
+
+
 # ---------------- GPU end-to-end runs of the native executables ----------------
 
 @pytest.mark.gpu
