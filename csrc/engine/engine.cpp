@@ -119,6 +119,7 @@ PipelineEngine::PipelineEngine(const EngineConfig& cfg, int n_slots)
     native_fft_ = (be == 0 || be == 2) && NativeFft::supported(nc_);
     native_bwd_ = native_fft_ && NativeFft::supported(l_) &&
                   (be == 0 || l_ >= (1ull << 17));
+    fuse_r2c_ = false;
     if (native_fft_) {
       s.nfwd.plan(nc_, 1, -1, s.stream);
       if (native_bwd_) {
@@ -142,6 +143,23 @@ PipelineEngine::PipelineEngine(const EngineConfig& cfg, int n_slots)
         s.plans.create_c2c_only(l_, s_, s.stream);
       }
       check_hip(hipStreamSynchronize(s.stream), "fft table sync");
+      // r2c-into-backward fusion: the pair-combine runs at the backward
+      // first pass's load and the RFI mean comes from Parseval on the
+      // packed spectrum (fwd-DIF power partials) — the standalone 8.6 GB
+      // r2c pass disappears.  SRTB_FUSE_R2C=0 reverts.
+      {
+        const char* fe = std::getenv("SRTB_FUSE_R2C");
+        const bool want = !fe || std::atoi(fe) != 0;
+        const int fwgs = s.nfwd.dif_sk_wgs_per_row();
+        fuse_r2c_ = want && native_bwd_ && s.nbwd.first_pass_fusable() &&
+                    fwgs > 0;
+        if (fuse_r2c_) {
+          fwd_pw_n_ = (size_t)fwgs;  // batch = 1
+          check_hip(hipMalloc(&s.xbuf, nc_ * sizeof(float2)), "xbuf alloc");
+          check_hip(hipMalloc(&s.fwd_pw, fwd_pw_n_ * sizeof(float2)),
+                    "fwd pw alloc");
+        }
+      }
     } else {
       s.plans.create(n_, l_, s_, s.stream);
     }
@@ -181,6 +199,8 @@ PipelineEngine::~PipelineEngine() {
     (void)hipFree(s.spec);
     (void)hipFree(s.s2s4);
     if (s.sk_dif_partials) (void)hipFree(s.sk_dif_partials);
+    if (s.xbuf) (void)hipFree(s.xbuf);
+    if (s.fwd_pw) (void)hipFree(s.fwd_pw);
     (void)hipFree(s.flags);
     (void)hipFree(s.ts);
     (void)hipFree(s.ts_partial);
@@ -238,7 +258,20 @@ void PipelineEngine::enqueue_chain(Slot& s, const uint8_t* dev_raw,
   } else if (!dev_raw) {
     fft_in = dev_samples;
   }
-  if (native_fft_) {
+  const bool fr2c = fuse_r2c_;  // r2c pair-combine fused into bwd load
+  if (native_fft_ && fr2c) {
+    // 2. forward C2C of the packed-real view; the DIF store accumulates
+    //    per-WG sum|Z|^2 and the RFI mean comes from Parseval — the
+    //    standalone r2c pass (8.6 GB) is gone, its pair-combine runs at
+    //    the backward first pass's load below.
+    s.nfwd.exec(reinterpret_cast<float2*>(s.samples), s.spec, st,
+                nullptr, cfg_.enable_rfi_s1 ? s.fwd_pw : nullptr,
+                fuse_unpack ? dev_raw : nullptr, in_bits);
+    if (cfg_.enable_rfi_s1)
+      check_hip(r2c_mean_from_power(s.fwd_pw, fwd_pw_n_, s.spec, nc_,
+                                    s.mean_power, st),
+                "r2c mean");
+  } else if (native_fft_) {
     // 2. forward C2C of the packed-real view + r2c post-process with FUSED
     //    mean-|X|^2 (saves the separate 4 GB mean_power pass)
     s.nfwd.exec(reinterpret_cast<float2*>(s.samples), s.spec, st,
@@ -274,8 +307,10 @@ void PipelineEngine::enqueue_chain(Slot& s, const uint8_t* dev_raw,
     pre.df = df_;
     pre.dm = dm;
     pre.table = table;
+    pre.r2c_m = fr2c ? nc_ : 0;
     s.nbwd.exec(s.spec, reinterpret_cast<float2*>(s.samples), st, &pre,
-                watfft_window_ ? nullptr : s.sk_dif_partials);
+                watfft_window_ ? nullptr : s.sk_dif_partials, nullptr, 2,
+                fr2c ? s.xbuf : nullptr);
     wf = reinterpret_cast<float2*>(s.samples);
   } else {
     check_hip(rfi_dedisperse_fused(

@@ -153,6 +153,8 @@ class PipelineEngine {
   bool native_fft_ = false;        // hand-written FORWARD FFT active
   bool fused_unpack_off_ = false;  // SRTB_NO_FUSED_UNPACK kill switch
   bool native_bwd_ = false;        // hand-written BACKWARD (waterfall) FFT
+  bool fuse_r2c_ = false;          // r2c pair-combine fused into bwd load
+  size_t fwd_pw_n_ = 0;            // fwd power partial count
 
 
   struct Slot {
@@ -164,6 +166,8 @@ class PipelineEngine {
     uint8_t* raw = nullptr;         // device raw bytes
     float* samples = nullptr;       // [N] unpacked
     float2* spec = nullptr;         // [Nc+1] spectrum / waterfall (in-place)
+    float2* xbuf = nullptr;         // [Nc] bwd working set (r2c fusion)
+    float2* fwd_pw = nullptr;       // fwd-DIF per-WG power partials
     float2* s2s4 = nullptr;         // [S]
     float2* sk_dif_partials = nullptr;  // [S * wgs_per_row] (fused SK stats)
     uint8_t* flags = nullptr;       // [S]

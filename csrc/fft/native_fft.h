@@ -272,13 +272,20 @@ class NativeFft {
   // Execute the planned transform.  out may equal in only for 1-pass plans.
   // preop (optional) is applied to the FIRST pass's loads (requires
   // first_pass_fusable()).
+  // first_pass_out (optional): write the FIRST column pass out-of-place
+  // into this buffer (later passes continue in place there).  REQUIRED
+  // when preop->r2c_m is set: the pair-combine reads element m-k of the
+  // input, which an in-place pass would be overwriting concurrently.
   void exec(const float2* in, float2* out, hipStream_t stream,
             const FftPreop* preop = nullptr,
             float2* dif_sk_partials = nullptr,
-            const uint8_t* decode2_raw = nullptr, int decode_bits = 2) {
+            const uint8_t* decode2_raw = nullptr, int decode_bits = 2,
+            float2* first_pass_out = nullptr) {
     if (passes_.empty()) throw std::runtime_error("NativeFft: not planned");
-    if (passes_.size() > 1 && in == out)
+    if (passes_.size() > 1 && in == out && !first_pass_out)
       throw std::runtime_error("NativeFft: multi-pass needs out != in");
+    if (preop && preop->r2c_m && !first_pass_out)
+      throw std::runtime_error("NativeFft: r2c fusion needs first_pass_out");
     if (preop && !first_pass_fusable())
       throw std::runtime_error("NativeFft: preop needs a column first pass");
     if (decode2_raw && (preop || !first_pass_fusable()))
@@ -298,13 +305,15 @@ class NativeFft {
                     "fft_stockham_pass");
           break;
         }
-        case PassKind::kCol:
-          check_hip(fft_col_pass(cur, cur, p.d, p.n_ffts, sign_, p.tw_n,
+        case PassKind::kCol: {
+          float2* o = (i == 0 && first_pass_out) ? first_pass_out : cur;
+          check_hip(fft_col_pass(cur, o, p.d, p.n_ffts, sign_, p.tw_n,
                                  p.tw_hi, p.tw_lo, stream, pre,
                                  i == 0 ? decode2_raw : nullptr, decode_bits),
                     "fft_col_pass");
-          dst = cur;  // in place
+          dst = o;  // in place except an out-of-place first pass
           break;
+        }
         case PassKind::kMid:
           check_hip(fft_mid512_pass(cur, cur, p.d, p.n_ffts, sign_, p.tw_n,
                                     stream),

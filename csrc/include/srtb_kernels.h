@@ -222,7 +222,20 @@ struct FftPreop {
   // table (one extra coalesced float2 load) instead of the fp64 phase
   // computation, which would otherwise double the pass's VALU cost
   const float2* table = nullptr;
+  // != 0: the pass input is the PACKED forward spectrum Z (length r2c_m);
+  // the r2c pair-combine X[k] = E + w_k O runs at load (eliminates the
+  // standalone r2c pass; needs an out-of-place first pass)
+  unsigned long long r2c_m = 0;
 };
+
+// mean |X|^2 over the r2c spectrum from the PACKED spectrum Z via
+// Parseval: sum_{k<m}|X_k|^2 = sum|Z|^2 + (X0^2 - XM^2)/2 with
+// X0 = Re Z0 + Im Z0, XM = Re Z0 - Im Z0.  power_partials: per-workgroup
+// <sum|Z|^2, _> pairs accumulated by the forward DIF store (the SK-fusion
+// machinery), z0 = device pointer to Z[0].
+hipError_t r2c_mean_from_power(const float2* power_partials,
+                               size_t n_partials, const float2* z0, size_t m,
+                               double* out_mean, hipStream_t stream);
 
 // Register-resident column FFT pass (N in {2,4,8,16,32,64}; one FFT per
 // thread fully in VGPRs; in-place: out must alias layout of in addressing;

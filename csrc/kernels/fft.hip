@@ -438,7 +438,31 @@ struct FftPreopDev {
   ZapRange zap[16];
   double f_min, f_c, df, dm;
   const float2* table;  // cached dedispersion factors (null = compute fp64)
+  // != 0: the pass input is the PACKED forward spectrum Z (length r2c_m);
+  // the r2c pair-combine X[k] = E + w_k O runs at load (k_r2c_post math),
+  // eliminating the standalone r2c pass.  Requires an out-of-place first
+  // pass: element k also reads Z[r2c_m - k].
+  unsigned long long r2c_m;
 };
+
+// r2c pair-combine at load (identical math/precision to k_r2c_post)
+__device__ inline float2 r2c_combine_load(const float2* __restrict__ zbase,
+                                          float2 zk, unsigned long long k,
+                                          unsigned long long m) {
+  if (k == 0) return make_float2(zk.x + zk.y, 0.0f);
+  const float2 zm = zbase[m - k];
+  const float2 zmc = make_float2(zm.x, -zm.y);
+  const float2 even = make_float2(0.5f * (zk.x + zmc.x),
+                                  0.5f * (zk.y + zmc.y));
+  const float2 dif = make_float2(zk.x - zmc.x, zk.y - zmc.y);
+  const float2 odd = make_float2(0.5f * dif.y, -0.5f * dif.x);
+  const float phiw = (float)(-M_PI * (double)k / (double)m);
+  float sw, cw;
+  __sincosf(phiw, &sw, &cw);
+  const float2 wo = make_float2(cw * odd.x - sw * odd.y,
+                                cw * odd.y + sw * odd.x);
+  return make_float2(even.x + wo.x, even.y + wo.y);
+}
 
 template <int N, bool TWIDDLE, int SIGN, bool PREOP, int DEC = 0>
 __global__ void __launch_bounds__(256)
@@ -474,6 +498,7 @@ __global__ void __launch_bounds__(256)
     if constexpr (PREOP) {
       // fused rfi_dedisperse (spectrum.hip k_rfi_dedisp_fused semantics);
       // `flat` IS the spectrum bin index for in-place column passes
+      if (pre.r2c_m) x = r2c_combine_load(in, x, flat, pre.r2c_m);
       bool zap = pre.mean_power && (norm2(x) > thr_mean);
       for (int z = 0; z < pre.n_zap; ++z)
         zap |= (flat >= pre.zap[z].lo) & (flat <= pre.zap[z].hi);
@@ -563,6 +588,7 @@ __global__ void __launch_bounds__(256)
     else x = ld_stream(colbase + off, d.tuning);
     if constexpr (PREOP) {
       const unsigned long long flat = base + off;
+      if (pre.r2c_m) x = r2c_combine_load(in, x, flat, pre.r2c_m);
       bool zap = pre.mean_power && (norm2(x) > thr_mean);
       for (int z = 0; z < pre.n_zap; ++z)
         zap |= (flat >= pre.zap[z].lo) & (flat <= pre.zap[z].hi);
@@ -668,6 +694,7 @@ __global__ void __launch_bounds__(256)
     else x = ld_stream(colbase + off, d.tuning);
     if constexpr (PREOP) {
       const unsigned long long flat = base + off;
+      if (pre.r2c_m) x = r2c_combine_load(in, x, flat, pre.r2c_m);
       bool zap = pre.mean_power && (norm2(x) > thr_mean);
       for (int z = 0; z < pre.n_zap; ++z)
         zap |= (flat >= pre.zap[z].lo) & (flat <= pre.zap[z].hi);
@@ -1323,6 +1350,7 @@ hipError_t fft_col_pass(const float2* in, float2* out, const FftPassDesc& hd,
     pre.df = preop->df;
     pre.dm = preop->dm;
     pre.table = preop->table;
+    pre.r2c_m = preop->r2c_m;
   }
   FftPassDescDev d;
   d.n = hd.n;
@@ -1577,6 +1605,33 @@ hipError_t fft_wave_pass(const float2* in, float2* out, uint32_t n,
     else WAVE_LAUNCH(1, 32);
   }
 #undef WAVE_LAUNCH
+  SRTB_CHECK_LAUNCH();
+  return hipSuccess;
+}
+
+namespace {
+__global__ void k_r2c_mean_from_power(const float2* __restrict__ partials,
+                                      size_t n_partials,
+                                      const float2* __restrict__ z0, size_t m,
+                                      double* __restrict__ out_mean) {
+  double acc = 0.0;
+  for (size_t i = threadIdx.x; i < n_partials; i += blockDim.x)
+    acc += (double)partials[i].x;
+  const double p = block_reduce_sum(acc);
+  if (threadIdx.x == 0) {
+    const float2 z = *z0;
+    const double x0 = (double)z.x + z.y;
+    const double xm = (double)z.x - z.y;
+    *out_mean = (p + 0.5 * (x0 * x0 - xm * xm)) / (double)m;
+  }
+}
+}  // namespace
+
+hipError_t r2c_mean_from_power(const float2* power_partials,
+                               size_t n_partials, const float2* z0, size_t m,
+                               double* out_mean, hipStream_t stream) {
+  hipLaunchKernelGGL(k_r2c_mean_from_power, dim3(1), dim3(256), 0, stream,
+                     power_partials, n_partials, z0, m, out_mean);
   SRTB_CHECK_LAUNCH();
   return hipSuccess;
 }
