@@ -143,13 +143,17 @@ PipelineEngine::PipelineEngine(const EngineConfig& cfg, int n_slots)
         s.plans.create_c2c_only(l_, s_, s.stream);
       }
       check_hip(hipStreamSynchronize(s.stream), "fft table sync");
-      // r2c-into-backward fusion: the pair-combine runs at the backward
-      // first pass's load and the RFI mean comes from Parseval on the
-      // packed spectrum (fwd-DIF power partials) — the standalone 8.6 GB
-      // r2c pass disappears.  SRTB_FUSE_R2C=0 reverts.
+      // r2c-into-backward fusion (opt-in SRTB_FUSE_R2C=1): the
+      // pair-combine runs at the backward first pass's load and the RFI
+      // mean comes from Parseval on the packed spectrum — the standalone
+      // 8.6 GB r2c pass disappears.  MEASURED SLOWER end-to-end (50.8 vs
+      // 61.5 Gsps, r02): the combine adds a second 4.3 GB read + sincos
+      // to the pass that is already the chain's critical kernel (fp64
+      // dedispersion), and the lost overlap outweighs the traffic saved.
+      // Kept opt-in; numerics are exact (57 GPU tests pass either way).
       {
         const char* fe = std::getenv("SRTB_FUSE_R2C");
-        const bool want = !fe || std::atoi(fe) != 0;
+        const bool want = fe && std::atoi(fe) != 0;
         const int fwgs = s.nfwd.dif_sk_wgs_per_row();
         fuse_r2c_ = want && native_bwd_ && s.nbwd.first_pass_fusable() &&
                     fwgs > 0;
