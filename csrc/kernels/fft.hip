@@ -948,24 +948,36 @@ __device__ inline float2 shfl_xor_f2(float2 x, int mask) {
   return make_float2(__shfl_xor(x.x, mask, 64), __shfl_xor(x.y, mask, 64));
 }
 
-template <int SIGN, int E>
+template <int SIGN, int E, bool GT = false>
 __global__ void __launch_bounds__(256)
     k_fft_wave(const float2* __restrict__ in, float2* __restrict__ out,
                unsigned long long n_ffts, const float2* __restrict__ tw_n) {
   constexpr int N = 64 * E;
   constexpr int EL2 = col_ilog2(E);
   extern __shared__ float2 lds[];
-  // layout: [ltw: N][w64: 64][per-wave transpose buffers: 4 * 64*(E+1)]
-  float2* ltw = lds;
-  float2* w64 = lds + N;
+  // layout: [ltw: N][w64: 64][per-wave transpose buffers: 4 * 64*(E+1)];
+  // GT = twiddles read from GLOBAL (L2-hot) — frees N+64 float2 of LDS
+  // for one more workgroup per CU
   const int wave = (int)(threadIdx.x >> 6);
   const int lane = (int)(threadIdx.x & 63);
-  float2* buf = w64 + 64 + (size_t)wave * 64 * (E + 1);
-
-  for (int j = (int)threadIdx.x; j < N; j += (int)blockDim.x)
-    ltw[j] = tw_n[j];
-  if (threadIdx.x < 64) w64[threadIdx.x] = tw_n[(size_t)threadIdx.x * E];
-  __syncthreads();  // tables only; the data path below is wave-local
+  const float2* ltw;
+  const float2* w64;
+  float2* buf;
+  if constexpr (GT) {
+    ltw = tw_n;
+    w64 = nullptr;
+    buf = lds + (size_t)wave * 64 * (E + 1);
+  } else {
+    float2* ltw_l = lds;
+    float2* w64_l = lds + N;
+    for (int j = (int)threadIdx.x; j < N; j += (int)blockDim.x)
+      ltw_l[j] = tw_n[j];
+    if (threadIdx.x < 64) w64_l[threadIdx.x] = tw_n[(size_t)threadIdx.x * E];
+    __syncthreads();  // tables only; the data path below is wave-local
+    ltw = ltw_l;
+    w64 = w64_l;
+    buf = w64_l + 64 + (size_t)wave * 64 * (E + 1);
+  }
 
   const unsigned rev = (__brev((unsigned)lane) >> 26);  // bitrev6(lane)
   const unsigned long long wave0 =
@@ -1012,7 +1024,8 @@ __global__ void __launch_bounds__(256)
     for (int M = 32; M >= 1; M >>= 1) {
       const int upper = lane & M;
       const int j = lane & (M - 1);
-      const float2 tw = w64[j * (32 / M)];
+      const float2 tw = GT ? tw_n[(size_t)(j * (32 / M)) * E]
+                           : w64[j * (32 / M)];
 #pragma unroll
       for (int k2 = 0; k2 < E; ++k2) {
         const float2 t = shfl_xor_f2(v[k2], M);
@@ -1587,23 +1600,32 @@ hipError_t fft_wave_pass(const float2* in, float2* out, uint32_t n,
   if (n != 256 && n != 512 && n != 1024 && n != 2048)
     return hipErrorInvalidValue;
   const int E = (int)(n / 64);
-  const size_t lds_bytes = ((size_t)n + 64 + 4ull * 64 * (E + 1)) *
-                           sizeof(float2);
+  // SRTB_FFT_WAVE_GT=1: twiddles from GLOBAL (L2-hot) instead of LDS —
+  // frees n+64 float2 of LDS for one more workgroup per CU (A/B probe)
+  static const bool gt = [] {
+    const char* e = std::getenv("SRTB_FFT_WAVE_GT");
+    return e && std::atoi(e) != 0;
+  }();
+  const size_t lds_bytes =
+      (gt ? 4ull * 64 * (E + 1)
+          : (size_t)n + 64 + 4ull * 64 * (E + 1)) * sizeof(float2);
   // enough waves to fill the chip ~8x; each wave strides over FFTs
   uint32_t grid = (uint32_t)((n_ffts + 3) / 4);
   if (grid > 16384) grid = 16384;
-#define WAVE_LAUNCH(SG, EE)                                                    hipLaunchKernelGGL((k_fft_wave<SG, EE>), dim3(grid), dim3(256),                                 lds_bytes, stream, in, out, n_ffts, tw_n)
+#define WAVE_LAUNCH(SG, EE, G) hipLaunchKernelGGL((k_fft_wave<SG, EE, G>), dim3(grid), dim3(256), lds_bytes, stream, in, out, n_ffts, tw_n)
+#define WAVE_DISPATCH(SG, EE) do { if (gt) WAVE_LAUNCH(SG, EE, true); else WAVE_LAUNCH(SG, EE, false); } while (0)
   if (sign < 0) {
-    if (E == 4) WAVE_LAUNCH(-1, 4);
-    else if (E == 8) WAVE_LAUNCH(-1, 8);
-    else if (E == 16) WAVE_LAUNCH(-1, 16);
-    else WAVE_LAUNCH(-1, 32);
+    if (E == 4) WAVE_DISPATCH(-1, 4);
+    else if (E == 8) WAVE_DISPATCH(-1, 8);
+    else if (E == 16) WAVE_DISPATCH(-1, 16);
+    else WAVE_DISPATCH(-1, 32);
   } else {
-    if (E == 4) WAVE_LAUNCH(1, 4);
-    else if (E == 8) WAVE_LAUNCH(1, 8);
-    else if (E == 16) WAVE_LAUNCH(1, 16);
-    else WAVE_LAUNCH(1, 32);
+    if (E == 4) WAVE_DISPATCH(1, 4);
+    else if (E == 8) WAVE_DISPATCH(1, 8);
+    else if (E == 16) WAVE_DISPATCH(1, 16);
+    else WAVE_DISPATCH(1, 32);
   }
+#undef WAVE_DISPATCH
 #undef WAVE_LAUNCH
   SRTB_CHECK_LAUNCH();
   return hipSuccess;
