@@ -33,15 +33,31 @@ _PAGE = """<!DOCTYPE html>
 <style>
  body {{ background:#101018; color:#d0d0e0; font-family:monospace; }}
  .wf {{ border:1px solid #445; margin:4px; image-rendering:pixelated; }}
+ .sp {{ border:1px solid #445; margin:4px; display:block; }}
  #status {{ margin:8px; white-space:pre; }}
 </style></head>
 <body>
-<h3>srtb_amd &mdash; live waterfall</h3>
+<h3>srtb_amd &mdash; live waterfall + spectrum</h3>
 <div id="imgs"></div>
 <div id="status"></div>
 <script>
 const REFRESH_MS = {refresh_ms};
 let streams = [];
+function drawSpectrum(canvas, values) {{
+  const ctx = canvas.getContext('2d');
+  const w = canvas.width, h = canvas.height;
+  ctx.fillStyle = '#101018'; ctx.fillRect(0, 0, w, h);
+  if (!values || !values.length) return;
+  let lo = Math.min(...values), hi = Math.max(...values);
+  if (hi <= lo) hi = lo + 1;
+  ctx.strokeStyle = '#6fc36f'; ctx.beginPath();
+  for (let i = 0; i < values.length; i++) {{
+    const x = i * (w - 1) / (values.length - 1 || 1);
+    const y = h - 1 - (values[i] - lo) / (hi - lo) * (h - 2);
+    if (i === 0) ctx.moveTo(x, y); else ctx.lineTo(x, y);
+  }}
+  ctx.stroke();
+}}
 async function poll() {{
   try {{
     const r = await fetch('/status.json'); const st = await r.json();
@@ -54,11 +70,22 @@ async function poll() {{
         const img = document.createElement('img');
         img.id = 'wf' + s; img.className = 'wf';
         div.appendChild(img);
+        const cv = document.createElement('canvas');
+        cv.id = 'sp' + s; cv.className = 'sp';
+        cv.width = 640; cv.height = 120;
+        div.appendChild(cv);
       }}
     }}
     for (const s of streams) {{
       document.getElementById('wf' + s).src =
         '/frame' + s + '.bmp?t=' + Date.now();
+      try {{
+        const sr = await fetch('/spectrum' + s + '.json');
+        if (sr.ok) {{
+          const sj = await sr.json();
+          drawSpectrum(document.getElementById('sp' + s), sj.values);
+        }}
+      }} catch (e) {{}}
     }}
   }} catch (e) {{ /* server gone */ }}
 }}
@@ -98,6 +125,15 @@ class _Handler(BaseHTTPRequestHandler):
         elif path == "/status.json":
             self._reply(200, "application/json",
                         json.dumps(srv.status()).encode())
+        elif path.startswith("/spectrum") and path.endswith(".json"):
+            try:
+                sid = int(path[len("/spectrum"):-len(".json")])
+            except ValueError:
+                return self._reply(404, "text/plain", b"bad stream")
+            spec = srv.get_spectrum_json(sid)
+            if spec is None:
+                return self._reply(404, "text/plain", b"no spectrum yet")
+            self._reply(200, "application/json", spec)
         elif path.startswith("/frame") and path.endswith(".bmp"):
             try:
                 sid = int(path[len("/frame"):-len(".bmp")])
@@ -135,6 +171,7 @@ class WaterfallServer:
         self.port = port
         self.refresh_ms = refresh_ms
         self._frames: dict[int, bytes] = {}
+        self._spectra: dict[int, bytes] = {}
         self._status: dict = {}
         self._lock = threading.Lock()
         self._httpd: ThreadingHTTPServer | None = None
@@ -166,6 +203,18 @@ class WaterfallServer:
         with self._lock:
             self._frames[stream_id] = bmp
 
+    def push_spectrum(self, stream_id: int, values) -> None:
+        """Publish the latest per-channel intensity line of one stream
+        (the reference's spectrum.qml view); downsampled to <= 1024 points."""
+        import numpy as np
+        v = np.asarray(values, dtype=np.float64).ravel()
+        if v.size > 1024:
+            n = (v.size // 1024) * 1024
+            v = v[:n].reshape(1024, -1).mean(axis=1)
+        body = json.dumps({"values": [round(float(x), 6) for x in v]})
+        with self._lock:
+            self._spectra[stream_id] = body.encode()
+
     def update_status(self, **kv) -> None:
         with self._lock:
             self._status.update(kv)
@@ -178,6 +227,10 @@ class WaterfallServer:
     def get_frame_bmp(self, stream_id: int) -> bytes | None:
         with self._lock:
             return self._frames.get(stream_id)
+
+    def get_spectrum_json(self, stream_id: int) -> bytes | None:
+        with self._lock:
+            return self._spectra.get(stream_id)
 
     def __enter__(self):
         return self.start()

@@ -117,6 +117,12 @@ class GpuStreamPipeline:
                                      ref.COLOR_OVERFLOW)
         return pix.cpu().numpy().view(np.uint32)
 
+    def spectrum_lines(self) -> list[np.ndarray]:
+        # per-channel mean intensity of the latest waterfall (the
+        # reference's live spectrum view, src/spectrum.qml)
+        wf = self.eng.waterfall(self.last_slot)
+        return [(wf.real ** 2 + wf.imag ** 2).mean(dim=1).cpu().numpy()]
+
 
 class GpuMultiPolPipeline:
     """One PACKED packet stream fanning out into per-polarization sample
@@ -179,6 +185,14 @@ class GpuMultiPolPipeline:
         return [self._frame(e, s, width, height)
                 for e, s in zip(self.engines, self.last_slots)]
 
+    def spectrum_lines(self) -> list[np.ndarray]:
+        out = []
+        for e, sl in zip(self.engines, self.last_slots):
+            wf = e.waterfall(sl)
+            out.append((wf.real ** 2 + wf.imag ** 2).mean(dim=1)
+                       .cpu().numpy())
+        return out
+
 
 class CpuMultiPolPipeline:
     """CPU/NumPy twin of GpuMultiPolPipeline (oracle + plumbing runs)."""
@@ -230,6 +244,10 @@ class CpuMultiPolPipeline:
             out.append(ref.generate_pixmap(img))
         return out
 
+    def spectrum_lines(self) -> list[np.ndarray]:
+        return [(np.abs(r["waterfall"].astype(np.complex64)) ** 2
+                 ).mean(axis=1) for r in self.last_results]
+
 
 class CpuStreamPipeline:
     """CPU/NumPy path (plumbing runs, no GPU)."""
@@ -257,6 +275,10 @@ class CpuStreamPipeline:
         img = ref.resample_power_2d(p, height, width)
         img = ref.normalize_by_mean(img)
         return ref.generate_pixmap(img)
+
+    def spectrum_lines(self) -> list[np.ndarray]:
+        wf = self.last_result["waterfall"]
+        return [(np.abs(wf.astype(np.complex64)) ** 2).mean(axis=1)]
 
 
 def main(argv=None) -> int:
@@ -343,6 +365,11 @@ def main(argv=None) -> int:
             return  # no block processed yet
         for sid, frame in enumerate(frames):
             gui.push_frame(sid, frame)
+        try:
+            for sid, line in enumerate(pipe.spectrum_lines()):
+                gui.push_spectrum(sid, line)
+        except Exception:
+            pass
         gui.update_status(blocks=blocks_done, written=written)
 
     agg = DetectionAggregator()

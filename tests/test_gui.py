@@ -50,6 +50,12 @@ def test_server_serves_page_status_and_frames():
         assert bmp == encode_bmp(f0)
         bmp1 = get(base + "/frame1.bmp")
         assert bmp1 == encode_bmp(f1)
+        # live spectrum line view (reference spectrum.qml equivalent)
+        gui.push_spectrum(0, np.linspace(0.0, 2.0, 4096))
+        sj = json.loads(get(base + "/spectrum0.json"))
+        assert len(sj["values"]) == 1024  # downsampled
+        assert abs(sj["values"][0] - 0.000733) < 1e-3
+        assert sj["values"][-1] > 1.9
         # latest-frame-wins (drop-under-load semantics)
         f0b = np.full((8, 16), 0xFF010203, dtype=np.uint32)
         gui.push_frame(0, f0b)
@@ -102,6 +108,7 @@ gui_pixmap_height = 32
                 st = json.loads(get(base + "/status.json"))
                 if st.get("streams"):
                     captured["bmp"] = get(base + "/frame0.bmp")
+                    captured["spec"] = get(base + "/spectrum0.json")
                     captured["status"] = st
                     return
             except (urllib.error.URLError, OSError):
@@ -115,6 +122,8 @@ gui_pixmap_height = 32
     t.join()
     assert rc == 0
     assert "bmp" in captured, "no live frame observed during the run"
+    spec = json.loads(captured["spec"])
+    assert len(spec["values"]) >= 32  # live spectrum line served too
     bmp = captured["bmp"]
     assert bmp[:2] == b"BM"
     _, w, h = struct.unpack_from("<Iii", bmp, 14)
