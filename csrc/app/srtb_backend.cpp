@@ -417,6 +417,29 @@ int main(int argc, char** argv) {
   srtb_app::WritePool writers(2);
   const std::string out_prefix = cfg.baseband_output_file_prefix;
 
+  // baseband_write_all: record every block minus the overlap tail into one
+  // rolling file per rank (the reference replaces the write_signal tail
+  // with write_file_pipe, write_file_pipe.hpp:41-94).  Appends go to the
+  // pool as pwrite-at-offset so two writer threads can't reorder them.
+  int write_all_fd = -1;
+  size_t wa_valid = 0;
+  uint64_t wa_index = 0;
+  if (cfg.baseband_write_all) {
+    size_t res_b = reserved * (size_t)n_streams *
+                   (size_t)std::abs(cfg.baseband_input_bits) / 8;
+    if (res_b >= raw_bytes) res_b = 0;
+    wa_valid = raw_bytes - res_b;
+    const std::string path = out_prefix + "all_r" +
+                             std::to_string(comm.rank()) + ".bin";
+    write_all_fd = ::open(path.c_str(), O_WRONLY | O_CREAT | O_TRUNC, 0644);
+    if (write_all_fd < 0) {
+      SRTB_APP_LOGE("cannot open " << path);
+      return 3;
+    }
+    SRTB_APP_LOGI("baseband_write_all -> " << path << " (" << wa_valid
+                                           << " valid bytes/block)");
+  }
+
   // async dump staging, lazily pinned on first detection
   std::vector<std::array<DumpStaging, 2>> staging(n_streams);
 
@@ -537,6 +560,31 @@ int main(int argc, char** argv) {
     for (int si = 0; si < n_streams; ++si)
       res[si] = engines[si]->wait(w.slots[si]);
     ++blocks;
+    if (write_all_fd >= 0) {
+      // record-everything tail: append the valid region and release the
+      // buffer from the pool; detections are still counted/logged below
+      // by the normal gate, but product dumps are replaced by the file
+      const uint64_t off = wa_index++ * (uint64_t)wa_valid;
+      uint8_t* p = bufs[w.buf_index];
+      const int bi = w.buf_index;
+      const size_t nb = wa_valid;
+      const int fd = write_all_fd;
+      writers.post([&q_free, p, off, nb, bi, fd] {
+        size_t done = 0;
+        while (done < nb) {
+          const ssize_t k =
+              ::pwrite(fd, p + done, nb - done, (off_t)(off + done));
+          if (k <= 0) break;
+          done += (size_t)k;
+        }
+        q_free.put(bi);
+      });
+      bool any = false;
+      for (int si = 0; si < n_streams; ++si)
+        for (auto& [len, cnt] : res[si].counts) any |= cnt > 0;
+      if (any) ++detections;
+      return;
+    }
     if (have_last && w.counter > last_counter)
       block_delta = w.counter - last_counter;
     last_counter = w.counter;
@@ -595,6 +643,11 @@ int main(int argc, char** argv) {
   }
   while (!inflight.empty()) drain_one();
   if (held_buf >= 0) q_free.put(held_buf);
+  if (write_all_fd >= 0) {
+    writers.drain();  // all appends on disk before fdatasync+close
+    ::fdatasync(write_all_fd);
+    ::close(write_all_fd);
+  }
   stop.store(true);
   for (auto& t : input_threads)
     if (t.joinable()) t.join();

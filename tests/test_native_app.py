@@ -395,6 +395,38 @@ def test_srtb_backend_rccl_comm_single_rank(tmp_path):
 
 
 @pytest.mark.gpu
+def test_srtb_backend_write_all(tmp_path):
+    """baseband_write_all=1 replaces the product tail with one rolling
+    file holding every block minus the overlap (reference
+    write_file_pipe.hpp:41-94)."""
+    rng = np.random.default_rng(2)
+    n = 1 << 17
+    raw = np.clip(np.round(rng.normal(0, 8, 2 * n)), -128,
+                  127).astype(np.int8).view(np.uint8)
+    rec = tmp_path / "wa.bin"
+    raw.tofile(rec)
+    out = subprocess.run(
+        [BACKEND, "--input_file_path", str(rec),
+         "--baseband_input_count", str(n),
+         "--baseband_input_bits", "-8",
+         "--spectrum_channel_count", "64",
+         "--baseband_freq_low", "1400", "--baseband_bandwidth", "64",
+         "--baseband_sample_rate", "128e6", "--dm", "0",
+         "--baseband_reserve_sample", "0", "--baseband_write_all", "1",
+         "--mitigate_rfi_average_method_threshold", "1e30",
+         "--mitigate_rfi_spectral_kurtosis_threshold", "1e30",
+         "--signal_detect_signal_noise_threshold", "1e30",
+         "--baseband_output_file_prefix", str(tmp_path) + "/wa_"],
+        capture_output=True, text=True, timeout=300)
+    assert out.returncode == 0, out.stderr
+    data = np.fromfile(tmp_path / "wa_all_r0.bin", dtype=np.uint8)
+    np.testing.assert_array_equal(data, raw)
+    # the product tail is replaced: no per-detection files
+    import glob
+    assert not glob.glob(str(tmp_path / "wa_*.npy"))
+
+
+@pytest.mark.gpu
 def test_srtb_correlator_native(tmp_path):
     rng = np.random.default_rng(1)
     n = 1 << 14
