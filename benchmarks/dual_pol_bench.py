@@ -60,21 +60,29 @@ def main():
 
     counts = torch.zeros(2, dtype=torch.int64, device="cuda")
 
-    def step():
-        results = pipe.process_block(raw)
+    def agg(results):
         c = [sum(cnt for _, cnt in r["counts"]) for r in results]
         counts.copy_(torch.tensor(c, dtype=torch.int64))
         if world > 1:
             torch.distributed.all_reduce(counts)
 
-    for _ in range(args.warmup):
-        step()
+    def run(k):
+        # keep both engine slots in flight: block i+1's fan-out + chains
+        # enqueue while block i drains
+        inflight = []
+        for _ in range(k):
+            inflight.append(pipe.submit_block(raw))
+            if len(inflight) >= 2:
+                agg(pipe.wait_block(inflight.pop(0)))
+        while inflight:
+            agg(pipe.wait_block(inflight.pop(0)))
+
+    run(args.warmup)
     if world > 1:
         torch.distributed.barrier()
     torch.cuda.synchronize()
     t0 = time.perf_counter()
-    for _ in range(args.steps):
-        step()
+    run(args.steps)
     torch.cuda.synchronize()
     if world > 1:
         torch.distributed.barrier()

@@ -54,9 +54,12 @@ class DualPolPipeline:
             _make_engine(self.C, cfg, nsamps_reserved, nbits=-8),
         ]
 
-    def process_block(self, raw: np.ndarray) -> list[dict]:
-        """raw: packed bytes holding 2 * baseband_input_count int8 samples.
-        Returns one result dict per polarization."""
+    def submit_block(self, raw: np.ndarray) -> dict:
+        """Fan one packed block out and enqueue both polarization chains;
+        returns a handle for wait_block.  The handle owns the fan-out
+        tensors: they must stay alive until the engines drained them (the
+        caching allocator would otherwise recycle their memory under the
+        engine streams)."""
         torch = self.torch
         n = self.cfg.baseband_input_count
         raw_t = torch.from_numpy(np.ascontiguousarray(raw)).cuda()
@@ -64,16 +67,24 @@ class DualPolPipeline:
             pols = self.C.unpack_gznupsr_a1(raw_t, 2)
         else:
             pols = self.C.unpack_2pol(raw_t, self.kind)
-        out = []
         slots = []
         for eng, pol in zip(self.engines, pols):
             assert pol.numel() == n
             slots.append(eng.submit_samples(pol))
-        for eng, slot in zip(self.engines, slots):
+        return {"slots": slots, "pols": pols, "raw": raw_t}
+
+    def wait_block(self, handle: dict) -> list[dict]:
+        out = []
+        for eng, slot in zip(self.engines, handle["slots"]):
             res = eng.wait(slot)
             res["slot"] = slot
             out.append(res)
         return out
+
+    def process_block(self, raw: np.ndarray) -> list[dict]:
+        """raw: packed bytes holding 2 * baseband_input_count int8 samples.
+        Returns one result dict per polarization."""
+        return self.wait_block(self.submit_block(raw))
 
 
 @dataclass
